@@ -1,0 +1,301 @@
+"""Concrete splitters.
+
+Parity with reference replay/splitters/: RatioSplitter (ratio_splitter.py:13),
+LastNSplitter (last_n_splitter.py:24), TimeSplitter (time_splitter.py:20),
+RandomSplitter (random_splitter.py:6), NewUsersSplitter
+(new_users_splitter.py:12), ColdUserRandomSplitter
+(cold_user_random_splitter.py:17), RandomNextNSplitter
+(random_next_n_splitter.py:20), TwoStageSplitter (two_stage_splitter.py:17),
+KFolds (k_folds.py:16).
+"""
+
+from __future__ import annotations
+
+from typing import Iterator, Optional, Tuple
+
+import numpy as np
+import pandas as pd
+
+from .base_splitter import Splitter, SplitterReturnType
+
+
+class RatioSplitter(Splitter):
+    """Per-user temporal split: last ``test_size`` fraction of each user's
+    interactions go to test."""
+
+    _init_arg_names = Splitter._init_arg_names + ("test_size", "divide_column", "min_interactions_per_group")
+
+    def __init__(
+        self,
+        test_size: float = 0.2,
+        divide_column: str = "query_id",
+        min_interactions_per_group: Optional[int] = None,
+        split_by_fractions: bool = True,
+        **kwargs,
+    ) -> None:
+        super().__init__(**kwargs)
+        if not 0 < test_size < 1:
+            raise ValueError("test_size must be in (0, 1)")
+        self.test_size = test_size
+        self.divide_column = divide_column
+        self.min_interactions_per_group = min_interactions_per_group
+        self.split_by_fractions = split_by_fractions
+
+    def _core_split(self, interactions: pd.DataFrame) -> SplitterReturnType:
+        df = interactions.sort_values([self.divide_column, self.timestamp_column], kind="stable")
+        sizes = df.groupby(self.divide_column)[self.divide_column].transform("size")
+        pos = df.groupby(self.divide_column).cumcount()
+        if self.split_by_fractions:
+            n_test = np.floor(sizes * self.test_size).astype(int)
+        else:
+            n_test = np.minimum(sizes - 1, int(np.ceil(self.test_size)))
+        is_test = pos >= (sizes - n_test)
+        if self.min_interactions_per_group is not None:
+            is_test &= sizes >= self.min_interactions_per_group
+        is_test = self._recalculate_with_session_id_column(df, is_test)
+        return df[~is_test].sort_index(), df[is_test].sort_index()
+
+
+class LastNSplitter(Splitter):
+    """Last N interactions (or last N seconds) per user to test
+    (reference last_n_splitter.py:24)."""
+
+    _init_arg_names = Splitter._init_arg_names + ("N", "divide_column", "strategy")
+
+    def __init__(
+        self,
+        N: int = 1,
+        divide_column: str = "query_id",
+        strategy: str = "interactions",
+        **kwargs,
+    ) -> None:
+        super().__init__(**kwargs)
+        if strategy not in ("interactions", "timedelta"):
+            raise ValueError("strategy must be 'interactions' or 'timedelta'")
+        self.N = N
+        self.divide_column = divide_column
+        self.strategy = strategy
+
+    def _core_split(self, interactions: pd.DataFrame) -> SplitterReturnType:
+        df = interactions.sort_values([self.divide_column, self.timestamp_column], kind="stable")
+        if self.strategy == "interactions":
+            sizes = df.groupby(self.divide_column)[self.divide_column].transform("size")
+            pos = df.groupby(self.divide_column).cumcount()
+            is_test = pos >= (sizes - self.N)
+        else:
+            ts = df[self.timestamp_column]
+            if pd.api.types.is_datetime64_any_dtype(ts):
+                seconds = ts.astype("int64") // 10**9
+            else:
+                seconds = ts.astype("int64")
+            last = seconds.groupby(df[self.divide_column]).transform("max")
+            is_test = seconds > (last - self.N)
+        is_test = self._recalculate_with_session_id_column(df, is_test)
+        return df[~is_test].sort_index(), df[is_test].sort_index()
+
+
+class TimeSplitter(Splitter):
+    """Global time split at ``time_threshold`` (fraction or timestamp)
+    (reference time_splitter.py:20)."""
+
+    _init_arg_names = Splitter._init_arg_names + ("time_threshold",)
+
+    def __init__(self, time_threshold, time_column_format: str = "%Y-%m-%d %H:%M:%S", **kwargs) -> None:
+        super().__init__(**kwargs)
+        self.time_threshold = time_threshold
+        self.time_column_format = time_column_format
+
+    def _core_split(self, interactions: pd.DataFrame) -> SplitterReturnType:
+        ts = interactions[self.timestamp_column]
+        threshold = self.time_threshold
+        if isinstance(threshold, float) and 0 < threshold < 1:
+            sorted_ts = ts.sort_values()
+            threshold = sorted_ts.iloc[min(len(sorted_ts) - 1, int(len(sorted_ts) * (1 - threshold)))]
+            is_test = ts >= threshold
+        else:
+            if isinstance(threshold, str):
+                threshold = pd.to_datetime(threshold, format=self.time_column_format)
+                ts = pd.to_datetime(ts)
+            is_test = ts >= threshold
+        is_test = self._recalculate_with_session_id_column(interactions, is_test)
+        return interactions[~is_test], interactions[is_test]
+
+
+class RandomSplitter(Splitter):
+    """Uniform random row split (reference random_splitter.py:6)."""
+
+    _init_arg_names = Splitter._init_arg_names + ("test_size", "seed")
+
+    def __init__(self, test_size: float = 0.2, seed: Optional[int] = None, **kwargs) -> None:
+        super().__init__(**kwargs)
+        if not 0 <= test_size <= 1:
+            raise ValueError("test_size must be in [0, 1]")
+        self.test_size = test_size
+        self.seed = seed
+
+    def _core_split(self, interactions: pd.DataFrame) -> SplitterReturnType:
+        rng = np.random.default_rng(self.seed)
+        mask = rng.random(len(interactions)) < self.test_size
+        is_test = pd.Series(mask, index=interactions.index)
+        is_test = self._recalculate_with_session_id_column(interactions, is_test)
+        return interactions[~is_test], interactions[is_test]
+
+
+class NewUsersSplitter(Splitter):
+    """Users whose first interaction is in the last ``test_size`` share of the
+    timeline go entirely to test (reference new_users_splitter.py:12)."""
+
+    _init_arg_names = Splitter._init_arg_names + ("test_size",)
+
+    def __init__(self, test_size: float = 0.2, **kwargs) -> None:
+        super().__init__(**kwargs)
+        if not 0 < test_size < 1:
+            raise ValueError("test_size must be in (0, 1)")
+        self.test_size = test_size
+
+    def _core_split(self, interactions: pd.DataFrame) -> SplitterReturnType:
+        first_ts = interactions.groupby(self.query_column)[self.timestamp_column].transform("min")
+        threshold = first_ts.quantile(1 - self.test_size)
+        is_test = first_ts > threshold
+        is_test = self._recalculate_with_session_id_column(interactions, is_test)
+        return interactions[~is_test], interactions[is_test]
+
+
+class ColdUserRandomSplitter(Splitter):
+    """A random ``test_size`` fraction of users moves entirely to test
+    (reference cold_user_random_splitter.py:17)."""
+
+    _init_arg_names = Splitter._init_arg_names + ("test_size", "seed")
+
+    def __init__(self, test_size: float = 0.2, seed: Optional[int] = None, **kwargs) -> None:
+        super().__init__(**kwargs)
+        self.test_size = test_size
+        self.seed = seed
+
+    def _core_split(self, interactions: pd.DataFrame) -> SplitterReturnType:
+        users = interactions[self.query_column].unique()
+        rng = np.random.default_rng(self.seed)
+        test_users = set(rng.choice(users, size=int(len(users) * self.test_size), replace=False).tolist())
+        is_test = interactions[self.query_column].isin(test_users)
+        return interactions[~is_test], interactions[is_test]
+
+
+class RandomNextNSplitter(Splitter):
+    """Pick a random cut position per user; the following N interactions form
+    the test part (reference random_next_n_splitter.py:20)."""
+
+    _init_arg_names = Splitter._init_arg_names + ("N", "seed", "divide_column")
+
+    def __init__(self, N: int = 1, seed: Optional[int] = None, divide_column: str = "query_id", **kwargs) -> None:
+        super().__init__(**kwargs)
+        self.N = N
+        self.seed = seed
+        self.divide_column = divide_column
+
+    def _core_split(self, interactions: pd.DataFrame) -> SplitterReturnType:
+        df = interactions.sort_values([self.divide_column, self.timestamp_column], kind="stable")
+        rng = np.random.default_rng(self.seed)
+        sizes = df.groupby(self.divide_column)[self.divide_column].transform("size").to_numpy()
+        pos = df.groupby(self.divide_column).cumcount().to_numpy()
+        # one random cut per user, broadcast to rows
+        users, first_idx = np.unique(df[self.divide_column].to_numpy(), return_index=True)
+        cuts_per_user = {}
+        for u, fi in zip(users, first_idx):
+            size = sizes[fi]
+            cuts_per_user[u] = rng.integers(1, max(2, size))
+        cut = df[self.divide_column].map(cuts_per_user).to_numpy()
+        is_test = (pos >= cut) & (pos < cut + self.N)
+        is_test = pd.Series(is_test, index=df.index)
+        is_test = self._recalculate_with_session_id_column(df, is_test)
+        # interactions after cut+N are dropped (they are future relative to test)
+        keep = pd.Series(pos < cut + self.N, index=df.index)
+        df = df[keep]
+        is_test = is_test[keep]
+        return df[~is_test].sort_index(), df[is_test].sort_index()
+
+
+class TwoStageSplitter(Splitter):
+    """Split by users first (``second_divide_size`` share of users), then for
+    those users take ``first_divide_size`` of interactions to test
+    (reference two_stage_splitter.py:17)."""
+
+    _init_arg_names = Splitter._init_arg_names + ("second_divide_size", "first_divide_size", "seed")
+
+    def __init__(
+        self,
+        second_divide_size: float = 0.5,
+        first_divide_size: float = 0.5,
+        shuffle: bool = False,
+        seed: Optional[int] = None,
+        **kwargs,
+    ) -> None:
+        super().__init__(**kwargs)
+        self.second_divide_size = second_divide_size
+        self.first_divide_size = first_divide_size
+        self.shuffle = shuffle
+        self.seed = seed
+
+    def _core_split(self, interactions: pd.DataFrame) -> SplitterReturnType:
+        users = np.sort(interactions[self.query_column].unique())
+        rng = np.random.default_rng(self.seed)
+        n_test_users = (
+            int(self.second_divide_size)
+            if self.second_divide_size >= 1
+            else int(len(users) * self.second_divide_size)
+        )
+        test_users = set(rng.choice(users, size=n_test_users, replace=False).tolist())
+        in_test_users = interactions[self.query_column].isin(test_users)
+        df_test_users = interactions[in_test_users]
+        if self.shuffle:
+            df_test_users = df_test_users.sample(frac=1.0, random_state=self.seed)
+        else:
+            df_test_users = df_test_users.sort_values(
+                [self.query_column, self.timestamp_column], kind="stable"
+            )
+        sizes = df_test_users.groupby(self.query_column)[self.query_column].transform("size")
+        pos = df_test_users.groupby(self.query_column).cumcount()
+        if self.first_divide_size >= 1:
+            n_test = np.minimum(sizes - 1, int(self.first_divide_size))
+        else:
+            n_test = np.floor(sizes * self.first_divide_size).astype(int)
+        is_test = pos >= (sizes - n_test)
+        test = df_test_users[is_test]
+        train = pd.concat([interactions[~in_test_users], df_test_users[~is_test]]).sort_index()
+        return train, test.sort_index()
+
+
+class KFolds:
+    """K-fold user-wise random splitter (reference k_folds.py:16).
+
+    Yields (train, test) pairs; fold assignment is per interaction within a
+    user (strategy='query').
+    """
+
+    def __init__(
+        self,
+        n_folds: int = 5,
+        strategy: str = "query",
+        seed: Optional[int] = None,
+        query_column: str = "query_id",
+        item_column: str = "item_id",
+        timestamp_column: str = "timestamp",
+        session_id_column: Optional[str] = None,
+        session_id_processing_strategy: str = "test",
+    ) -> None:
+        if strategy not in ("query",):
+            raise ValueError("strategy must be 'query'")
+        self.n_folds = n_folds
+        self.strategy = strategy
+        self.seed = seed
+        self.query_column = query_column
+
+    def split(self, interactions: pd.DataFrame) -> Iterator[Tuple[pd.DataFrame, pd.DataFrame]]:
+        rng = np.random.default_rng(self.seed)
+        fold = np.empty(len(interactions), dtype=np.int64)
+        df = interactions.reset_index(drop=True)
+        for _, group in df.groupby(self.query_column):
+            idx = group.index.to_numpy()
+            fold[idx] = rng.permutation(len(idx)) % self.n_folds
+        for k in range(self.n_folds):
+            mask = fold == k
+            yield interactions[~mask], interactions[mask]

@@ -1,0 +1,124 @@
+import numpy as np
+import pandas as pd
+import pytest
+
+from replay_amd.splitters import (
+    ColdUserRandomSplitter,
+    KFolds,
+    LastNSplitter,
+    NewUsersSplitter,
+    RandomNextNSplitter,
+    RandomSplitter,
+    RatioSplitter,
+    TimeSplitter,
+    TwoStageSplitter,
+)
+
+pytestmark = pytest.mark.core
+
+
+def test_ratio_splitter(interactions_pandas):
+    train, test = RatioSplitter(test_size=0.5).split(interactions_pandas)
+    assert len(train) + len(test) == len(interactions_pandas)
+    # per-user: last half goes to test
+    u3_test = test[test["query_id"] == 3]["timestamp"]
+    u3_train = train[train["query_id"] == 3]["timestamp"]
+    assert u3_train.max() < u3_test.min()
+
+
+def test_last_n_splitter(interactions_pandas):
+    train, test = LastNSplitter(N=1).split(interactions_pandas)
+    assert len(test) == interactions_pandas["query_id"].nunique()
+    merged = test.groupby("query_id")["timestamp"].max()
+    orig = interactions_pandas.groupby("query_id")["timestamp"].max()
+    assert (merged == orig).all()
+
+
+def test_last_n_splitter_timedelta(interactions_pandas):
+    train, test = LastNSplitter(N=100, strategy="timedelta").split(interactions_pandas)
+    assert len(train) + len(test) == len(interactions_pandas)
+    assert len(test) >= interactions_pandas["query_id"].nunique()
+
+
+def test_time_splitter(interactions_pandas):
+    train, test = TimeSplitter(time_threshold=300).split(interactions_pandas)
+    assert (train["timestamp"] < 300).all()
+    assert (test["timestamp"] >= 300).all()
+
+
+def test_time_splitter_fraction(interactions_pandas):
+    train, test = TimeSplitter(time_threshold=0.3).split(interactions_pandas)
+    assert len(test) >= 1
+    assert train["timestamp"].max() <= test["timestamp"].min()
+
+
+def test_random_splitter(interactions_pandas):
+    train, test = RandomSplitter(test_size=0.3, seed=42).split(interactions_pandas)
+    assert len(train) + len(test) == len(interactions_pandas)
+    train2, test2 = RandomSplitter(test_size=0.3, seed=42).split(interactions_pandas)
+    assert train.index.tolist() == train2.index.tolist()
+
+
+def test_new_users_splitter(interactions_pandas):
+    train, test = NewUsersSplitter(test_size=0.25).split(interactions_pandas)
+    assert set(test["query_id"]).isdisjoint(set(train["query_id"]))
+
+
+def test_cold_user_random_splitter(interactions_pandas):
+    train, test = ColdUserRandomSplitter(test_size=0.5, seed=1).split(interactions_pandas)
+    assert set(test["query_id"]).isdisjoint(set(train["query_id"]))
+    assert len(set(test["query_id"])) == 2
+
+
+def test_random_next_n_splitter(interactions_pandas):
+    train, test = RandomNextNSplitter(N=1, seed=0).split(interactions_pandas)
+    assert len(test) <= interactions_pandas["query_id"].nunique()
+    assert len(test) > 0
+
+
+def test_two_stage_splitter(interactions_pandas):
+    train, test = TwoStageSplitter(second_divide_size=0.5, first_divide_size=0.5, seed=3).split(
+        interactions_pandas
+    )
+    assert len(train) + len(test) == len(interactions_pandas)
+    assert 0 < len(set(test["query_id"])) <= 2
+
+
+def test_kfolds(interactions_pandas):
+    folds = list(KFolds(n_folds=2, seed=0, query_column="query_id").split(interactions_pandas))
+    assert len(folds) == 2
+    total_test = sum(len(test) for _, test in folds)
+    assert total_test == len(interactions_pandas)
+
+
+def test_drop_cold_items(interactions_pandas):
+    train, test = LastNSplitter(N=1, drop_cold_items=True).split(interactions_pandas)
+    assert set(test["item_id"]).issubset(set(train["item_id"]))
+
+
+def test_splitter_save_load(tmp_path, interactions_pandas):
+    splitter = RatioSplitter(test_size=0.4)
+    splitter.save(tmp_path / "s")
+    from replay_amd.splitters import Splitter
+
+    loaded = Splitter.load(tmp_path / "s")
+    assert isinstance(loaded, RatioSplitter)
+    assert loaded.test_size == 0.4
+    t1, v1 = splitter.split(interactions_pandas)
+    t2, v2 = loaded.split(interactions_pandas)
+    assert t1.index.tolist() == t2.index.tolist()
+
+
+def test_session_id_strategy():
+    df = pd.DataFrame(
+        {
+            "query_id": [1, 1, 1, 1],
+            "item_id": [1, 2, 3, 4],
+            "timestamp": [1, 2, 3, 4],
+            "session_id": [0, 0, 1, 1],
+        }
+    )
+    train, test = RatioSplitter(test_size=0.5, session_id_column="session_id").split(df)
+    # session 1 has test rows -> whole session 1 in test
+    assert set(test["session_id"]) == {1}
+    assert set(train["session_id"]) == {0}
