@@ -1,0 +1,141 @@
+"""Item-item KNN with modified cosine similarity.
+
+Parity with reference ItemKNN (replay/models/knn.py:15): tf-idf / bm25
+reweighting (reference knn.py:92-156), item-item dot products (:158-198 — the
+reference does a Spark self-join; here it is a scipy CSR ``A.T @ A``), top-k
+neighbour truncation (:197-220) and the NeighbourRec predict
+(base_neighbour_rec.py:55-96: interactions x similarity -> groupby-sum).
+"""
+
+from __future__ import annotations
+
+from typing import Dict, Optional
+
+import numpy as np
+import pandas as pd
+from scipy.sparse import csr_matrix
+
+from .base_rec import Recommender
+
+
+class NeighbourRec(Recommender):
+    """Base for models that keep an item-to-item similarity frame
+    (reference replay/models/base_neighbour_rec.py:23)."""
+
+    similarity: Optional[pd.DataFrame] = None
+
+    @property
+    def _dataframes(self):
+        return {"similarity": self.similarity}
+
+    def _predict_by_similarity(self, dataset, k, queries, items) -> pd.DataFrame:
+        """recs(q, j) = sum over seen i of sim(i, j)
+        (reference base_neighbour_rec.py:55-96)."""
+        inter = dataset.interactions[[self.query_column, self.item_column]]
+        inter = inter.merge(queries, on=self.query_column)
+        joined = inter.merge(
+            self.similarity.rename(columns={"item_idx_one": self.item_column}), on=self.item_column
+        )
+        joined = joined.rename(columns={"item_idx_two": "rec_item"})
+        scores = (
+            joined.groupby([self.query_column, "rec_item"])["similarity"].sum().rename(self.rating_column)
+        ).reset_index()
+        scores = scores.rename(columns={"rec_item": self.item_column})
+        scores = scores.merge(items, on=self.item_column)
+        return scores
+
+    def _predict(self, dataset, k, queries, items, filter_seen_items=True) -> pd.DataFrame:
+        return self._predict_by_similarity(dataset, k, queries, items)
+
+    def get_nearest_items(self, items, k: int, metric: Optional[str] = None) -> pd.DataFrame:
+        sim = self.similarity
+        sel = sim[sim["item_idx_one"].isin(set(items))]
+        sel = sel.sort_values(["item_idx_one", "similarity"], ascending=[True, False], kind="stable")
+        return sel.groupby("item_idx_one", sort=False).head(k).reset_index(drop=True)
+
+
+class ItemKNN(NeighbourRec):
+    """Item-based KNN (reference replay/models/knn.py:15)."""
+
+    def __init__(
+        self,
+        num_neighbours: int = 10,
+        use_rating: bool = False,
+        shrink: float = 0.0,
+        weighting: Optional[str] = None,
+    ) -> None:
+        super().__init__()
+        if weighting not in (None, "tf_idf", "bm25"):
+            raise ValueError("weighting must be None, 'tf_idf' or 'bm25'")
+        self.num_neighbours = num_neighbours
+        self.use_rating = use_rating
+        self.shrink = shrink
+        self.weighting = weighting
+        self.bm25_k1 = 1.2
+        self.bm25_b = 0.75
+
+    @property
+    def _init_args(self):
+        return {
+            "num_neighbours": self.num_neighbours,
+            "use_rating": self.use_rating,
+            "shrink": self.shrink,
+            "weighting": self.weighting,
+        }
+
+    _search_space = {
+        "num_neighbours": {"type": "int", "args": [1, 100]},
+        "shrink": {"type": "int", "args": [0, 100]},
+        "weighting": {"type": "categorical", "args": [None, "tf_idf", "bm25"]},
+    }
+
+    def _build_matrix(self, dataset) -> csr_matrix:
+        inter = dataset.interactions
+        rows = inter[self.query_column].to_numpy(dtype=np.int64)
+        cols = inter[self.item_column].to_numpy(dtype=np.int64)
+        if self.use_rating and self.rating_column in inter.columns:
+            data = inter[self.rating_column].to_numpy(dtype=np.float64)
+        else:
+            data = np.ones(len(inter), dtype=np.float64)
+        return csr_matrix((data, (rows, cols)), shape=(self._query_dim_size, self._item_dim_size))
+
+    def _reweight(self, mat: csr_matrix) -> csr_matrix:
+        """tf-idf / bm25 over the user axis (reference knn.py:92-156)."""
+        if self.weighting is None:
+            return mat
+        n_users = mat.shape[0]
+        df_item = np.asarray((mat > 0).sum(axis=0)).ravel()  # users per item? no: df over users
+        if self.weighting == "tf_idf":
+            # idf over users: how many items each user has is the doc length;
+            # reference weights a user-item cell by idf of the USER frequency
+            n_items_per_user = np.asarray((mat > 0).sum(axis=1)).ravel()
+            idf = np.log1p(mat.shape[1] / np.maximum(1, n_items_per_user))
+            d = mat.tocoo()
+            data = d.data * idf[d.row]
+            return csr_matrix((data, (d.row, d.col)), shape=mat.shape)
+        # bm25 over users
+        k1, b = self.bm25_k1, self.bm25_b
+        n_users_per_item = np.asarray((mat > 0).sum(axis=0)).ravel()
+        idf = np.log1p((n_users - n_users_per_item + 0.5) / (n_users_per_item + 0.5))
+        doc_len = np.asarray(mat.sum(axis=1)).ravel()
+        avg_len = doc_len.mean() if len(doc_len) else 1.0
+        d = mat.tocoo()
+        tf = d.data
+        denom = tf + k1 * (1 - b + b * doc_len[d.row] / max(avg_len, 1e-9))
+        data = idf[d.col] * tf * (k1 + 1) / np.maximum(denom, 1e-12)
+        return csr_matrix((data, (d.row, d.col)), shape=mat.shape)
+
+    def _fit(self, dataset) -> None:
+        mat = self._reweight(self._build_matrix(dataset))
+        dot = (mat.T @ mat).tocoo()  # item-item co-occurrence dot products
+        norms = np.sqrt(np.asarray(mat.multiply(mat).sum(axis=0)).ravel())
+        rows, cols, data = dot.row, dot.col, dot.data
+        off = rows != cols
+        rows, cols, data = rows[off], cols[off], data[off]
+        denom = norms[rows] * norms[cols] + self.shrink + 1e-12
+        sim = data / denom
+        df = pd.DataFrame({"item_idx_one": rows, "item_idx_two": cols, "similarity": sim})
+        df = df.sort_values(["item_idx_one", "similarity"], ascending=[True, False], kind="stable")
+        self.similarity = (
+            df.groupby("item_idx_one", sort=False).head(self.num_neighbours).reset_index(drop=True)
+        )
