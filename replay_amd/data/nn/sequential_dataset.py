@@ -1,0 +1,98 @@
+"""Per-query sequence stores.
+
+Parity with reference replay/data/nn/sequential_dataset.py
+(SequentialDataset:18, PandasSequentialDataset:142,
+keep_common_query_ids:91): random access to per-user sequences matched to a
+TensorSchema.
+"""
+
+from __future__ import annotations
+
+from typing import Tuple
+
+import numpy as np
+import pandas as pd
+
+from .schema import TensorSchema
+
+
+class SequentialDataset:
+    """Abstract per-query sequence store."""
+
+    def __len__(self) -> int:  # pragma: no cover
+        raise NotImplementedError
+
+    def get_query_id(self, index: int):  # pragma: no cover
+        raise NotImplementedError
+
+    def get_sequence(self, index: int, feature_name: str) -> np.ndarray:  # pragma: no cover
+        raise NotImplementedError
+
+    def get_sequence_length(self, index: int) -> int:  # pragma: no cover
+        raise NotImplementedError
+
+    @property
+    def schema(self) -> TensorSchema:  # pragma: no cover
+        raise NotImplementedError
+
+    @staticmethod
+    def keep_common_query_ids(
+        lhs: "SequentialDataset", rhs: "SequentialDataset"
+    ) -> Tuple["SequentialDataset", "SequentialDataset"]:
+        """Restrict both datasets to the intersection of query ids
+        (reference sequential_dataset.py:91)."""
+        lhs_ids = set(lhs.get_all_query_ids().tolist())
+        rhs_ids = set(rhs.get_all_query_ids().tolist())
+        common = lhs_ids & rhs_ids
+        return lhs.filter_by_query_ids(common), rhs.filter_by_query_ids(common)
+
+
+class PandasSequentialDataset(SequentialDataset):
+    """Sequences in a pandas frame: one row per query, cells are np arrays."""
+
+    def __init__(self, tensor_schema: TensorSchema, query_id_column: str, item_id_column: str, sequences: pd.DataFrame) -> None:
+        self._schema = tensor_schema
+        self._query_id_column = query_id_column
+        self._item_id_column = item_id_column
+        self._sequences = sequences.reset_index(drop=True)
+        for feature in tensor_schema.all_features:
+            if feature.is_seq and feature.name not in sequences.columns:
+                raise ValueError(f"Sequence column {feature.name} missing from sequences frame")
+
+    def __len__(self) -> int:
+        return len(self._sequences)
+
+    @property
+    def schema(self) -> TensorSchema:
+        return self._schema
+
+    @property
+    def sequences(self) -> pd.DataFrame:
+        return self._sequences
+
+    def get_query_id(self, index: int):
+        return self._sequences[self._query_id_column].iloc[index]
+
+    def get_all_query_ids(self) -> np.ndarray:
+        return self._sequences[self._query_id_column].to_numpy()
+
+    def get_sequence(self, index: int, feature_name: str) -> np.ndarray:
+        return np.asarray(self._sequences[feature_name].iloc[index])
+
+    def get_sequence_by_query_id(self, query_id, feature_name: str) -> np.ndarray:
+        rows = self._sequences[self._sequences[self._query_id_column] == query_id]
+        if not len(rows):
+            return np.array([], dtype=np.int64)
+        return np.asarray(rows[feature_name].iloc[0])
+
+    def get_sequence_length(self, index: int) -> int:
+        return len(self.get_sequence(index, self._item_id_column))
+
+    def get_max_sequence_length(self) -> int:
+        return max((self.get_sequence_length(i) for i in range(len(self))), default=0)
+
+    def filter_by_query_ids(self, query_ids) -> "PandasSequentialDataset":
+        mask = self._sequences[self._query_id_column].isin(set(query_ids))
+        return PandasSequentialDataset(
+            self._schema, self._query_id_column, self._item_id_column, self._sequences[mask]
+        )

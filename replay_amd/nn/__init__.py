@@ -1,0 +1,29 @@
+from .agg import ConcatAggregator, SumAggregator
+from .attention import MultiheadAttention, MultiHeadDifferentialAttention
+from .embedding import (
+    CategoricalEmbedding,
+    CategoricalListEmbedding,
+    IdentityEmbedding,
+    NumericalEmbedding,
+    SequenceEmbedding,
+)
+from .ffn import PointWiseFeedForward, SwiGLU, SwiGLUEncoder
+from .head import EmbeddingTyingHead
+from .mask import DefaultAttentionMask
+
+__all__ = [
+    "ConcatAggregator",
+    "SumAggregator",
+    "MultiheadAttention",
+    "MultiHeadDifferentialAttention",
+    "CategoricalEmbedding",
+    "CategoricalListEmbedding",
+    "IdentityEmbedding",
+    "NumericalEmbedding",
+    "SequenceEmbedding",
+    "PointWiseFeedForward",
+    "SwiGLU",
+    "SwiGLUEncoder",
+    "EmbeddingTyingHead",
+    "DefaultAttentionMask",
+]

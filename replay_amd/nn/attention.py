@@ -1,0 +1,154 @@
+"""Attention modules.
+
+``MultiheadAttention`` here is this framework's own implementation (same math
+as the reference's use of torch.nn.MultiheadAttention inside
+replay/nn/sequential/sasrec/transformer.py:38,99-106): QKV projection GEMMs +
+scores + additive mask + softmax + PV + out projection.  It is written so the
+scores->softmax->PV middle runs through replay_amd.ops (HIP flash-style
+fused attention on GPU, K1/K2 in SURVEY §2.12).
+
+``MultiHeadDifferentialAttention`` gives parity with reference
+replay/nn/attention.py:7 (dual softmax attentions subtracted with a learned
+lambda, per-head RMSNorm, reference :67-157).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+
+
+def _attention_core(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    attn_mask: Optional[torch.Tensor],
+    dropout_p: float,
+    training: bool,
+) -> torch.Tensor:
+    """scores -> +mask -> softmax -> PV.  q,k,v: [B*H, L, Dh];
+    attn_mask additive float [B*H, L, L] or None."""
+    from replay_amd.ops.attention import fused_attention
+
+    return fused_attention(q, k, v, attn_mask, dropout_p if training else 0.0)
+
+
+class MultiheadAttention(torch.nn.Module):
+    """Batch-first multi-head self-attention with additive float mask."""
+
+    def __init__(self, embed_dim: int, num_heads: int, dropout: float = 0.0, bias: bool = True) -> None:
+        super().__init__()
+        if embed_dim % num_heads != 0:
+            raise ValueError("embed_dim must divide num_heads")
+        self.embed_dim = embed_dim
+        self.num_heads = num_heads
+        self.head_dim = embed_dim // num_heads
+        self.dropout = dropout
+        self.in_proj = torch.nn.Linear(embed_dim, 3 * embed_dim, bias=bias)
+        self.out_proj = torch.nn.Linear(embed_dim, embed_dim, bias=bias)
+
+    def forward(
+        self,
+        x: torch.Tensor,
+        attn_mask: Optional[torch.Tensor] = None,
+        key_padding_mask: Optional[torch.Tensor] = None,
+    ) -> torch.Tensor:
+        """x: [B, L, E]; attn_mask: [B*H, L, L] additive float;
+        key_padding_mask: [B, L] bool, True = PAD (torch convention)."""
+        B, L, E = x.shape
+        H, Dh = self.num_heads, self.head_dim
+        qkv = self.in_proj(x)  # [B, L, 3E]
+        q, k, v = qkv.chunk(3, dim=-1)
+
+        def split(t):
+            return t.view(B, L, H, Dh).transpose(1, 2).reshape(B * H, L, Dh)
+
+        q, k, v = split(q), split(k), split(v)
+        if key_padding_mask is not None:
+            kp = torch.zeros(B, L, dtype=q.dtype, device=q.device)
+            fill = float("-inf") if self.training else torch.finfo(torch.float32).min
+            kp = kp.masked_fill(key_padding_mask, fill)
+            kp = kp[:, None, None, :].expand(B, H, L, L).reshape(B * H, L, L)
+            attn_mask = kp if attn_mask is None else attn_mask + kp
+        out = _attention_core(q, k, v, attn_mask, self.dropout, self.training)
+        out = out.view(B, H, L, Dh).transpose(1, 2).reshape(B, L, E)
+        return self.out_proj(out)
+
+
+class RMSNormPerHead(torch.nn.Module):
+    def __init__(self, head_dim: int, eps: float = 1e-5) -> None:
+        super().__init__()
+        self.eps = eps
+        self.weight = torch.nn.Parameter(torch.ones(head_dim))
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        norm = x * torch.rsqrt(x.pow(2).mean(-1, keepdim=True) + self.eps)
+        return norm * self.weight
+
+
+class MultiHeadDifferentialAttention(torch.nn.Module):
+    """Differential attention (reference replay/nn/attention.py:7)."""
+
+    def __init__(
+        self,
+        embed_dim: int,
+        num_heads: int,
+        lambda_init: float = 0.8,
+        dropout: float = 0.0,
+        depth: int = 1,
+    ) -> None:
+        super().__init__()
+        if embed_dim % (2 * num_heads) != 0:
+            raise ValueError("embed_dim must divide 2*num_heads")
+        self.embed_dim = embed_dim
+        self.num_heads = num_heads
+        self.head_dim = embed_dim // (2 * num_heads)
+        self.dropout = dropout
+        self.q_proj = torch.nn.Linear(embed_dim, embed_dim, bias=False)
+        self.k_proj = torch.nn.Linear(embed_dim, embed_dim, bias=False)
+        self.v_proj = torch.nn.Linear(embed_dim, embed_dim, bias=False)
+        self.out_proj = torch.nn.Linear(embed_dim, embed_dim, bias=False)
+        self.lambda_init = lambda_init - 0.6 * math.exp(-0.3 * (depth - 1))
+        d = self.head_dim
+        self.lambda_q1 = torch.nn.Parameter(torch.randn(d) * 0.1)
+        self.lambda_k1 = torch.nn.Parameter(torch.randn(d) * 0.1)
+        self.lambda_q2 = torch.nn.Parameter(torch.randn(d) * 0.1)
+        self.lambda_k2 = torch.nn.Parameter(torch.randn(d) * 0.1)
+        self.subln = RMSNormPerHead(2 * self.head_dim)
+
+    def forward(
+        self,
+        x: torch.Tensor,
+        attn_mask: Optional[torch.Tensor] = None,
+        key_padding_mask: Optional[torch.Tensor] = None,
+    ) -> torch.Tensor:
+        B, L, E = x.shape
+        H, Dh = self.num_heads, self.head_dim
+        q = self.q_proj(x).view(B, L, 2 * H, Dh).transpose(1, 2)  # [B, 2H, L, Dh]
+        k = self.k_proj(x).view(B, L, 2 * H, Dh).transpose(1, 2)
+        v = self.v_proj(x).view(B, L, H, 2 * Dh).transpose(1, 2)  # [B, H, L, 2Dh]
+
+        scores = q @ k.transpose(-1, -2) / math.sqrt(Dh)  # [B, 2H, L, L]
+        if key_padding_mask is not None:
+            fill = float("-inf") if self.training else torch.finfo(torch.float32).min
+            scores = scores.masked_fill(key_padding_mask[:, None, None, :], fill)
+        if attn_mask is not None:
+            if attn_mask.dim() == 3 and attn_mask.shape[0] == B * H:
+                am = attn_mask.view(B, H, L, L).repeat_interleave(2, dim=1)
+            else:
+                am = attn_mask
+            scores = scores + am
+        probs = torch.softmax(scores, dim=-1)
+        probs = torch.nan_to_num(probs, nan=0.0)
+        probs = probs.view(B, H, 2, L, L)
+        lam1 = torch.exp((self.lambda_q1 * self.lambda_k1).sum())
+        lam2 = torch.exp((self.lambda_q2 * self.lambda_k2).sum())
+        lam = lam1 - lam2 + self.lambda_init
+        attn = probs[:, :, 0] - lam * probs[:, :, 1]  # [B, H, L, L]
+        attn = torch.nn.functional.dropout(attn, self.dropout, self.training)
+        out = attn @ v  # [B, H, L, 2Dh]
+        out = self.subln(out) * (1.0 - self.lambda_init)
+        out = out.transpose(1, 2).reshape(B, L, E)
+        return self.out_proj(out)

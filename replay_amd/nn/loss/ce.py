@@ -1,0 +1,116 @@
+"""Cross-entropy losses.
+
+Parity with reference replay/nn/loss/ce.py (CE:10 full-softmax over [B,L,V]
+with pad ignore :49-82; CEWeighted:84; CESampled:146 sampled
+positives+negatives with collision masking; CESampledWeighted:252).
+
+MI355X note: full-softmax CE at catalog scale is K10 in SURVEY §2.12 (chunked
+online-logsumexp HIP kernel); the sampled variant is K9 (fused
+gather+dot+logsumexp).  The eager forms below are the numerics reference.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from .base import LossBase, SampledLossBase
+
+
+class CE(LossBase):
+    """Full-softmax cross entropy over the catalog."""
+
+    def forward(
+        self,
+        embeddings: torch.Tensor,  # [B, L, E]
+        positive_labels: torch.Tensor,  # [B, L]
+        padding_mask: torch.Tensor,  # [B, L] bool True=valid
+        target_padding_mask: Optional[torch.Tensor] = None,
+        negative_labels: Optional[torch.Tensor] = None,
+        weights: Optional[torch.Tensor] = None,
+    ) -> torch.Tensor:
+        logits = self.logits_callback(embeddings)  # [B, L, V]
+        mask = target_padding_mask if target_padding_mask is not None else padding_mask
+        labels = positive_labels.masked_fill(~mask, -100)
+        return torch.nn.functional.cross_entropy(
+            logits.reshape(-1, logits.shape[-1]).float(),
+            labels.reshape(-1),
+            ignore_index=-100,
+        )
+
+
+class CEWeighted(LossBase):
+    """Per-position weighted full CE (reference ce.py:84)."""
+
+    def forward(
+        self,
+        embeddings: torch.Tensor,
+        positive_labels: torch.Tensor,
+        padding_mask: torch.Tensor,
+        target_padding_mask: Optional[torch.Tensor] = None,
+        negative_labels: Optional[torch.Tensor] = None,
+        weights: Optional[torch.Tensor] = None,
+    ) -> torch.Tensor:
+        logits = self.logits_callback(embeddings)
+        mask = target_padding_mask if target_padding_mask is not None else padding_mask
+        labels = positive_labels.masked_fill(~mask, -100)
+        per_pos = torch.nn.functional.cross_entropy(
+            logits.reshape(-1, logits.shape[-1]).float(),
+            labels.reshape(-1),
+            ignore_index=-100,
+            reduction="none",
+        ).reshape(labels.shape)
+        if weights is None:
+            weights = torch.ones_like(per_pos)
+        weights = weights * mask.to(per_pos.dtype)
+        return (per_pos * weights).sum() / weights.sum().clamp(min=1e-12)
+
+
+class CESampled(SampledLossBase):
+    """CE over [positive | negatives] sampled logits (reference ce.py:146),
+    with optional uniform-sampling log-correction matching the legacy sampled
+    CE (reference models/nn/sequential/sasrec/lightning.py:357-381:
+    neg logits corrected by +log(V-1) - log(n_negatives))."""
+
+    def __init__(self, log_correction: bool = False, vocab_size: Optional[int] = None) -> None:
+        super().__init__()
+        self.log_correction = log_correction
+        self.vocab_size = vocab_size
+
+    def forward(
+        self,
+        embeddings: torch.Tensor,
+        positive_labels: torch.Tensor,
+        padding_mask: torch.Tensor,
+        target_padding_mask: Optional[torch.Tensor] = None,
+        negative_labels: Optional[torch.Tensor] = None,
+        weights: Optional[torch.Tensor] = None,
+    ) -> torch.Tensor:
+        if negative_labels is None:
+            raise ValueError("CESampled requires negative_labels")
+        pos, neg = self.get_sampled_logits(embeddings, positive_labels, negative_labels)
+        if self.log_correction:
+            n_neg = neg.shape[-1]
+            vocab = self.vocab_size or (int(negative_labels.max()) + 1)
+            rejected = torch.isinf(neg).sum(-1, keepdim=True)
+            correction = torch.log(torch.tensor(float(max(vocab - 1, 1)), device=neg.device)) - torch.log(
+                (n_neg - rejected).clamp(min=1).to(neg.dtype)
+            )
+            neg = neg + correction
+        logits = torch.cat([pos, neg], dim=-1).float()  # [B, L, 1+n]
+        mask = target_padding_mask if target_padding_mask is not None else padding_mask
+        target = torch.zeros(logits.shape[:-1], dtype=torch.long, device=logits.device)
+        per_pos = torch.nn.functional.cross_entropy(
+            logits.reshape(-1, logits.shape[-1]),
+            target.reshape(-1),
+            reduction="none",
+        ).reshape(target.shape)
+        valid = mask.to(per_pos.dtype)
+        if weights is not None:
+            valid = valid * weights
+        return (per_pos * valid).sum() / valid.sum().clamp(min=1e-12)
+
+
+class CESampledWeighted(CESampled):
+    """Weighted variant (reference ce.py:252): weights flow through forward."""
