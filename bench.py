@@ -1,0 +1,178 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: SASRec training throughput (interactions/sec).
+
+BASELINE config: SASRec (2 blocks, d=64, seq_len=50), bf16, ML-20M-shape
+synthetic data (27278 items), random-init weights.  Weak scaling: per-GPU
+batch is fixed; value is the WHOLE-JOB aggregate interactions/sec over all
+ranks.
+
+Single GPU:   python bench.py --gpus 1 --steps 20 --warmup 5
+Multi GPU:    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+                  --master-addr 127.0.0.1 bench.py --gpus N ...
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+N_ITEMS = 27278  # ML-20M catalog size
+SEQ_LEN = 50
+EMB_DIM = 64
+N_BLOCKS = 2
+N_HEADS = 2
+
+
+def build_model(device):
+    from replay_amd.data.nn import TensorFeatureInfo, TensorSchema
+    from replay_amd.data.schema import FeatureHint, FeatureType
+    from replay_amd.nn.loss import CE
+    from replay_amd.nn.sequential.sasrec import SasRec
+
+    schema = TensorSchema(
+        [
+            TensorFeatureInfo(
+                "item_id",
+                FeatureType.CATEGORICAL,
+                is_seq=True,
+                feature_hint=FeatureHint.ITEM_ID,
+                cardinality=N_ITEMS,
+                embedding_dim=EMB_DIM,
+            )
+        ]
+    )
+    model = SasRec.from_params(
+        schema,
+        max_sequence_length=SEQ_LEN,
+        embedding_dim=EMB_DIM,
+        num_blocks=N_BLOCKS,
+        num_heads=N_HEADS,
+        dropout=0.0,
+        loss=CE(),
+    ).to(device)
+    return model
+
+
+def make_batches(n_batches, batch_size, device, seed):
+    """Synthetic ML-20M-shape sequence batches, generated on device."""
+    gen = torch.Generator(device="cpu").manual_seed(seed)
+    batches = []
+    for _ in range(n_batches):
+        items = torch.randint(0, N_ITEMS, (batch_size, SEQ_LEN + 1), generator=gen)
+        batch = {
+            "item_id": items[:, :-1].to(device),
+            "labels": items[:, 1:].to(device),
+            "padding_mask": torch.ones(batch_size, SEQ_LEN, dtype=torch.bool, device=device),
+        }
+        batch["labels_padding_mask"] = batch["padding_mask"]
+        batches.append(batch)
+    return batches
+
+
+def main() -> None:
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--gpus", type=int, default=1)
+    parser.add_argument("--steps", type=int, default=20)
+    parser.add_argument("--warmup", type=int, default=5)
+    parser.add_argument("--batch", type=int, default=1024, help="per-GPU batch size")
+    parser.add_argument("--lr", type=float, default=1e-3)
+    args = parser.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    use_cuda = torch.cuda.is_available()
+    if world > 1:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        torch.distributed.init_process_group(backend="nccl" if use_cuda else "gloo")
+    device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
+    if use_cuda:
+        torch.cuda.set_device(device)
+    torch.manual_seed(1234 + rank)
+
+    model = build_model(device)
+    if world > 1:
+        model = torch.nn.parallel.DistributedDataParallel(
+            model,
+            device_ids=[device.index] if use_cuda else None,
+            bucket_cap_mb=64,
+            gradient_as_bucket_view=True,
+        )
+    optimizer = torch.optim.Adam(model.parameters(), lr=args.lr)
+    batches = make_batches(4, args.batch, device, seed=1000 + rank)
+
+    amp_dtype = torch.bfloat16
+    autocast = torch.autocast(device_type=device.type, dtype=amp_dtype, enabled=use_cuda)
+
+    def step(i: int) -> None:
+        batch = batches[i % len(batches)]
+        with autocast:
+            loss = model(batch)
+        optimizer.zero_grad(set_to_none=True)
+        loss.backward()
+        optimizer.step()
+
+    for i in range(args.warmup):
+        step(i)
+
+    if world > 1:
+        torch.distributed.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(i)
+    if use_cuda:
+        torch.cuda.synchronize()
+    if world > 1:
+        torch.distributed.barrier()
+    elapsed = time.perf_counter() - t0
+
+    if world > 1:
+        t = torch.tensor([elapsed], device=device if use_cuda else "cpu", dtype=torch.float64)
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    n_gpus = world if world > 1 else args.gpus
+    global_batch = args.batch * n_gpus
+    interactions = args.steps * global_batch * SEQ_LEN
+    value = interactions / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        print(
+            json.dumps(
+                {
+                    "metric": "training interactions/sec, SASRec ML-20M-shape",
+                    "value": value,
+                    "unit": "interactions/sec",
+                    "n_gpus": n_gpus,
+                    "steps": args.steps,
+                    "warmup": args.warmup,
+                    "ms_per_step": ms_per_step,
+                    "higher_is_better": True,
+                    "scaling": "weak",
+                    "vs_baseline": None,
+                    "dtype": "bf16" if use_cuda else "fp32",
+                    "data": "synthetic",
+                    "config": {
+                        "model": f"sasrec_{N_BLOCKS}blocks_d{EMB_DIM}",
+                        "global_batch": global_batch,
+                        "seq_len": SEQ_LEN,
+                        "n_items": N_ITEMS,
+                        "loss": "full-softmax CE",
+                        "parallelism": f"dp{n_gpus}",
+                    },
+                }
+            )
+        )
+    if world > 1:
+        torch.distributed.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

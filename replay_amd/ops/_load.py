@@ -37,7 +37,10 @@ def build_extension(verbose: bool = True):
     from torch.utils.cpp_extension import load
 
     src_dir = _HERE / "hip"
-    sources = sorted(str(p) for p in src_dir.glob("*.hip")) + sorted(str(p) for p in src_dir.glob("*.cpp"))
+    # exclude torch-hipify's generated *_hip.hip copies of our sources
+    sources = sorted(
+        str(p) for p in src_dir.glob("*.hip") if not p.name.endswith("_hip.hip")
+    ) + sorted(str(p) for p in src_dir.glob("*.cpp"))
     if not sources:
         raise RuntimeError(f"No HIP sources under {src_dir}")
     build_dir = _HERE / "build"

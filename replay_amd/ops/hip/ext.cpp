@@ -1,0 +1,14 @@
+// Python bindings for the replay_amd gfx950 HIP extension.
+#include <torch/extension.h>
+#include <vector>
+
+std::vector<torch::Tensor> layer_norm_fwd(torch::Tensor x, torch::Tensor weight,
+                                          torch::Tensor bias, double eps);
+std::vector<torch::Tensor> layer_norm_bwd(torch::Tensor x, torch::Tensor dy,
+                                          torch::Tensor weight, torch::Tensor mean,
+                                          torch::Tensor rstd);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("layer_norm_fwd", &layer_norm_fwd, "fused LayerNorm forward (gfx950)");
+  m.def("layer_norm_bwd", &layer_norm_bwd, "fused LayerNorm backward (gfx950)");
+}
