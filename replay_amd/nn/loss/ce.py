@@ -33,8 +33,16 @@ class CE(LossBase):
         logits = self.logits_callback(embeddings)  # [B, L, V]
         mask = target_padding_mask if target_padding_mask is not None else padding_mask
         labels = positive_labels.masked_fill(~mask, -100)
+        logits2d = logits.reshape(-1, logits.shape[-1])
+        if logits.is_cuda:
+            from replay_amd.ops import hip_ext, require_hip_on_gpu
+
+            if require_hip_on_gpu(logits) and hasattr(hip_ext(), "ce_fwd"):
+                from replay_amd.ops.autograd import fused_cross_entropy
+
+                return fused_cross_entropy(logits2d.contiguous(), labels.reshape(-1), -100)
         return torch.nn.functional.cross_entropy(
-            logits.reshape(-1, logits.shape[-1]).float(),
+            logits2d.float(),
             labels.reshape(-1),
             ignore_index=-100,
         )

@@ -7,8 +7,15 @@ std::vector<torch::Tensor> layer_norm_fwd(torch::Tensor x, torch::Tensor weight,
 std::vector<torch::Tensor> layer_norm_bwd(torch::Tensor x, torch::Tensor dy,
                                           torch::Tensor weight, torch::Tensor mean,
                                           torch::Tensor rstd);
+std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor labels,
+                                  int64_t ignore_index);
+torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor labels, torch::Tensor lse,
+                     torch::Tensor grad_scale, torch::Tensor count, int64_t ignore_index,
+                     bool inplace);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layer_norm_fwd", &layer_norm_fwd, "fused LayerNorm forward (gfx950)");
   m.def("layer_norm_bwd", &layer_norm_bwd, "fused LayerNorm backward (gfx950)");
+  m.def("ce_fwd", &ce_fwd, "fused softmax-CE forward (gfx950)");
+  m.def("ce_bwd", &ce_bwd, "fused softmax-CE backward (gfx950)");
 }

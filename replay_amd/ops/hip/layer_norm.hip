@@ -52,10 +52,13 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x,
 }
 
 // dx for one row needs two row-reductions of dy*w and dy*w*xhat.
-// dweight/dbias are column reductions over all rows: accumulate per-wave
-// partials with device-scope atomics into fp32 buffers (guide Guideline 12:
-// per-wave partials first; here each lane owns distinct columns so a plain
-// atomicAdd per (lane, col) stripe is already conflict-light).
+// dweight/dbias are column reductions over all rows.  A per-element
+// atomicAdd is catastrophic here (measured 1.25 ms per call at [51200,64]
+// from contention on 64 fp32 words): instead each wave accumulates its
+// grid-stride rows' partials in registers (E <= 8*WAVE columns per lane) and
+// issues ONE atomicAdd per column at the end (guide Guideline 12).
+#define LN_MAX_COLS_PER_LANE 8  // supports E up to 512
+
 template <typename T>
 __global__ void ln_bwd_kernel(const T* __restrict__ x,
                               const T* __restrict__ dy,
@@ -69,6 +72,11 @@ __global__ void ln_bwd_kernel(const T* __restrict__ x,
   const int wave_id = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int n_waves = (gridDim.x * blockDim.x) / WAVE;
+
+  float acc_dw[LN_MAX_COLS_PER_LANE];
+  float acc_db[LN_MAX_COLS_PER_LANE];
+#pragma unroll
+  for (int i = 0; i < LN_MAX_COLS_PER_LANE; ++i) acc_dw[i] = acc_db[i] = 0.f;
 
   for (int64_t row = wave_id; row < n_rows; row += n_waves) {
     const T* xr = x + row * n_cols;
@@ -85,14 +93,20 @@ __global__ void ln_bwd_kernel(const T* __restrict__ x,
     s1 = wave_reduce_sum(s1) / n_cols;
     s2 = wave_reduce_sum(s2) / n_cols;
     T* dxr = dx + row * n_cols;
-    for (int c = lane; c < n_cols; c += WAVE) {
+    int i = 0;
+    for (int c = lane; c < n_cols; c += WAVE, ++i) {
       float xhat = (to_f32<T>(xr[c]) - m) * rs;
       float dyv = to_f32<T>(dyr[c]);
       float dyw = dyv * weight[c];
       dxr[c] = from_f32<T>((dyw - s1 - xhat * s2) * rs);
-      atomicAdd(&dweight[c], dyv * xhat);
-      atomicAdd(&dbias[c], dyv);
+      acc_dw[i] += dyv * xhat;
+      acc_db[i] += dyv;
     }
+  }
+  int i = 0;
+  for (int c = lane; c < n_cols; c += WAVE, ++i) {
+    atomicAdd(&dweight[c], acc_dw[i]);
+    atomicAdd(&dbias[c], acc_db[i]);
   }
 }
 
@@ -119,9 +133,12 @@ void ln_bwd_launch(const torch::Tensor& x, const torch::Tensor& dy, const torch:
                    torch::Tensor& dw, torch::Tensor& db) {
   const int64_t n_rows = x.numel() / x.size(-1);
   const int n_cols = x.size(-1);
+  TORCH_CHECK(n_cols <= WAVE * LN_MAX_COLS_PER_LANE, "LN bwd supports E<=512");
   const int threads = 256;
   const int waves_per_block = threads / WAVE;
-  int blocks = (int)std::min<int64_t>((n_rows + waves_per_block - 1) / waves_per_block, 8192);
+  // cap waves: each wave does one atomicAdd per column at the end, so more
+  // waves = more atomic traffic; 512 blocks = 2048 waves fills 256 CUs twice
+  int blocks = (int)std::min<int64_t>((n_rows + waves_per_block - 1) / waves_per_block, 512);
   auto stream = at::cuda::getCurrentHIPStream();
   hipLaunchKernelGGL(ln_bwd_kernel<T>, dim3(blocks), dim3(threads), 0, stream,
                      reinterpret_cast<const T*>(x.data_ptr()),
