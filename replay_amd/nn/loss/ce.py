@@ -30,19 +30,27 @@ class CE(LossBase):
         negative_labels: Optional[torch.Tensor] = None,
         weights: Optional[torch.Tensor] = None,
     ) -> torch.Tensor:
-        logits = self.logits_callback(embeddings)  # [B, L, V]
         mask = target_padding_mask if target_padding_mask is not None else padding_mask
         labels = positive_labels.masked_fill(~mask, -100)
-        logits2d = logits.reshape(-1, logits.shape[-1])
-        if logits.is_cuda:
+        if embeddings.is_cuda:
             from replay_amd.ops import hip_ext, require_hip_on_gpu
 
-            if require_hip_on_gpu(logits) and hasattr(hip_ext(), "ce_fwd"):
-                from replay_amd.ops.autograd import fused_cross_entropy
+            head = self.logits_callback
+            if (
+                require_hip_on_gpu(embeddings)
+                and hasattr(hip_ext(), "ce_fwd")
+                and hasattr(head, "get_item_weights")
+            ):
+                # chunked fused CE (K10): logits chunks stay L3-resident,
+                # the [N, V] logits never materialize in HBM
+                from replay_amd.ops.fused_ce import chunked_fused_ce
 
-                return fused_cross_entropy(logits2d.contiguous(), labels.reshape(-1), -100)
+                hidden2d = embeddings.reshape(-1, embeddings.shape[-1])
+                weight = head.get_item_weights()
+                return chunked_fused_ce(hidden2d, weight, labels.reshape(-1), -100)
+        logits = self.logits_callback(embeddings)  # [B, L, V]
         return torch.nn.functional.cross_entropy(
-            logits2d.float(),
+            logits.reshape(-1, logits.shape[-1]).float(),
             labels.reshape(-1),
             ignore_index=-100,
         )

@@ -1,14 +1,10 @@
 // Fused softmax cross-entropy over large catalogs for gfx950 (CDNA4).
 //
-// K10 in SURVEY §2.12.  The eager reference path materializes fp32
-// log-softmax of [B*L, V] (measured 6.5 ms/step at V=27278 plus ~3 ms of
-// bf16<->fp32 copies); here the loss reads the bf16 logits once
-// (fused max+logsumexp, one wave per row, vectorized) and the backward
-// writes bf16 dlogits in a second single pass: 3 passes over the logits
-// instead of ~8, no fp32 copy ever.
-//
-// Layout: logits [N, V] bf16/fp32 (row-contiguous), labels [N] int64 with
-// ignore_index for padded positions.  Loss = mean over valid rows.
+// K10 in SURVEY §2.12.  One 64-lane wave per row of [N, V] logits:
+// forward = ONE vectorized pass (online max + rescaled sum-exp, 8 bf16 =
+// 16 B per lane per iteration, guide G13), backward = one vectorized pass
+// writing bf16 dlogits in place.  Used standalone and as the per-chunk core
+// of the chunked CE (ops/fused_ce.py) whose chunks stay L3-resident.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -16,13 +12,47 @@
 
 namespace {
 
-// one 64-lane wave per row; 256-thread workgroups = 4 rows per block
-template <typename T>
+// load 8 bf16 (16 B) as uint4
+__device__ __forceinline__ void load8_bf16(const __hip_bfloat16* p, float* out) {
+  const uint4 raw = *reinterpret_cast<const uint4*>(p);
+  const unsigned w[4] = {raw.x, raw.y, raw.z, raw.w};
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    __hip_bfloat162 pair = *reinterpret_cast<const __hip_bfloat162*>(&w[i]);
+    out[2 * i] = __bfloat162float(pair.x);
+    out[2 * i + 1] = __bfloat162float(pair.y);
+  }
+}
+
+__device__ __forceinline__ void store8_bf16(__hip_bfloat16* p, const float* in) {
+  uint4 raw;
+  unsigned* w = reinterpret_cast<unsigned*>(&raw);
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    __hip_bfloat162 pair = __float22bfloat162_rn(float2{in[2 * i], in[2 * i + 1]});
+    w[i] = *reinterpret_cast<const unsigned*>(&pair);
+  }
+  *reinterpret_cast<uint4*>(p) = raw;
+}
+
+// online (max, sum) combine across lanes
+__device__ __forceinline__ void wave_reduce_online(float& m, float& s) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    float m2 = __shfl_xor(m, off, WAVE);
+    float s2 = __shfl_xor(s, off, WAVE);
+    float mn = fmaxf(m, m2);
+    s = s * __expf(m - mn) + s2 * __expf(m2 - mn);
+    m = mn;
+  }
+}
+
+template <typename T, bool VEC8>
 __global__ void ce_fwd_kernel(const T* __restrict__ logits,
                               const int64_t* __restrict__ labels,
                               float* __restrict__ lse_out,
-                              float* __restrict__ loss_out,  // scalar, atomic
-                              int* __restrict__ count_out,   // scalar, atomic
+                              float* __restrict__ loss_out,
+                              int* __restrict__ count_out,
                               int64_t n_rows, int64_t n_cols, int64_t ignore_index) {
   const int wave_id = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
@@ -31,59 +61,68 @@ __global__ void ce_fwd_kernel(const T* __restrict__ logits,
   float local_loss = 0.f;
   int local_count = 0;
   for (int64_t row = wave_id; row < n_rows; row += n_waves) {
-    const int64_t label = labels[row];
     const T* lr = logits + row * n_cols;
-    // pass 1: row max (vectorized 4-wide when aligned)
-    float m = -INFINITY;
-    int64_t c = lane * 4;
-    if ((n_cols & 3) == 0) {
-      for (; c + 3 < n_cols; c += WAVE * 4) {
-        float4 v4;
-        const T* p = lr + c;
-        v4.x = to_f32<T>(p[0]); v4.y = to_f32<T>(p[1]);
-        v4.z = to_f32<T>(p[2]); v4.w = to_f32<T>(p[3]);
-        m = fmaxf(m, fmaxf(fmaxf(v4.x, v4.y), fmaxf(v4.z, v4.w)));
+    float m = -INFINITY, s = 0.f;
+    if constexpr (VEC8) {
+      // per-row 16B alignment: scalar head, vec8 body, scalar tail
+      const int head = (int)(((16 - (reinterpret_cast<uintptr_t>(lr) & 15)) & 15) >> 1);
+      const int64_t n_vec = head + ((n_cols - head) & ~int64_t(7));
+      float v[8];
+      for (int64_t c = lane; c < head; c += WAVE) {
+        float x = to_f32<T>(lr[c]);
+        if (x > m) { s *= __expf(m - x); m = x; }
+        s += __expf(x - m);
+      }
+      for (int64_t c = head + lane * 8; c + 7 < n_vec; c += WAVE * 8) {
+        load8_bf16(reinterpret_cast<const __hip_bfloat16*>(lr) + c, v);
+        float m8 = v[0];
+#pragma unroll
+        for (int i = 1; i < 8; ++i) m8 = fmaxf(m8, v[i]);
+        if (m8 > m) {
+          s *= __expf(m - m8);
+          m = m8;
+        }
+#pragma unroll
+        for (int i = 0; i < 8; ++i) s += __expf(v[i] - m);
+      }
+      for (int64_t c = n_vec + lane; c < n_cols; c += WAVE) {
+        float x = to_f32<T>(lr[c]);
+        if (x > m) { s *= __expf(m - x); m = x; }
+        s += __expf(x - m);
       }
     } else {
-      for (int64_t cc = lane; cc < n_cols; cc += WAVE) m = fmaxf(m, to_f32<T>(lr[cc]));
-    }
-    m = wave_reduce_max(m);
-    // pass 2: sum exp
-    float s = 0.f;
-    if ((n_cols & 3) == 0) {
-      for (c = lane * 4; c + 3 < n_cols; c += WAVE * 4) {
-        const T* p = lr + c;
-        s += __expf(to_f32<T>(p[0]) - m) + __expf(to_f32<T>(p[1]) - m) +
-             __expf(to_f32<T>(p[2]) - m) + __expf(to_f32<T>(p[3]) - m);
+      for (int64_t c = lane; c < n_cols; c += WAVE) {
+        float v = to_f32<T>(lr[c]);
+        if (v > m) {
+          s *= __expf(m - v);
+          m = v;
+        }
+        s += __expf(v - m);
       }
-    } else {
-      for (int64_t cc = lane; cc < n_cols; cc += WAVE) s += __expf(to_f32<T>(lr[cc]) - m);
     }
-    s = wave_reduce_sum(s);
+    wave_reduce_online(m, s);
     const float lse = m + __logf(s);
     if (lane == 0) {
       lse_out[row] = lse;
+      const int64_t label = labels[row];
       if (label != ignore_index) {
         local_loss += lse - to_f32<T>(lr[label]);
         local_count += 1;
       }
     }
   }
-  if (lane == 0) {
-    if (local_count > 0) {
-      atomicAdd(loss_out, local_loss);
-      atomicAdd(count_out, local_count);
-    }
+  if (lane == 0 && local_count > 0) {
+    atomicAdd(loss_out, local_loss);
+    atomicAdd(count_out, local_count);
   }
 }
 
-// dlogits[r, c] = (exp(logit - lse) - 1{c==label}) * scale  (scale = dloss/N)
-template <typename T>
+template <typename T, bool VEC8>
 __global__ void ce_bwd_kernel(const T* __restrict__ logits,
                               const int64_t* __restrict__ labels,
                               const float* __restrict__ lse,
                               T* __restrict__ dlogits,
-                              const float* __restrict__ grad_scale,  // dLoss (scalar tensor)
+                              const float* __restrict__ grad_scale,
                               const int* __restrict__ count,
                               int64_t n_rows, int64_t n_cols, int64_t ignore_index) {
   const int wave_id = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
@@ -95,15 +134,49 @@ __global__ void ce_bwd_kernel(const T* __restrict__ logits,
     const int64_t label = labels[row];
     const T* lr = logits + row * n_cols;
     T* dr = dlogits + row * n_cols;
-    if (label == ignore_index) {
-      for (int64_t c = lane; c < n_cols; c += WAVE) dr[c] = from_f32<T>(0.f);
-      continue;
-    }
-    const float l = lse[row];
-    for (int64_t c = lane; c < n_cols; c += WAVE) {
-      float p = __expf(to_f32<T>(lr[c]) - l);
-      float g = (p - (c == label ? 1.f : 0.f)) * scale;
-      dr[c] = from_f32<T>(g);
+    const bool ignored = (label == ignore_index);
+    const float l = ignored ? 0.f : lse[row];
+    if constexpr (VEC8) {
+      const int head = (int)(((16 - (reinterpret_cast<uintptr_t>(lr) & 15)) & 15) >> 1);
+      const int64_t n_vec = head + ((n_cols - head) & ~int64_t(7));
+      float v[8];
+      for (int64_t c = lane; c < head; c += WAVE) {
+        if (ignored) { dr[c] = from_f32<T>(0.f); }
+        else {
+          float p = __expf(to_f32<T>(lr[c]) - l);
+          dr[c] = from_f32<T>((p - (c == label ? 1.f : 0.f)) * scale);
+        }
+      }
+      for (int64_t c = head + lane * 8; c + 7 < n_vec; c += WAVE * 8) {
+        if (ignored) {
+#pragma unroll
+          for (int i = 0; i < 8; ++i) v[i] = 0.f;
+        } else {
+          load8_bf16(reinterpret_cast<const __hip_bfloat16*>(lr) + c, v);
+#pragma unroll
+          for (int i = 0; i < 8; ++i) {
+            float p = __expf(v[i] - l);
+            v[i] = (p - ((c + i) == label ? 1.f : 0.f)) * scale;
+          }
+        }
+        store8_bf16(reinterpret_cast<__hip_bfloat16*>(dr) + c, v);
+      }
+      for (int64_t c = n_vec + lane; c < n_cols; c += WAVE) {
+        if (ignored) { dr[c] = from_f32<T>(0.f); }
+        else {
+          float p = __expf(to_f32<T>(lr[c]) - l);
+          dr[c] = from_f32<T>((p - (c == label ? 1.f : 0.f)) * scale);
+        }
+      }
+    } else {
+      for (int64_t c = lane; c < n_cols; c += WAVE) {
+        if (ignored) {
+          dr[c] = from_f32<T>(0.f);
+        } else {
+          float p = __expf(to_f32<T>(lr[c]) - l);
+          dr[c] = from_f32<T>((p - (c == label ? 1.f : 0.f)) * scale);
+        }
+      }
     }
   }
 }
@@ -123,18 +196,21 @@ std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor labels,
   int blocks = (int)std::min<int64_t>((n_rows + 3) / 4, 4096);
   auto stream = at::cuda::getCurrentHIPStream();
   auto lab = labels.contiguous();
-#define LAUNCH_CE_FWD(T)                                                              \
-  hipLaunchKernelGGL(ce_fwd_kernel<T>, dim3(blocks), dim3(threads), 0, stream,        \
+  // row alignment handled in-kernel (scalar head); vec8 for any bf16 shape
+  const bool vec8 = (logits.scalar_type() == torch::kBFloat16);
+#define LAUNCH_CE_FWD(T, V)                                                           \
+  hipLaunchKernelGGL((ce_fwd_kernel<T, V>), dim3(blocks), dim3(threads), 0, stream,   \
                      reinterpret_cast<const T*>(logits.data_ptr()),                   \
                      lab.data_ptr<int64_t>(), lse.data_ptr<float>(),                  \
                      loss.data_ptr<float>(), count.data_ptr<int>(), n_rows, n_cols,   \
                      ignore_index)
   if (logits.scalar_type() == torch::kBFloat16) {
-    LAUNCH_CE_FWD(__hip_bfloat16);
+    if (vec8) LAUNCH_CE_FWD(__hip_bfloat16, true);
+    else LAUNCH_CE_FWD(__hip_bfloat16, false);
   } else if (logits.scalar_type() == torch::kFloat32) {
-    LAUNCH_CE_FWD(float);
+    LAUNCH_CE_FWD(float, false);
   } else if (logits.scalar_type() == torch::kHalf) {
-    LAUNCH_CE_FWD(__half);
+    LAUNCH_CE_FWD(__half, false);
   } else {
     TORCH_CHECK(false, "unsupported dtype");
   }
@@ -152,18 +228,20 @@ torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor labels, torch::Tensor l
   auto stream = at::cuda::getCurrentHIPStream();
   auto lab = labels.contiguous();
   auto gs = grad_scale.to(torch::kFloat32).contiguous();
-#define LAUNCH_CE_BWD(T)                                                              \
-  hipLaunchKernelGGL(ce_bwd_kernel<T>, dim3(blocks), dim3(threads), 0, stream,        \
+  const bool vec8 = (logits.scalar_type() == torch::kBFloat16);
+#define LAUNCH_CE_BWD(T, V)                                                           \
+  hipLaunchKernelGGL((ce_bwd_kernel<T, V>), dim3(blocks), dim3(threads), 0, stream,   \
                      reinterpret_cast<const T*>(logits.data_ptr()),                   \
                      lab.data_ptr<int64_t>(), lse.data_ptr<float>(),                  \
                      reinterpret_cast<T*>(dlogits.data_ptr()), gs.data_ptr<float>(),  \
                      count.data_ptr<int>(), n_rows, n_cols, ignore_index)
   if (logits.scalar_type() == torch::kBFloat16) {
-    LAUNCH_CE_BWD(__hip_bfloat16);
+    if (vec8) LAUNCH_CE_BWD(__hip_bfloat16, true);
+    else LAUNCH_CE_BWD(__hip_bfloat16, false);
   } else if (logits.scalar_type() == torch::kFloat32) {
-    LAUNCH_CE_BWD(float);
+    LAUNCH_CE_BWD(float, false);
   } else if (logits.scalar_type() == torch::kHalf) {
-    LAUNCH_CE_BWD(__half);
+    LAUNCH_CE_BWD(__half, false);
   } else {
     TORCH_CHECK(false, "unsupported dtype");
   }

@@ -98,6 +98,35 @@ class TestFusedCrossEntropy:
         loss = fused_cross_entropy(logits, labels)
         assert float(loss) == 0.0
 
+    @pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
+    @pytest.mark.parametrize("shape", [(300, 64, 1000), (512, 32, 27278)])
+    def test_chunked_fused_ce_matches_reference(self, dtype, shape):
+        from replay_amd.ops.fused_ce import chunked_fused_ce
+
+        torch.manual_seed(3)
+        N, E, V = shape
+        hidden = torch.randn(N, E, device="cuda", dtype=dtype, requires_grad=True)
+        weight = torch.randn(V, E, device="cuda", dtype=torch.float32, requires_grad=True)
+        labels = torch.randint(0, V, (N,), device="cuda")
+        labels[::7] = -100
+
+        loss = chunked_fused_ce(hidden, weight, labels, chunk_rows=128)
+        loss.backward()
+
+        h_ref = hidden.detach().float().clone().requires_grad_(True)
+        w_ref = weight.detach().clone().requires_grad_(True)
+        ref = torch.nn.functional.cross_entropy(h_ref @ w_ref.t(), labels, ignore_index=-100)
+        ref.backward()
+
+        tol = 5e-2 if dtype == torch.bfloat16 else 2e-4
+        assert abs(float(loss) - float(ref)) < tol * max(1.0, abs(float(ref)))
+        torch.testing.assert_close(
+            hidden.grad.float(), h_ref.grad, atol=tol * 0.1, rtol=tol
+        )
+        torch.testing.assert_close(
+            weight.grad.float(), w_ref.grad, atol=tol * 0.1, rtol=tol
+        )
+
 
 @requires_gpu
 class TestModelOnGPU:
