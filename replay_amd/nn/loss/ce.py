@@ -46,6 +46,10 @@ class CE(LossBase):
                 from replay_amd.ops.fused_ce import chunked_fused_ce
 
                 hidden2d = embeddings.reshape(-1, embeddings.shape[-1])
+                if torch.is_autocast_enabled("cuda"):
+                    # custom autograd functions bypass autocast: cast here so
+                    # the chunk GEMMs and CE kernels run in bf16 (MFMA path)
+                    hidden2d = hidden2d.to(torch.get_autocast_dtype("cuda"))
                 weight = head.get_item_weights()
                 return chunked_fused_ce(hidden2d, weight, labels.reshape(-1), -100)
         logits = self.logits_callback(embeddings)  # [B, L, V]

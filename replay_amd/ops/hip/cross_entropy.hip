@@ -42,7 +42,10 @@ __device__ __forceinline__ void wave_reduce_online(float& m, float& s) {
     float m2 = __shfl_xor(m, off, WAVE);
     float s2 = __shfl_xor(s, off, WAVE);
     float mn = fmaxf(m, m2);
-    s = s * __expf(m - mn) + s2 * __expf(m2 - mn);
+    // lanes that saw no elements carry (m=-inf, s=0): guard 0*exp(-inf-(-inf))
+    float a = (s == 0.f) ? 0.f : s * __expf(m - mn);
+    float b = (s2 == 0.f) ? 0.f : s2 * __expf(m2 - mn);
+    s = a + b;
     m = mn;
   }
 }
