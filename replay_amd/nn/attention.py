@@ -49,18 +49,56 @@ class MultiheadAttention(torch.nn.Module):
         self.in_proj = torch.nn.Linear(embed_dim, 3 * embed_dim, bias=bias)
         self.out_proj = torch.nn.Linear(embed_dim, embed_dim, bias=bias)
 
+    def _can_use_flash(self, x: torch.Tensor, attn_mask) -> bool:
+        from replay_amd.nn.mask import MaskSpec
+        from replay_amd.ops import hip_ext
+
+        if not isinstance(attn_mask, MaskSpec) or not x.is_cuda:
+            return False
+        if self.dropout > 0.0 and self.training:
+            return False
+        if x.dtype not in (torch.bfloat16, torch.float16, torch.float32):
+            return False
+        ext = hip_ext()
+        if ext is None or not hasattr(ext, "attention_fwd"):
+            return False
+        L, Dh = x.shape[1], self.head_dim
+        if Dh > 64 or 256 % Dh != 0 or L > 512:
+            return False
+        # backward LDS budget check (fp32 staging): D*Lpad + 5*L*D + extras
+        lpad = (L + 63) & ~63
+        lds_bwd = 4 * (Dh * lpad + 5 * L * Dh + L + 8 * L) + L
+        return lds_bwd <= 160 * 1024
+
     def forward(
         self,
         x: torch.Tensor,
-        attn_mask: Optional[torch.Tensor] = None,
+        attn_mask=None,
         key_padding_mask: Optional[torch.Tensor] = None,
     ) -> torch.Tensor:
-        """x: [B, L, E]; attn_mask: [B*H, L, L] additive float;
+        """x: [B, L, E]; attn_mask: [B*H, L, L] additive float or a MaskSpec;
         key_padding_mask: [B, L] bool, True = PAD (torch convention)."""
+        from replay_amd.nn.mask import MaskSpec
+
         B, L, E = x.shape
         H, Dh = self.num_heads, self.head_dim
         qkv = self.in_proj(x)  # [B, L, 3E]
         q, k, v = qkv.chunk(3, dim=-1)
+
+        if key_padding_mask is None and self._can_use_flash(x, attn_mask):
+            from replay_amd.ops.autograd import FlashAttentionFunction
+
+            def split4(t):
+                return t.view(B, L, H, Dh).transpose(1, 2).contiguous()
+
+            out = FlashAttentionFunction.apply(
+                split4(q), split4(k), split4(v), attn_mask.padding_mask, attn_mask.causal
+            )
+            out = out.transpose(1, 2).reshape(B, L, E)
+            return self.out_proj(out)
+
+        if isinstance(attn_mask, MaskSpec):
+            attn_mask = attn_mask.materialize()
 
         def split(t):
             return t.view(B, L, H, Dh).transpose(1, 2).reshape(B * H, L, Dh)
@@ -124,6 +162,10 @@ class MultiHeadDifferentialAttention(torch.nn.Module):
         attn_mask: Optional[torch.Tensor] = None,
         key_padding_mask: Optional[torch.Tensor] = None,
     ) -> torch.Tensor:
+        from replay_amd.nn.mask import MaskSpec
+
+        if isinstance(attn_mask, MaskSpec):
+            attn_mask = attn_mask.materialize()
         B, L, E = x.shape
         H, Dh = self.num_heads, self.head_dim
         q = self.q_proj(x).view(B, L, 2 * H, Dh).transpose(1, 2)  # [B, 2H, L, Dh]

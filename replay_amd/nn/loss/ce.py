@@ -41,17 +41,26 @@ class CE(LossBase):
                 and hasattr(hip_ext(), "ce_fwd")
                 and hasattr(head, "get_item_weights")
             ):
-                # chunked fused CE (K10): logits chunks stay L3-resident,
-                # the [N, V] logits never materialize in HBM
-                from replay_amd.ops.fused_ce import chunked_fused_ce
-
                 hidden2d = embeddings.reshape(-1, embeddings.shape[-1])
                 if torch.is_autocast_enabled("cuda"):
                     # custom autograd functions bypass autocast: cast here so
-                    # the chunk GEMMs and CE kernels run in bf16 (MFMA path)
+                    # the GEMMs and CE kernels run in bf16 (MFMA path)
                     hidden2d = hidden2d.to(torch.get_autocast_dtype("cuda"))
                 weight = head.get_item_weights()
-                return chunked_fused_ce(hidden2d, weight, labels.reshape(-1), -100)
+                n_elems = hidden2d.shape[0] * weight.shape[0]
+                if n_elems * hidden2d.element_size() > 8 * 2**30:
+                    # huge catalogs (e.g. 10M items): chunked CE, logits are
+                    # recomputed in backward and never fully materialized
+                    from replay_amd.ops.fused_ce import chunked_fused_ce
+
+                    return chunked_fused_ce(hidden2d, weight, labels.reshape(-1), -100)
+                # materialized path: ONE bf16 logits buffer, fused one-pass
+                # LSE forward + in-place dlogits backward (measured faster
+                # than chunking at V<=1e5: no recompute, full-width GEMMs)
+                from replay_amd.ops.autograd import fused_cross_entropy
+
+                logits2d = hidden2d @ weight.to(hidden2d.dtype).t()
+                return fused_cross_entropy(logits2d, labels.reshape(-1), -100)
         logits = self.logits_callback(embeddings)  # [B, L, V]
         return torch.nn.functional.cross_entropy(
             logits.reshape(-1, logits.shape[-1]).float(),

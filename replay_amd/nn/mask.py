@@ -4,6 +4,11 @@ Parity with reference replay/nn/mask.py (DefaultAttentionMask:58, float mask
 build :30-51): causal tril AND key-padding, OR'd with the diagonal so a fully
 masked row still attends to itself; fill value -inf in train, finfo.min in
 eval (reference :39-42).
+
+MI355X design note: the float [B*H, L, L] mask is built LAZILY via
+:class:`MaskSpec` — on the GPU path the fused attention kernel (K1/K2)
+consumes the [B, L] bool padding mask + causal flag directly and the float
+tensor never materializes.
 """
 
 from __future__ import annotations
@@ -11,18 +16,20 @@ from __future__ import annotations
 import torch
 
 
-class DefaultAttentionMask(torch.nn.Module):
-    def __init__(self, num_heads: int = 1, causal: bool = True) -> None:
-        super().__init__()
+class MaskSpec:
+    """Lazy attention-mask: holds what the fused kernel needs; materializes
+    the additive float mask only for eager consumers."""
+
+    def __init__(self, padding_mask: torch.Tensor, num_heads: int, causal: bool, training: bool) -> None:
+        self.padding_mask = padding_mask  # [B, L] bool, True = valid
         self.num_heads = num_heads
         self.causal = causal
+        self.training = training
 
-    def forward(self, padding_mask: torch.Tensor) -> torch.Tensor:
-        """padding_mask: [B, L] bool (True = valid).  Returns float mask
-        [B*H, L, L] additive."""
-        B, L = padding_mask.shape
-        device = padding_mask.device
-        allowed = padding_mask[:, None, :].expand(B, L, L)  # key validity
+    def materialize(self) -> torch.Tensor:
+        B, L = self.padding_mask.shape
+        device = self.padding_mask.device
+        allowed = self.padding_mask[:, None, :].expand(B, L, L)
         if self.causal:
             causal = torch.tril(torch.ones(L, L, dtype=torch.bool, device=device))
             allowed = allowed & causal[None]
@@ -34,3 +41,15 @@ class DefaultAttentionMask(torch.nn.Module):
         if self.num_heads > 1:
             mask = mask.repeat_interleave(self.num_heads, dim=0)
         return mask
+
+
+class DefaultAttentionMask(torch.nn.Module):
+    def __init__(self, num_heads: int = 1, causal: bool = True) -> None:
+        super().__init__()
+        self.num_heads = num_heads
+        self.causal = causal
+
+    def forward(self, padding_mask: torch.Tensor) -> MaskSpec:
+        """padding_mask: [B, L] bool (True = valid).  Returns a MaskSpec;
+        call ``.materialize()`` for the additive float [B*H, L, L] mask."""
+        return MaskSpec(padding_mask, self.num_heads, self.causal, self.training)

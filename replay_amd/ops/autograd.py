@@ -52,14 +52,25 @@ def fused_cross_entropy(logits2d: torch.Tensor, labels: torch.Tensor, ignore_ind
 
 
 class FlashAttentionFunction(torch.autograd.Function):
-    """Placeholder until the HIP flash kernel lands; the dispatch in
-    ops/attention.py only routes here when the extension exports
-    attention_fwd."""
+    """Fused attention (K1/K2): q,k,v [B,H,L,Dh]; padding_mask [B,L] bool
+    (True = valid); causal flag.  Backward recomputes P from the saved LSE."""
 
     @staticmethod
-    def forward(ctx, q, k, v, attn_mask):  # pragma: no cover
-        raise NotImplementedError
+    def forward(ctx, q, k, v, padding_mask, causal):
+        ext = hip_ext()
+        scale = 1.0 / (q.shape[-1] ** 0.5)
+        need_lse = q.requires_grad or k.requires_grad or v.requires_grad
+        out, lse = ext.attention_fwd(q, k, v, padding_mask, scale, causal, need_lse)
+        ctx.save_for_backward(q, k, v, out, lse, padding_mask)
+        ctx.causal = causal
+        ctx.scale = scale
+        return out
 
     @staticmethod
-    def backward(ctx, do):  # pragma: no cover
-        raise NotImplementedError
+    def backward(ctx, dout):
+        ext = hip_ext()
+        q, k, v, out, lse, padding_mask = ctx.saved_tensors
+        dq, dk, dv = ext.attention_bwd(
+            q, k, v, out, dout, lse, padding_mask, ctx.scale, ctx.causal
+        )
+        return dq, dk, dv, None, None

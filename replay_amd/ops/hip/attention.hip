@@ -1,0 +1,317 @@
+// Fused attention forward/backward for recommender sequence lengths (gfx950).
+//
+// K1/K2 in SURVEY §2.12: causal (SASRec) or bidirectional (BERT4Rec)
+// self-attention over item sequences, L <= 512, head_dim <= 64.  The eager
+// path runs tiny batched GEMMs (measured ~2 TF/s at [50x32]x[32x50]) plus a
+// float [B*H, L, L] mask + fp32 softmax chain; here ONE kernel per direction:
+//   - workgroup = one (batch, head); K^T and V staged in LDS
+//     (K transposed so lane k reads K_t[d][k]: bank = k%32, conflict-free);
+//   - 4 waves split the query rows; online softmax per row; LSE saved;
+//   - masking comes straight from the [B, L] bool padding mask + causal flag
+//     (the float mask tensor never exists); a row's own diagonal is always
+//     allowed (reference replay/nn/mask.py:30-51 semantics);
+//   - backward recomputes P from the saved LSE (flash-style), accumulates
+//     dK/dV in LDS with an owner-computes thread mapping (no atomics).
+//
+// Dropout is not fused (dispatch falls back to torch eager when p > 0).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "common.h"
+
+namespace {
+
+constexpr int MAX_L = 512;
+
+template <typename T>
+__global__ __launch_bounds__(256) void attn_fwd_kernel(
+    const T* __restrict__ q,    // [B, H, L, D]
+    const T* __restrict__ k,
+    const T* __restrict__ v,
+    const bool* __restrict__ valid,  // [B, L] true = real token
+    T* __restrict__ out,             // [B, H, L, D]
+    float* __restrict__ lse_out,     // [B, H, L]
+    int B, int H, int L, int D, float scale, bool causal) {
+  const int bh = blockIdx.x;
+  const int b = bh / H;
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int n_waves = blockDim.x / WAVE;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // K_t [D][Lpad] f32, V [L][D] f32, valid flags
+  const int Lpad = (L + WAVE - 1) & ~(WAVE - 1);
+  float* k_t = reinterpret_cast<float*>(smem);              // D * Lpad
+  float* v_s = k_t + (size_t)D * Lpad;                      // L * D
+  float* p_s = v_s + (size_t)L * D;                         // n_waves * L
+  unsigned char* val_s = reinterpret_cast<unsigned char*>(p_s + (size_t)n_waves * L);
+
+  const size_t base = ((size_t)bh) * L * D;
+  // cooperative load: K transposed, V row-major, both fp32 in LDS
+  for (int i = threadIdx.x; i < L * D; i += blockDim.x) {
+    const int kk = i / D, d = i % D;
+    const float kv = to_f32<T>(k[base + i]);
+    k_t[d * Lpad + kk] = kv;
+    v_s[i] = to_f32<T>(v[base + i]);
+  }
+  for (int i = threadIdx.x; i < L; i += blockDim.x) {
+    val_s[i] = valid ? (unsigned char)valid[(size_t)b * L + i] : 1;
+  }
+  __syncthreads();
+
+  float* my_p = p_s + wave * L;
+  for (int qi = wave; qi < L; qi += n_waves) {
+    // scores: lane kk handles keys kk, kk+64, ...
+    float m = -INFINITY;
+    float s_sum = 0.f;
+    // load Q row into registers (all lanes broadcast-read the same value)
+    // per-lane score loop
+    for (int kk = lane; kk < L; kk += WAVE) {
+      bool allowed = (kk == qi) || ((!causal || kk <= qi) && val_s[kk]);
+      float s;
+      if (allowed) {
+        float acc = 0.f;
+        const T* qr = q + base + (size_t)qi * D;
+        for (int d = 0; d < D; ++d) {
+          acc += to_f32<T>(qr[d]) * k_t[d * Lpad + kk];
+        }
+        s = acc * scale;
+      } else {
+        s = -INFINITY;
+      }
+      my_p[kk] = s;
+      if (s > m) m = s;
+    }
+    m = wave_reduce_max(m);
+    for (int kk = lane; kk < L; kk += WAVE) {
+      float e = (my_p[kk] == -INFINITY) ? 0.f : __expf(my_p[kk] - m);
+      my_p[kk] = e;
+      s_sum += e;
+    }
+    s_sum = wave_reduce_sum(s_sum);
+    const float inv = 1.f / s_sum;
+    if (lane == 0 && lse_out != nullptr) {
+      lse_out[(size_t)bh * L + qi] = m + __logf(s_sum);
+    }
+    // out[qi][d] = sum_k P[k] * V[k][d]; lane d owns output dims d, d+64...
+    for (int d = lane; d < D; d += WAVE) {
+      float acc = 0.f;
+      for (int kk = 0; kk < L; ++kk) {
+        acc += my_p[kk] * v_s[kk * D + d];
+      }
+      out[base + (size_t)qi * D + d] = from_f32<T>(acc * inv);
+    }
+  }
+}
+
+// Backward: one workgroup per (b, h).  Waves compute dS rows for a batch of
+// n_waves queries; then an owner-computes phase updates dK/dV in LDS.
+template <typename T>
+__global__ __launch_bounds__(256) void attn_bwd_kernel(
+    const T* __restrict__ q,
+    const T* __restrict__ k,
+    const T* __restrict__ v,
+    const T* __restrict__ out,
+    const T* __restrict__ dout,
+    const float* __restrict__ lse,   // [B, H, L]
+    const bool* __restrict__ valid,  // [B, L]
+    T* __restrict__ dq,
+    T* __restrict__ dk,
+    T* __restrict__ dv,
+    int B, int H, int L, int D, float scale, bool causal) {
+  const int bh = blockIdx.x;
+  const int b = bh / H;
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int n_waves = blockDim.x / WAVE;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int Lpad = (L + WAVE - 1) & ~(WAVE - 1);
+  float* k_t = reinterpret_cast<float*>(smem);          // D * Lpad  (K^T)
+  float* q_s = k_t + (size_t)D * Lpad;                  // L * D
+  float* v_s = q_s + (size_t)L * D;                     // L * D
+  float* do_s = v_s + (size_t)L * D;                    // L * D
+  float* dk_s = do_s + (size_t)L * D;                   // L * D (fp32 accum)
+  float* dv_s = dk_s + (size_t)L * D;                   // L * D
+  float* delta_s = dv_s + (size_t)L * D;                // L
+  float* p_rows = delta_s + L;                          // n_waves * L
+  float* ds_rows = p_rows + (size_t)n_waves * L;        // n_waves * L
+  unsigned char* val_s = reinterpret_cast<unsigned char*>(ds_rows + (size_t)n_waves * L);
+
+  const size_t base = ((size_t)bh) * L * D;
+  for (int i = threadIdx.x; i < L * D; i += blockDim.x) {
+    const int kk = i / D, d = i % D;
+    k_t[d * Lpad + kk] = to_f32<T>(k[base + i]);
+    q_s[i] = to_f32<T>(q[base + i]);
+    v_s[i] = to_f32<T>(v[base + i]);
+    do_s[i] = to_f32<T>(dout[base + i]);
+    dk_s[i] = 0.f;
+    dv_s[i] = 0.f;
+  }
+  for (int i = threadIdx.x; i < L; i += blockDim.x) {
+    val_s[i] = valid ? (unsigned char)valid[(size_t)b * L + i] : 1;
+  }
+  __syncthreads();
+  // delta[q] = sum_d dO[q][d] * O[q][d]; one wave per query row
+  for (int qi = wave; qi < L; qi += n_waves) {
+    float acc = 0.f;
+    for (int d = lane; d < D; d += WAVE) {
+      acc += do_s[qi * D + d] * to_f32<T>(out[base + (size_t)qi * D + d]);
+    }
+    acc = wave_reduce_sum(acc);
+    if (lane == 0) delta_s[qi] = acc;
+  }
+  __syncthreads();
+
+  const int KK = blockDim.x / D;  // keys updated in parallel per iteration
+  for (int q0 = 0; q0 < L; q0 += n_waves) {
+    const int qi = q0 + wave;
+    float* my_p = p_rows + wave * L;
+    float* my_ds = ds_rows + wave * L;
+    if (qi < L) {
+      const float l = lse[(size_t)bh * L + qi];
+      const float dlt = delta_s[qi];
+      for (int kk = lane; kk < L; kk += WAVE) {
+        bool allowed = (kk == qi) || ((!causal || kk <= qi) && val_s[kk]);
+        float p = 0.f, ds_v = 0.f;
+        if (allowed) {
+          float acc = 0.f;
+          float dp = 0.f;
+          for (int d = 0; d < D; ++d) {
+            acc += q_s[qi * D + d] * k_t[d * Lpad + kk];
+            dp += do_s[qi * D + d] * v_s[kk * D + d];
+          }
+          p = __expf(acc * scale - l);
+          ds_v = p * (dp - dlt) * scale;
+        }
+        my_p[kk] = p;
+        my_ds[kk] = ds_v;
+      }
+    }
+    __syncthreads();
+    // dK[k][d] += sum_{q in batch} dS_q[k] * Q[q][d]
+    // dV[k][d] += sum_{q in batch} P_q[k]  * dO[q][d]
+    // thread t owns (k = t/D + i*KK, d = t%D): no write conflicts
+    {
+      const int d = threadIdx.x % D;
+      for (int kk = threadIdx.x / D; kk < L; kk += KK) {
+        float acc_dk = 0.f, acc_dv = 0.f;
+        const int qmax = min(n_waves, L - q0);
+        for (int j = 0; j < qmax; ++j) {
+          acc_dk += ds_rows[j * L + kk] * q_s[(q0 + j) * D + d];
+          acc_dv += p_rows[j * L + kk] * do_s[(q0 + j) * D + d];
+        }
+        dk_s[kk * D + d] += acc_dk;
+        dv_s[kk * D + d] += acc_dv;
+      }
+    }
+    // dQ[qi][d] = sum_k dS[k] * K[k][d]
+    if (qi < L) {
+      for (int d = lane; d < D; d += WAVE) {
+        float acc = 0.f;
+        for (int kk = 0; kk < L; ++kk) {
+          acc += my_ds[kk] * k_t[d * Lpad + kk];
+        }
+        dq[base + (size_t)qi * D + d] = from_f32<T>(acc);
+      }
+    }
+    __syncthreads();
+  }
+  for (int i = threadIdx.x; i < L * D; i += blockDim.x) {
+    dk[base + i] = from_f32<T>(dk_s[i]);
+    dv[base + i] = from_f32<T>(dv_s[i]);
+  }
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> attention_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                         c10::optional<torch::Tensor> valid, double scale,
+                                         bool causal, bool need_lse) {
+  TORCH_CHECK(q.dim() == 4 && q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  const int B = q.size(0), H = q.size(1), L = q.size(2), D = q.size(3);
+  TORCH_CHECK(L <= MAX_L && D <= 64, "attention kernel supports L<=512, D<=64");
+  auto out = torch::empty_like(q);
+  auto lse = torch::empty({B, H, L}, q.options().dtype(torch::kFloat32));
+  const int threads = 256;
+  const int n_waves = threads / WAVE;
+  const int Lpad = (L + WAVE - 1) & ~(WAVE - 1);
+  size_t lds = sizeof(float) * ((size_t)D * Lpad + (size_t)L * D + (size_t)n_waves * L) + L;
+  lds = (lds + 15) & ~size_t(15);
+  TORCH_CHECK(lds <= 160 * 1024, "LDS overflow");
+  auto stream = at::cuda::getCurrentHIPStream();
+  const bool* valid_ptr = nullptr;
+  torch::Tensor valid_c;
+  if (valid.has_value()) {
+    valid_c = valid->contiguous();
+    TORCH_CHECK(valid_c.scalar_type() == torch::kBool);
+    valid_ptr = valid_c.data_ptr<bool>();
+  }
+#define LAUNCH_ATTN_FWD(T)                                                              \
+  hipLaunchKernelGGL(attn_fwd_kernel<T>, dim3(B * H), dim3(threads), lds, stream,       \
+                     reinterpret_cast<const T*>(q.data_ptr()),                          \
+                     reinterpret_cast<const T*>(k.data_ptr()),                          \
+                     reinterpret_cast<const T*>(v.data_ptr()), valid_ptr,               \
+                     reinterpret_cast<T*>(out.data_ptr()),                              \
+                     need_lse ? lse.data_ptr<float>() : nullptr, B, H, L, D,            \
+                     (float)scale, causal)
+  if (q.scalar_type() == torch::kBFloat16) {
+    LAUNCH_ATTN_FWD(__hip_bfloat16);
+  } else if (q.scalar_type() == torch::kFloat32) {
+    LAUNCH_ATTN_FWD(float);
+  } else if (q.scalar_type() == torch::kHalf) {
+    LAUNCH_ATTN_FWD(__half);
+  } else {
+    TORCH_CHECK(false, "unsupported dtype");
+  }
+#undef LAUNCH_ATTN_FWD
+  return {out, lse};
+}
+
+std::vector<torch::Tensor> attention_bwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                         torch::Tensor out, torch::Tensor dout,
+                                         torch::Tensor lse, c10::optional<torch::Tensor> valid,
+                                         double scale, bool causal) {
+  const int B = q.size(0), H = q.size(1), L = q.size(2), D = q.size(3);
+  auto dq = torch::empty_like(q);
+  auto dk = torch::empty_like(k);
+  auto dv = torch::empty_like(v);
+  const int threads = 256;
+  TORCH_CHECK(threads % D == 0, "head_dim must divide 256");
+  const int n_waves = threads / WAVE;
+  const int Lpad = (L + WAVE - 1) & ~(WAVE - 1);
+  size_t lds = sizeof(float) * ((size_t)D * Lpad + 5 * (size_t)L * D + L + 2 * (size_t)n_waves * L) + L;
+  lds = (lds + 15) & ~size_t(15);
+  TORCH_CHECK(lds <= 160 * 1024, "LDS overflow (reduce L or D)");
+  auto stream = at::cuda::getCurrentHIPStream();
+  const bool* valid_ptr = nullptr;
+  torch::Tensor valid_c;
+  if (valid.has_value()) {
+    valid_c = valid->contiguous();
+    valid_ptr = valid_c.data_ptr<bool>();
+  }
+  auto dout_c = dout.contiguous();
+#define LAUNCH_ATTN_BWD(T)                                                              \
+  hipLaunchKernelGGL(attn_bwd_kernel<T>, dim3(B * H), dim3(threads), lds, stream,       \
+                     reinterpret_cast<const T*>(q.data_ptr()),                          \
+                     reinterpret_cast<const T*>(k.data_ptr()),                          \
+                     reinterpret_cast<const T*>(v.data_ptr()),                          \
+                     reinterpret_cast<const T*>(out.data_ptr()),                        \
+                     reinterpret_cast<const T*>(dout_c.data_ptr()),                     \
+                     lse.data_ptr<float>(), valid_ptr,                                  \
+                     reinterpret_cast<T*>(dq.data_ptr()),                               \
+                     reinterpret_cast<T*>(dk.data_ptr()),                               \
+                     reinterpret_cast<T*>(dv.data_ptr()), B, H, L, D, (float)scale,     \
+                     causal)
+  if (q.scalar_type() == torch::kBFloat16) {
+    LAUNCH_ATTN_BWD(__hip_bfloat16);
+  } else if (q.scalar_type() == torch::kFloat32) {
+    LAUNCH_ATTN_BWD(float);
+  } else if (q.scalar_type() == torch::kHalf) {
+    LAUNCH_ATTN_BWD(__half);
+  } else {
+    TORCH_CHECK(false, "unsupported dtype");
+  }
+#undef LAUNCH_ATTN_BWD
+  return {dq, dk, dv};
+}

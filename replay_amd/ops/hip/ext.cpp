@@ -12,10 +12,19 @@ std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor labels,
 torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor labels, torch::Tensor lse,
                      torch::Tensor grad_scale, torch::Tensor count, int64_t ignore_index,
                      bool inplace);
+std::vector<torch::Tensor> attention_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                         c10::optional<torch::Tensor> valid, double scale,
+                                         bool causal, bool need_lse);
+std::vector<torch::Tensor> attention_bwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                         torch::Tensor out, torch::Tensor dout,
+                                         torch::Tensor lse, c10::optional<torch::Tensor> valid,
+                                         double scale, bool causal);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layer_norm_fwd", &layer_norm_fwd, "fused LayerNorm forward (gfx950)");
   m.def("layer_norm_bwd", &layer_norm_bwd, "fused LayerNorm backward (gfx950)");
   m.def("ce_fwd", &ce_fwd, "fused softmax-CE forward (gfx950)");
   m.def("ce_bwd", &ce_bwd, "fused softmax-CE backward (gfx950)");
+  m.def("attention_fwd", &attention_fwd, "fused attention forward (gfx950)");
+  m.def("attention_bwd", &attention_bwd, "fused attention backward (gfx950)");
 }
