@@ -16,7 +16,15 @@ from __future__ import annotations
 import argparse
 import json
 import os
+import sys
 import time
+
+if "--tunableop" in sys.argv:
+    # rocBLAS/hipBLASLt algorithm tuning for the skinny wgrad GEMM shapes;
+    # must be set before torch import
+    os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "1")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_MAX_TUNING_DURATION_MS", "120")
 
 import torch
 
@@ -80,6 +88,7 @@ def main() -> None:
     parser.add_argument("--warmup", type=int, default=5)
     parser.add_argument("--batch", type=int, default=1024, help="per-GPU batch size")
     parser.add_argument("--lr", type=float, default=1e-3)
+    parser.add_argument("--tunableop", action="store_true", help="enable rocBLAS TunableOp")
     args = parser.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))

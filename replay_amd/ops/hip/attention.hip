@@ -23,7 +23,7 @@ namespace {
 
 constexpr int MAX_L = 512;
 
-template <typename T>
+template <typename T, int DD>
 __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const T* __restrict__ q,    // [B, H, L, D]
     const T* __restrict__ k,
@@ -71,8 +71,9 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       float s;
       if (allowed) {
         float acc = 0.f;
-        const T* qr = q + base + (size_t)qi * D;
-        for (int d = 0; d < D; ++d) {
+        const T* qr = q + base + (size_t)qi * DD;
+#pragma unroll
+        for (int d = 0; d < DD; ++d) {
           acc += to_f32<T>(qr[d]) * k_t[d * Lpad + kk];
         }
         s = acc * scale;
@@ -94,19 +95,24 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       lse_out[(size_t)bh * L + qi] = m + __logf(s_sum);
     }
     // out[qi][d] = sum_k P[k] * V[k][d]; lane d owns output dims d, d+64...
-    for (int d = lane; d < D; d += WAVE) {
-      float acc = 0.f;
-      for (int kk = 0; kk < L; ++kk) {
-        acc += my_p[kk] * v_s[kk * D + d];
+    for (int d = lane; d < DD; d += WAVE) {
+      float acc0 = 0.f, acc1 = 0.f, acc2 = 0.f, acc3 = 0.f;
+      int kk = 0;
+      for (; kk + 3 < L; kk += 4) {
+        acc0 += my_p[kk] * v_s[kk * DD + d];
+        acc1 += my_p[kk + 1] * v_s[(kk + 1) * DD + d];
+        acc2 += my_p[kk + 2] * v_s[(kk + 2) * DD + d];
+        acc3 += my_p[kk + 3] * v_s[(kk + 3) * DD + d];
       }
-      out[base + (size_t)qi * D + d] = from_f32<T>(acc * inv);
+      for (; kk < L; ++kk) acc0 += my_p[kk] * v_s[kk * DD + d];
+      out[base + (size_t)qi * DD + d] = from_f32<T>((acc0 + acc1 + acc2 + acc3) * inv);
     }
   }
 }
 
 // Backward: one workgroup per (b, h).  Waves compute dS rows for a batch of
 // n_waves queries; then an owner-computes phase updates dK/dV in LDS.
-template <typename T>
+template <typename T, int DD>
 __global__ __launch_bounds__(256) void attn_bwd_kernel(
     const T* __restrict__ q,
     const T* __restrict__ k,
@@ -177,9 +183,10 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
         if (allowed) {
           float acc = 0.f;
           float dp = 0.f;
-          for (int d = 0; d < D; ++d) {
-            acc += q_s[qi * D + d] * k_t[d * Lpad + kk];
-            dp += do_s[qi * D + d] * v_s[kk * D + d];
+#pragma unroll
+          for (int d = 0; d < DD; ++d) {
+            acc += q_s[qi * DD + d] * k_t[d * Lpad + kk];
+            dp += do_s[qi * DD + d] * v_s[kk * DD + d];
           }
           p = __expf(acc * scale - l);
           ds_v = p * (dp - dlt) * scale;
@@ -193,26 +200,29 @@ __global__ __launch_bounds__(256) void attn_bwd_kernel(
     // dV[k][d] += sum_{q in batch} P_q[k]  * dO[q][d]
     // thread t owns (k = t/D + i*KK, d = t%D): no write conflicts
     {
-      const int d = threadIdx.x % D;
-      for (int kk = threadIdx.x / D; kk < L; kk += KK) {
+      const int d = threadIdx.x % DD;
+      for (int kk = threadIdx.x / DD; kk < L; kk += KK) {
         float acc_dk = 0.f, acc_dv = 0.f;
         const int qmax = min(n_waves, L - q0);
         for (int j = 0; j < qmax; ++j) {
-          acc_dk += ds_rows[j * L + kk] * q_s[(q0 + j) * D + d];
-          acc_dv += p_rows[j * L + kk] * do_s[(q0 + j) * D + d];
+          acc_dk += ds_rows[j * L + kk] * q_s[(q0 + j) * DD + d];
+          acc_dv += p_rows[j * L + kk] * do_s[(q0 + j) * DD + d];
         }
-        dk_s[kk * D + d] += acc_dk;
-        dv_s[kk * D + d] += acc_dv;
+        dk_s[kk * DD + d] += acc_dk;
+        dv_s[kk * DD + d] += acc_dv;
       }
     }
     // dQ[qi][d] = sum_k dS[k] * K[k][d]
     if (qi < L) {
-      for (int d = lane; d < D; d += WAVE) {
-        float acc = 0.f;
-        for (int kk = 0; kk < L; ++kk) {
-          acc += my_ds[kk] * k_t[d * Lpad + kk];
+      for (int d = lane; d < DD; d += WAVE) {
+        float acc0 = 0.f, acc1 = 0.f;
+        int kk = 0;
+        for (; kk + 1 < L; kk += 2) {
+          acc0 += my_ds[kk] * k_t[d * Lpad + kk];
+          acc1 += my_ds[kk + 1] * k_t[d * Lpad + kk + 1];
         }
-        dq[base + (size_t)qi * D + d] = from_f32<T>(acc);
+        for (; kk < L; ++kk) acc0 += my_ds[kk] * k_t[d * Lpad + kk];
+        dq[base + (size_t)qi * DD + d] = from_f32<T>(acc0 + acc1);
       }
     }
     __syncthreads();
@@ -247,14 +257,21 @@ std::vector<torch::Tensor> attention_fwd(torch::Tensor q, torch::Tensor k, torch
     TORCH_CHECK(valid_c.scalar_type() == torch::kBool);
     valid_ptr = valid_c.data_ptr<bool>();
   }
-#define LAUNCH_ATTN_FWD(T)                                                              \
-  hipLaunchKernelGGL(attn_fwd_kernel<T>, dim3(B * H), dim3(threads), lds, stream,       \
+#define LAUNCH_ATTN_FWD_D(T, DD)                                                        \
+  hipLaunchKernelGGL((attn_fwd_kernel<T, DD>), dim3(B * H), dim3(threads), lds, stream, \
                      reinterpret_cast<const T*>(q.data_ptr()),                          \
                      reinterpret_cast<const T*>(k.data_ptr()),                          \
                      reinterpret_cast<const T*>(v.data_ptr()), valid_ptr,               \
                      reinterpret_cast<T*>(out.data_ptr()),                              \
                      need_lse ? lse.data_ptr<float>() : nullptr, B, H, L, D,            \
                      (float)scale, causal)
+#define LAUNCH_ATTN_FWD(T)                                                              \
+  do {                                                                                  \
+    if (D == 16) LAUNCH_ATTN_FWD_D(T, 16);                                              \
+    else if (D == 32) LAUNCH_ATTN_FWD_D(T, 32);                                         \
+    else if (D == 64) LAUNCH_ATTN_FWD_D(T, 64);                                         \
+    else TORCH_CHECK(false, "head_dim must be 16/32/64");                               \
+  } while (0)
   if (q.scalar_type() == torch::kBFloat16) {
     LAUNCH_ATTN_FWD(__hip_bfloat16);
   } else if (q.scalar_type() == torch::kFloat32) {
@@ -291,8 +308,8 @@ std::vector<torch::Tensor> attention_bwd(torch::Tensor q, torch::Tensor k, torch
     valid_ptr = valid_c.data_ptr<bool>();
   }
   auto dout_c = dout.contiguous();
-#define LAUNCH_ATTN_BWD(T)                                                              \
-  hipLaunchKernelGGL(attn_bwd_kernel<T>, dim3(B * H), dim3(threads), lds, stream,       \
+#define LAUNCH_ATTN_BWD_D(T, DD)                                                        \
+  hipLaunchKernelGGL((attn_bwd_kernel<T, DD>), dim3(B * H), dim3(threads), lds, stream, \
                      reinterpret_cast<const T*>(q.data_ptr()),                          \
                      reinterpret_cast<const T*>(k.data_ptr()),                          \
                      reinterpret_cast<const T*>(v.data_ptr()),                          \
@@ -303,6 +320,13 @@ std::vector<torch::Tensor> attention_bwd(torch::Tensor q, torch::Tensor k, torch
                      reinterpret_cast<T*>(dk.data_ptr()),                               \
                      reinterpret_cast<T*>(dv.data_ptr()), B, H, L, D, (float)scale,     \
                      causal)
+#define LAUNCH_ATTN_BWD(T)                                                              \
+  do {                                                                                  \
+    if (D == 16) LAUNCH_ATTN_BWD_D(T, 16);                                              \
+    else if (D == 32) LAUNCH_ATTN_BWD_D(T, 32);                                         \
+    else if (D == 64) LAUNCH_ATTN_BWD_D(T, 64);                                         \
+    else TORCH_CHECK(false, "head_dim must be 16/32/64");                               \
+  } while (0)
   if (q.scalar_type() == torch::kBFloat16) {
     LAUNCH_ATTN_BWD(__hip_bfloat16);
   } else if (q.scalar_type() == torch::kFloat32) {
