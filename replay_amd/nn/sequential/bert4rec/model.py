@@ -93,25 +93,27 @@ class Bert4Rec(torch.nn.Module):
         """Inference: append-a-mask convention — the last valid position's id
         is replaced by the mask token and its hidden state scores the catalog
         (reference bert4rec predict flow)."""
+        from replay_amd.nn.utils import last_valid_index
+
         padding_mask = batch["padding_mask"]
         feats = dict(self._features_of(batch))
         items = feats[self.item_feature_name].clone()
-        lengths = padding_mask.long().sum(-1).clamp(min=1)
+        last_idx = last_valid_index(padding_mask)
         rows = torch.arange(items.shape[0], device=items.device)
-        items[rows, lengths - 1] = self.mask_token
+        items[rows, last_idx] = self.mask_token
         feats[self.item_feature_name] = items
         hidden = self.body(feats, padding_mask)
-        last = hidden[rows, lengths - 1]
+        last = hidden[rows, last_idx]
         return self.head(last, candidates_to_score)
 
     predict = forward_inference
 
     def get_query_embeddings(self, batch: Dict[str, torch.Tensor]) -> torch.Tensor:
+        from replay_amd.nn.utils import gather_last_valid
+
         padding_mask = batch["padding_mask"]
         hidden = self.body(self._features_of(batch), padding_mask)
-        lengths = padding_mask.long().sum(-1).clamp(min=1)
-        rows = torch.arange(hidden.shape[0], device=hidden.device)
-        return hidden[rows, lengths - 1]
+        return gather_last_valid(hidden, padding_mask)
 
     @classmethod
     def from_params(

@@ -42,10 +42,10 @@ class QueryTower(torch.nn.Module):
         return self.body.embedding_dim
 
     def forward(self, feature_tensors: Dict[str, torch.Tensor], padding_mask: torch.Tensor) -> torch.Tensor:
+        from replay_amd.nn.utils import gather_last_valid
+
         hidden = self.body(feature_tensors, padding_mask)
-        lengths = padding_mask.long().sum(-1).clamp(min=1)
-        idx = (lengths - 1).view(-1, 1, 1).expand(-1, 1, hidden.shape[-1])
-        return hidden.gather(1, idx).squeeze(1)
+        return gather_last_valid(hidden, padding_mask)
 
 
 class ItemTower(torch.nn.Module):
@@ -232,9 +232,11 @@ class TwoTower(torch.nn.Module):
         query = self._query_embedding(batch)  # [B, E]
         labels = batch["labels"]
         if labels.dim() == 2:  # last valid target position
-            lengths = batch["labels_padding_mask"].long().sum(-1).clamp(min=1)
+            from replay_amd.nn.utils import last_valid_index
+
+            last_idx = last_valid_index(batch["labels_padding_mask"])
             rows = torch.arange(labels.shape[0], device=labels.device)
-            labels = labels[rows, lengths - 1]
+            labels = labels[rows, last_idx]
         negatives = batch.get("negatives")
         if negatives is None:
             negatives = labels  # in-batch negatives

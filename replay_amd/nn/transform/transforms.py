@@ -131,8 +131,9 @@ class TokenMaskTransform(BatchTransform):
         # guarantee at least one masked position per row: mask the last valid
         rows_without = ~token_mask.any(-1)
         if rows_without.any():
-            lengths = mask.long().sum(-1).clamp(min=1)
-            last_pos = lengths - 1
+            from replay_amd.nn.utils import last_valid_index
+
+            last_pos = last_valid_index(mask)
             rows = torch.nonzero(rows_without).squeeze(-1)
             token_mask[rows, last_pos[rows]] = True
         batch[self.mask_column] = token_mask
