@@ -82,6 +82,81 @@ def make_batches(n_batches, batch_size, device, seed):
     return batches
 
 
+def serve_bench(args, device, rank, world) -> None:
+    """Inference benchmark: recs/sec@K with filter_seen over the full catalog
+    (BASELINE config 5 shape with --items/--emb-dim overrides)."""
+    import json as _json
+
+    from replay_amd.ops.topk import catalog_topk, sharded_catalog_topk
+
+    n_items = args.items
+    emb_dim = args.emb_dim
+    B = args.batch
+    torch.manual_seed(7 + rank)
+    dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
+    # random-init catalog + queries of the named shape (no checkpoints offline)
+    if world > 1:  # catalog-sharded across ranks
+        shard = n_items // world
+        lo = rank * shard
+        items_shard = torch.randn(shard, emb_dim, device=device, dtype=dtype)
+    else:
+        lo = 0
+        items_shard = torch.randn(n_items, emb_dim, device=device, dtype=dtype)
+    queries = torch.randn(B, emb_dim, device=device, dtype=dtype)
+    seen = torch.randint(0, n_items, (B, 64), device=device)
+
+    def step():
+        if world > 1:
+            return sharded_catalog_topk(queries, items_shard, args.k, lo, seen)
+        return catalog_topk(queries, items_shard, args.k, seen)
+
+    for _ in range(args.warmup):
+        step()
+    if world > 1:
+        torch.distributed.barrier()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    if world > 1:
+        torch.distributed.barrier()
+    elapsed = time.perf_counter() - t0
+    if world > 1:
+        t = torch.tensor([elapsed], device=device, dtype=torch.float64)
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+    value = args.steps * B / elapsed  # queries served per second (each -> K recs)
+    if rank == 0:
+        print(
+            _json.dumps(
+                {
+                    "metric": f"serving recs/sec@K={args.k}, full-catalog top-K with filter_seen",
+                    "value": value,
+                    "unit": "queries/sec",
+                    "n_gpus": world if world > 1 else args.gpus,
+                    "steps": args.steps,
+                    "warmup": args.warmup,
+                    "ms_per_step": elapsed / args.steps * 1000.0,
+                    "higher_is_better": True,
+                    "scaling": "strong" if world > 1 else "weak",
+                    "vs_baseline": None,
+                    "dtype": "bf16" if device.type == "cuda" else "fp32",
+                    "data": "synthetic",
+                    "config": {
+                        "model": f"catalog_topk_d{emb_dim}",
+                        "global_batch": B,
+                        "n_items": n_items,
+                        "k": args.k,
+                        "parallelism": f"catalog-sharded x{world}" if world > 1 else "single",
+                    },
+                }
+            )
+        )
+
+
 def main() -> None:
     parser = argparse.ArgumentParser()
     parser.add_argument("--gpus", type=int, default=1)
@@ -91,6 +166,10 @@ def main() -> None:
     parser.add_argument("--lr", type=float, default=1e-3)
     parser.add_argument("--tunableop", action="store_true", help="(default on)")
     parser.add_argument("--no-tunableop", action="store_true", help="disable rocBLAS TunableOp")
+    parser.add_argument("--mode", choices=["train", "serve"], default="train")
+    parser.add_argument("--items", type=int, default=10_000_000, help="serve: catalog size")
+    parser.add_argument("--emb-dim", type=int, default=256, help="serve: embedding dim")
+    parser.add_argument("--k", type=int, default=100, help="serve: top-K")
     args = parser.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -104,6 +183,12 @@ def main() -> None:
     if use_cuda:
         torch.cuda.set_device(device)
     torch.manual_seed(1234 + rank)
+
+    if args.mode == "serve":
+        serve_bench(args, device, rank, world)
+        if world > 1:
+            torch.distributed.destroy_process_group()
+        return
 
     model = build_model(device)
     if world > 1:
