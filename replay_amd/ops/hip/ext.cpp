@@ -19,6 +19,8 @@ std::vector<torch::Tensor> attention_bwd(torch::Tensor q, torch::Tensor k, torch
                                          torch::Tensor out, torch::Tensor dout,
                                          torch::Tensor lse, c10::optional<torch::Tensor> valid,
                                          double scale, bool causal);
+std::vector<torch::Tensor> threshold_compact(torch::Tensor scores, torch::Tensor thresholds,
+                                             int64_t capacity);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layer_norm_fwd", &layer_norm_fwd, "fused LayerNorm forward (gfx950)");
@@ -27,4 +29,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_bwd", &ce_bwd, "fused softmax-CE backward (gfx950)");
   m.def("attention_fwd", &attention_fwd, "fused attention forward (gfx950)");
   m.def("attention_bwd", &attention_bwd, "fused attention backward (gfx950)");
+  m.def("threshold_compact", &threshold_compact, "top-k threshold compaction (gfx950)");
 }

@@ -1,0 +1,109 @@
+// Threshold-compact candidate selection for large-catalog top-K (gfx950).
+//
+// Half of K8 in SURVEY §2.12.  torch.topk over [B, V~10M] score rows is the
+// serving bottleneck (measured 68 of 72 ms/step); exact top-k needs only ONE
+// full pass when a per-row threshold T ~ kth value is known from a strided
+// subsample: elements >= T are compacted (value, index) into a small buffer,
+// the final top-k is a cheap [B, ~4k] torch.topk.  The caller retries with a
+// relaxed/tightened threshold on under/overflow (expected never on real
+// score distributions).
+//
+// One atomicAdd per wave per 512 elements (ballot-aggregated), 16-B
+// vectorized bf16 loads (guide G13).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "common.h"
+
+namespace {
+
+template <typename T>
+__global__ void threshold_compact_kernel(
+    const T* __restrict__ scores,  // [B, C]
+    const float* __restrict__ thresholds,  // [B]
+    float* __restrict__ out_vals,   // [B, M]
+    int* __restrict__ out_idx,      // [B, M]
+    int* __restrict__ counts,       // [B]
+    int64_t B, int64_t C, int M) {
+  // grid: x = segments over C, y = row
+  const int row = blockIdx.y;
+  const float T_row = thresholds[row];
+  const T* sr = scores + (size_t)row * C;
+  const int lane = threadIdx.x & (WAVE - 1);
+
+  const int64_t seg_elems = 8;  // per lane per iteration
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x * seg_elems;
+  for (int64_t base = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * seg_elems; base < C;
+       base += stride) {
+    float v[8];
+    int n_here = (int)min((int64_t)8, C - base);
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      v[i] = (i < n_here) ? to_f32<T>(sr[base + i]) : -INFINITY;
+    }
+    unsigned cnt = 0;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) cnt += (v[i] >= T_row);
+    // wave-aggregate: prefix sum of cnt across lanes, one atomic per wave
+    unsigned prefix = cnt;
+#pragma unroll
+    for (int off = 1; off < WAVE; off <<= 1) {
+      unsigned up = __shfl_up(prefix, off, WAVE);
+      if (lane >= off) prefix += up;
+    }
+    unsigned total = __shfl(prefix, WAVE - 1, WAVE);
+    if (total == 0) continue;
+    int wave_base = 0;
+    if (lane == WAVE - 1) {
+      wave_base = atomicAdd(&counts[row], (int)total);
+    }
+    wave_base = __shfl(wave_base, WAVE - 1, WAVE);
+    int pos = wave_base + (int)(prefix - cnt);
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      if (v[i] >= T_row) {
+        if (pos < M) {
+          out_vals[(size_t)row * M + pos] = v[i];
+          out_idx[(size_t)row * M + pos] = (int)(base + i);
+        }
+        ++pos;
+      }
+    }
+  }
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> threshold_compact(torch::Tensor scores, torch::Tensor thresholds,
+                                             int64_t capacity) {
+  TORCH_CHECK(scores.is_cuda() && scores.dim() == 2 && scores.is_contiguous());
+  const int64_t B = scores.size(0), C = scores.size(1);
+  auto opts_f = scores.options().dtype(torch::kFloat32);
+  auto opts_i = scores.options().dtype(torch::kInt32);
+  auto out_vals = torch::full({B, capacity}, -std::numeric_limits<float>::infinity(), opts_f);
+  auto out_idx = torch::zeros({B, capacity}, opts_i);
+  auto counts = torch::zeros({B}, opts_i);
+  auto thr = thresholds.to(torch::kFloat32).contiguous();
+  const int threads = 256;
+  // enough segments to fill the chip: >> 256 workgroups total
+  int seg_blocks = (int)std::min<int64_t>((C + threads * 8 - 1) / (threads * 8), std::max<int64_t>(1, 4096 / B + 1));
+  dim3 grid(seg_blocks, B);
+  auto stream = at::cuda::getCurrentHIPStream();
+#define LAUNCH_TC(T)                                                                    \
+  hipLaunchKernelGGL(threshold_compact_kernel<T>, grid, dim3(threads), 0, stream,       \
+                     reinterpret_cast<const T*>(scores.data_ptr()),                     \
+                     thr.data_ptr<float>(), out_vals.data_ptr<float>(),                 \
+                     out_idx.data_ptr<int>(), counts.data_ptr<int>(), B, C,             \
+                     (int)capacity)
+  if (scores.scalar_type() == torch::kBFloat16) {
+    LAUNCH_TC(__hip_bfloat16);
+  } else if (scores.scalar_type() == torch::kFloat32) {
+    LAUNCH_TC(float);
+  } else if (scores.scalar_type() == torch::kHalf) {
+    LAUNCH_TC(__half);
+  } else {
+    TORCH_CHECK(false, "unsupported dtype");
+  }
+#undef LAUNCH_TC
+  return {out_vals, out_idx, counts};
+}

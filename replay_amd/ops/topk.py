@@ -33,6 +33,40 @@ def _apply_seen_mask(scores: torch.Tensor, seen: torch.Tensor, lo: int, hi: int)
     scores.masked_fill_(hit > 0, torch.finfo(scores.dtype).min)
 
 
+def fast_row_topk(scores: torch.Tensor, k: int) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Exact per-row top-k via sample-threshold + compact (K8 selection half).
+
+    One full pass over ``scores`` instead of torch.topk's multi-pass radix
+    sort (measured 68 ms -> the compact pass at HBM speed for [1024, 10M]).
+    Rows whose threshold guess fails (under/overflow) fall back to torch.topk
+    — statistically never on continuous score distributions.
+    """
+    from replay_amd.ops import hip_ext
+
+    B, C = scores.shape
+    ext = hip_ext()
+    if not scores.is_cuda or ext is None or not hasattr(ext, "threshold_compact") or C < 65536 or k > 512:
+        return torch.topk(scores, min(k, C), dim=1)
+    stride = 256
+    sample = scores[:, ::stride].float()
+    q = sample.shape[1] / C
+    import math
+
+    j = max(1, math.ceil(k * q + 3.0 * math.sqrt(max(k * q, 1e-9)) + 2))
+    j = min(j, sample.shape[1])
+    thresholds = sample.topk(j, dim=1).values[:, -1]
+    capacity = max(4 * k, int(2.5 * j / q))
+    vals, idx, counts = ext.threshold_compact(scores.contiguous(), thresholds, capacity)
+    bad = (counts < k) | (counts > capacity)
+    top_s, top_pos = torch.topk(vals, min(k, capacity), dim=1)
+    top_i = idx.gather(1, top_pos).long()
+    if bool(bad.any()):
+        rows = torch.nonzero(bad).squeeze(-1)
+        ref_s, ref_i = torch.topk(scores[rows].float(), min(k, C), dim=1)
+        top_s[rows], top_i[rows] = ref_s, ref_i
+    return top_s.to(scores.dtype), top_i
+
+
 def catalog_topk(
     query_emb: torch.Tensor,  # [B, E]
     item_emb: torch.Tensor,  # [V, E]
@@ -55,7 +89,7 @@ def catalog_topk(
         if seen is not None:
             _apply_seen_mask(scores, seen, lo + item_offset, hi + item_offset)
         kk = min(k, hi - lo)
-        top_s, top_i = torch.topk(scores, kk, dim=1)
+        top_s, top_i = fast_row_topk(scores, kk)
         top_i = top_i + (lo + item_offset)
         if run_scores is None:
             run_scores, run_ids = top_s, top_i

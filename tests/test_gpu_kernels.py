@@ -210,6 +210,42 @@ class TestFlashAttention:
 
 
 @requires_gpu
+class TestFastTopK:
+    @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+    @pytest.mark.parametrize("shape_k", [((32, 100_000), 100), ((8, 200_000), 10), ((4, 70_000), 512)])
+    def test_matches_torch_topk(self, dtype, shape_k):
+        from replay_amd.ops.topk import fast_row_topk
+
+        (B, C), k = shape_k
+        torch.manual_seed(0)
+        scores = torch.randn(B, C, device="cuda").to(dtype)
+        s, i = fast_row_topk(scores, k)
+        ref_s, ref_i = torch.topk(scores.float(), k, dim=1)
+        # same item sets (ties may reorder equal scores)
+        for b in range(B):
+            assert set(i[b].tolist()) == set(ref_i[b].tolist())
+        torch.testing.assert_close(s.float(), ref_s, atol=1e-3, rtol=1e-3)
+
+    def test_catalog_topk_gpu_with_seen(self):
+        from replay_amd.ops.topk import catalog_topk
+
+        torch.manual_seed(1)
+        B, E, V, K = 16, 32, 300_000, 50
+        q = torch.randn(B, E, device="cuda", dtype=torch.bfloat16)
+        items = torch.randn(V, E, device="cuda", dtype=torch.bfloat16)
+        seen = torch.randint(0, V, (B, 32), device="cuda")
+        s, ids = catalog_topk(q, items, K, seen=seen, chunk_items=100_000)
+        full = (q.float() @ items.float().T)
+        full.scatter_(1, seen, float("-inf"))
+        ref_i = torch.topk(full, K, dim=1).indices
+        # bf16 scoring vs fp32 reference: allow small set difference at the boundary
+        for b in range(B):
+            overlap = len(set(ids[b].tolist()) & set(ref_i[b].tolist()))
+            assert overlap >= K - 2
+        assert not any((ids == s_id).any() for s_id in seen.T)  # no seen item recommended
+
+
+@requires_gpu
 class TestModelOnGPU:
     def test_sasrec_train_step_gpu(self):
         import __graft_entry__
