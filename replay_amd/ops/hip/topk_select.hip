@@ -33,10 +33,13 @@ __global__ void threshold_compact_kernel(
 
   const int64_t seg_elems = 8;  // per lane per iteration
   const int64_t stride = (int64_t)gridDim.x * blockDim.x * seg_elems;
-  for (int64_t base = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * seg_elems; base < C;
-       base += stride) {
+  // LOOP CONDITION MUST BE WAVE-UNIFORM: a per-lane `base < C` lets boundary
+  // lanes exit while the rest shuffle against inactive lanes (undefined).
+  const int64_t wave_first = ((int64_t)blockIdx.x * blockDim.x + (threadIdx.x & ~(WAVE - 1))) * seg_elems;
+  for (int64_t wbase = wave_first; wbase < C; wbase += stride) {
+    const int64_t base = wbase + (int64_t)lane * seg_elems;
     float v[8];
-    int n_here = (int)min((int64_t)8, C - base);
+    const int n_here = (int)max((int64_t)0, min((int64_t)8, C - base));
 #pragma unroll
     for (int i = 0; i < 8; ++i) {
       v[i] = (i < n_here) ? to_f32<T>(sr[base + i]) : -INFINITY;
