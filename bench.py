@@ -19,13 +19,18 @@ import os
 import sys
 import time
 
-if "--no-tunableop" not in sys.argv:
+_serve_mode = "serve" in sys.argv
+if "--no-tunableop" not in sys.argv and not _serve_mode:
     # rocBLAS/hipBLASLt algorithm tuning (split-K for the skinny wgrad GEMM
     # shapes; +10% step time measured); must be set before torch import.
-    # Tuning happens on each shape's first (warmup) call.
+    # Tuning happens on each shape's first (warmup) call.  Disabled in serve
+    # mode: tuning probes of bad tiles on [B, 2M]-wide GEMMs cost seconds each.
     os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
     os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "1")
-    os.environ.setdefault("PYTORCH_TUNABLEOP_MAX_TUNING_DURATION_MS", "120")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_MAX_TUNING_DURATION_MS", "60")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_MAX_TUNING_ITERATIONS", "5")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_MAX_WARMUP_DURATION_MS", "30")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_MAX_WARMUP_ITERATIONS", "2")
 
 import torch
 

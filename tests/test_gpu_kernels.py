@@ -221,10 +221,14 @@ class TestFastTopK:
         scores = torch.randn(B, C, device="cuda").to(dtype)
         s, i = fast_row_topk(scores, k)
         ref_s, ref_i = torch.topk(scores.float(), k, dim=1)
-        # same item sets (ties may reorder equal scores)
+        # score multisets must match exactly (bf16 quantization TIES at the
+        # kth boundary make index sets legitimately ambiguous)
+        torch.testing.assert_close(s.float(), ref_s, atol=0, rtol=0)
+        # every selected index's score >= the kth reference score, no dupes
+        sel = scores.float().gather(1, i)
+        assert (sel >= ref_s[:, -1:]).all()
         for b in range(B):
-            assert set(i[b].tolist()) == set(ref_i[b].tolist())
-        torch.testing.assert_close(s.float(), ref_s, atol=1e-3, rtol=1e-3)
+            assert len(set(i[b].tolist())) == k
 
     def test_catalog_topk_gpu_with_seen(self):
         from replay_amd.ops.topk import catalog_topk
@@ -235,14 +239,16 @@ class TestFastTopK:
         items = torch.randn(V, E, device="cuda", dtype=torch.bfloat16)
         seen = torch.randint(0, V, (B, 32), device="cuda")
         s, ids = catalog_topk(q, items, K, seen=seen, chunk_items=100_000)
-        full = (q.float() @ items.float().T)
+        # reference computed in the SAME bf16 scoring precision
+        full = torch.cat([(q @ items[lo : lo + 100_000].T).float() for lo in range(0, V, 100_000)], dim=1)
         full.scatter_(1, seen, float("-inf"))
-        ref_i = torch.topk(full, K, dim=1).indices
-        # bf16 scoring vs fp32 reference: allow small set difference at the boundary
+        ref_s = torch.topk(full, K, dim=1).values
+        torch.testing.assert_close(s.float(), ref_s, atol=0, rtol=0)
+        sel = full.gather(1, ids)
+        assert (sel >= ref_s[:, -1:]).all()  # tie-valid selection
+        seen_sets = [set(r.tolist()) for r in seen]
         for b in range(B):
-            overlap = len(set(ids[b].tolist()) & set(ref_i[b].tolist()))
-            assert overlap >= K - 2
-        assert not any((ids == s_id).any() for s_id in seen.T)  # no seen item recommended
+            assert seen_sets[b].isdisjoint(set(ids[b].tolist()))
 
 
 @requires_gpu
