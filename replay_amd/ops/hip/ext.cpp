@@ -20,7 +20,9 @@ std::vector<torch::Tensor> attention_bwd(torch::Tensor q, torch::Tensor k, torch
                                          torch::Tensor lse, c10::optional<torch::Tensor> valid,
                                          double scale, bool causal);
 std::vector<torch::Tensor> threshold_compact(torch::Tensor scores, torch::Tensor thresholds,
-                                             int64_t capacity);
+                                             int64_t capacity,
+                                             c10::optional<torch::Tensor> seen,
+                                             int64_t col_offset);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layer_norm_fwd", &layer_norm_fwd, "fused LayerNorm forward (gfx950)");

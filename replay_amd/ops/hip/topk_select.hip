@@ -4,12 +4,17 @@
 // serving bottleneck (measured 68 of 72 ms/step); exact top-k needs only ONE
 // full pass when a per-row threshold T ~ kth value is known from a strided
 // subsample: elements >= T are compacted (value, index) into a small buffer,
-// the final top-k is a cheap [B, ~4k] torch.topk.  The caller retries with a
-// relaxed/tightened threshold on under/overflow (expected never on real
-// score distributions).
+// the final top-k is a cheap [B, ~4k] torch.topk.  The caller retries /
+// falls back on under/overflow (expected never on real score distributions).
 //
-// One atomicAdd per wave per 512 elements (ballot-aggregated), 16-B
-// vectorized bf16 loads (guide G13).
+// The seen-item filter (the other half of K8, reference
+// postprocessor/seen_items.py:56-83) is folded in: candidates whose global
+// id is in the query's seen list are dropped AT COMPACTION, so the eager
+// [B, C] mask passes (measured 21 ms/step of zeros+scatter+where) vanish.
+//
+// 16-B vectorized bf16 loads (guide G13); one atomicAdd per wave per 512
+// elements (ballot-aggregated); wave-uniform loop bounds (partial-wave
+// shuffles are undefined).
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -17,37 +22,73 @@
 
 namespace {
 
-template <typename T>
+__device__ __forceinline__ void load8_bf16_tk(const __hip_bfloat16* p, float* out) {
+  const uint4 raw = *reinterpret_cast<const uint4*>(p);
+  const unsigned w[4] = {raw.x, raw.y, raw.z, raw.w};
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    __hip_bfloat162 pair = *reinterpret_cast<const __hip_bfloat162*>(&w[i]);
+    out[2 * i] = __bfloat162float(pair.x);
+    out[2 * i + 1] = __bfloat162float(pair.y);
+  }
+}
+
+template <typename T, bool VEC8>
 __global__ void threshold_compact_kernel(
-    const T* __restrict__ scores,  // [B, C]
+    const T* __restrict__ scores,          // [B, C]
     const float* __restrict__ thresholds,  // [B]
-    float* __restrict__ out_vals,   // [B, M]
-    int* __restrict__ out_idx,      // [B, M]
-    int* __restrict__ counts,       // [B]
-    int64_t B, int64_t C, int M) {
-  // grid: x = segments over C, y = row
+    const int64_t* __restrict__ seen,      // [B, S] global ids or nullptr
+    float* __restrict__ out_vals,          // [B, M]
+    int* __restrict__ out_idx,             // [B, M]
+    int* __restrict__ counts,              // [B]
+    int64_t B, int64_t C, int M, int S, int64_t col_offset) {
   const int row = blockIdx.y;
   const float T_row = thresholds[row];
   const T* sr = scores + (size_t)row * C;
   const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t* seen_row = seen ? seen + (size_t)row * S : nullptr;
 
-  const int64_t seg_elems = 8;  // per lane per iteration
+  const int64_t seg_elems = 8;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x * seg_elems;
-  // LOOP CONDITION MUST BE WAVE-UNIFORM: a per-lane `base < C` lets boundary
-  // lanes exit while the rest shuffle against inactive lanes (undefined).
-  const int64_t wave_first = ((int64_t)blockIdx.x * blockDim.x + (threadIdx.x & ~(WAVE - 1))) * seg_elems;
+  const int64_t wave_first =
+      ((int64_t)blockIdx.x * blockDim.x + (threadIdx.x & ~(WAVE - 1))) * seg_elems;
   for (int64_t wbase = wave_first; wbase < C; wbase += stride) {
     const int64_t base = wbase + (int64_t)lane * seg_elems;
     float v[8];
     const int n_here = (int)max((int64_t)0, min((int64_t)8, C - base));
+    if constexpr (VEC8) {
+      if (n_here == 8) {
+        load8_bf16_tk(reinterpret_cast<const __hip_bfloat16*>(sr) + base, v);
+      } else {
+#pragma unroll
+        for (int i = 0; i < 8; ++i) v[i] = (i < n_here) ? to_f32<T>(sr[base + i]) : -INFINITY;
+      }
+    } else {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) v[i] = (i < n_here) ? to_f32<T>(sr[base + i]) : -INFINITY;
+    }
+    unsigned keep = 0;  // bitmask of kept candidates
 #pragma unroll
     for (int i = 0; i < 8; ++i) {
-      v[i] = (i < n_here) ? to_f32<T>(sr[base + i]) : -INFINITY;
+      if (v[i] >= T_row) keep |= (1u << i);
     }
-    unsigned cnt = 0;
+    if (keep && seen_row) {
+      // drop candidates present in the query's seen list (rare path:
+      // only candidates above threshold pay the scan)
 #pragma unroll
-    for (int i = 0; i < 8; ++i) cnt += (v[i] >= T_row);
-    // wave-aggregate: prefix sum of cnt across lanes, one atomic per wave
+      for (int i = 0; i < 8; ++i) {
+        if (keep & (1u << i)) {
+          const int64_t gid = col_offset + base + i;
+          for (int s = 0; s < S; ++s) {
+            if (seen_row[s] == gid) {
+              keep &= ~(1u << i);
+              break;
+            }
+          }
+        }
+      }
+    }
+    unsigned cnt = __popc(keep);
     unsigned prefix = cnt;
 #pragma unroll
     for (int off = 1; off < WAVE; off <<= 1) {
@@ -64,7 +105,7 @@ __global__ void threshold_compact_kernel(
     int pos = wave_base + (int)(prefix - cnt);
 #pragma unroll
     for (int i = 0; i < 8; ++i) {
-      if (v[i] >= T_row) {
+      if (keep & (1u << i)) {
         if (pos < M) {
           out_vals[(size_t)row * M + pos] = v[i];
           out_idx[(size_t)row * M + pos] = (int)(base + i);
@@ -78,7 +119,9 @@ __global__ void threshold_compact_kernel(
 }  // namespace
 
 std::vector<torch::Tensor> threshold_compact(torch::Tensor scores, torch::Tensor thresholds,
-                                             int64_t capacity) {
+                                             int64_t capacity,
+                                             c10::optional<torch::Tensor> seen,
+                                             int64_t col_offset) {
   TORCH_CHECK(scores.is_cuda() && scores.dim() == 2 && scores.is_contiguous());
   const int64_t B = scores.size(0), C = scores.size(1);
   auto opts_f = scores.options().dtype(torch::kFloat32);
@@ -87,23 +130,35 @@ std::vector<torch::Tensor> threshold_compact(torch::Tensor scores, torch::Tensor
   auto out_idx = torch::zeros({B, capacity}, opts_i);
   auto counts = torch::zeros({B}, opts_i);
   auto thr = thresholds.to(torch::kFloat32).contiguous();
+  const int64_t* seen_ptr = nullptr;
+  int S = 0;
+  torch::Tensor seen_c;
+  if (seen.has_value()) {
+    seen_c = seen->contiguous();
+    TORCH_CHECK(seen_c.scalar_type() == torch::kLong && seen_c.size(0) == B);
+    seen_ptr = seen_c.data_ptr<int64_t>();
+    S = (int)seen_c.size(1);
+  }
   const int threads = 256;
-  // enough segments to fill the chip: >> 256 workgroups total
-  int seg_blocks = (int)std::min<int64_t>((C + threads * 8 - 1) / (threads * 8), std::max<int64_t>(1, 4096 / B + 1));
+  int seg_blocks =
+      (int)std::min<int64_t>((C + threads * 8 - 1) / (threads * 8), std::max<int64_t>(1, 4096 / B + 1));
   dim3 grid(seg_blocks, B);
   auto stream = at::cuda::getCurrentHIPStream();
-#define LAUNCH_TC(T)                                                                    \
-  hipLaunchKernelGGL(threshold_compact_kernel<T>, grid, dim3(threads), 0, stream,       \
-                     reinterpret_cast<const T*>(scores.data_ptr()),                     \
-                     thr.data_ptr<float>(), out_vals.data_ptr<float>(),                 \
-                     out_idx.data_ptr<int>(), counts.data_ptr<int>(), B, C,             \
-                     (int)capacity)
+  // 16-B row alignment for the vector path (bf16 rows start 16B-aligned when
+  // C % 8 == 0; otherwise scalar)
+  const bool vec8 = scores.scalar_type() == torch::kBFloat16 && (C % 8 == 0);
+#define LAUNCH_TC(T, V)                                                                   \
+  hipLaunchKernelGGL((threshold_compact_kernel<T, V>), grid, dim3(threads), 0, stream,    \
+                     reinterpret_cast<const T*>(scores.data_ptr()), thr.data_ptr<float>(),\
+                     seen_ptr, out_vals.data_ptr<float>(), out_idx.data_ptr<int>(),       \
+                     counts.data_ptr<int>(), B, C, (int)capacity, S, col_offset)
   if (scores.scalar_type() == torch::kBFloat16) {
-    LAUNCH_TC(__hip_bfloat16);
+    if (vec8) LAUNCH_TC(__hip_bfloat16, true);
+    else LAUNCH_TC(__hip_bfloat16, false);
   } else if (scores.scalar_type() == torch::kFloat32) {
-    LAUNCH_TC(float);
+    LAUNCH_TC(float, false);
   } else if (scores.scalar_type() == torch::kHalf) {
-    LAUNCH_TC(__half);
+    LAUNCH_TC(__half, false);
   } else {
     TORCH_CHECK(false, "unsupported dtype");
   }
