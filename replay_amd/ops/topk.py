@@ -75,7 +75,11 @@ def fast_row_topk(
     if seen is not None:
         gid = idx.long() + col_offset  # [B, capacity] global ids
         written = torch.isfinite(vals)  # unwritten slots carry idx=0: exclude
-        hit = (gid.unsqueeze(-1) == seen.unsqueeze(1)).any(-1) & written
+        # membership via per-row binary search in the sorted seen list
+        # (a broadcast compare + .any over [B, cap, S] costs ~10 ms/step)
+        sorted_seen, _ = seen.sort(dim=1)
+        pos = torch.searchsorted(sorted_seen, gid).clamp(max=sorted_seen.shape[1] - 1)
+        hit = (sorted_seen.gather(1, pos) == gid) & written
         vals = vals.masked_fill(hit, float("-inf"))
         survivors = counts.clamp(max=capacity) - hit.sum(-1, dtype=counts.dtype)
         bad = (survivors < k) | (counts > capacity)
