@@ -1,0 +1,13 @@
+"""Legacy-surface NN models (reference replay/models/nn — layer 8).
+
+The reference ships two generations of its NN stack; this namespace keeps
+the legacy classes' constructor/loss surface while reusing the single
+MI355X transformer implementation underneath.
+"""
+
+from .sequential.bert4rec import Bert4Rec
+from .sequential.compiled import Bert4RecCompiled, SasRecCompiled
+from .sequential.sasrec import SasRec
+from .sequential.tisasrec import TiSasRec
+
+__all__ = ["Bert4Rec", "Bert4RecCompiled", "SasRecCompiled", "SasRec", "TiSasRec"]
