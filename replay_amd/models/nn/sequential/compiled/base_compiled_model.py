@@ -52,7 +52,7 @@ class BaseCompiledModel:
         self.device = device
         self.item_feature_name = item_feature_name
         model = model.to(device).eval()
-        wrapper = _InferenceWrapper(model, item_feature_name)
+        wrapper = _InferenceWrapper(model, item_feature_name).eval()
         example_b = self.batch_size or 2
         example = (
             torch.zeros(example_b, max_seq_len, dtype=torch.long, device=device),
