@@ -23,6 +23,8 @@ std::vector<torch::Tensor> threshold_compact(torch::Tensor scores, torch::Tensor
                                              int64_t capacity,
                                              c10::optional<torch::Tensor> seen,
                                              int64_t col_offset);
+std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
+                                            torch::Tensor thresholds, int64_t capacity);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layer_norm_fwd", &layer_norm_fwd, "fused LayerNorm forward (gfx950)");
@@ -32,4 +34,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attention_fwd", &attention_fwd, "fused attention forward (gfx950)");
   m.def("attention_bwd", &attention_bwd, "fused attention backward (gfx950)");
   m.def("threshold_compact", &threshold_compact, "top-k threshold compaction (gfx950)");
+  m.def("scored_topk_gemm", &scored_topk_gemm,
+        "fused MFMA score-GEMM + top-k candidate selection (gfx950)");
 }

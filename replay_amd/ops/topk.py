@@ -97,6 +97,76 @@ def fast_row_topk(
     return top_s.to(scores.dtype), top_i
 
 
+def fused_catalog_topk(
+    query_emb: torch.Tensor,  # [M, E] bf16
+    item_emb: torch.Tensor,  # [V, E] bf16
+    k: int,
+    seen: Optional[torch.Tensor] = None,
+    item_offset: int = 0,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Single-pass MFMA GEMM + in-epilogue selection (K7+K8 fused): the
+    [M, V] score matrix never reaches HBM.  Exactness contract as
+    fast_row_topk; falls back to the chunked path for rows whose threshold
+    guess fails."""
+    import math
+
+    from replay_amd.ops import hip_ext
+
+    ext = hip_ext()
+    M, E = query_emb.shape
+    V = item_emb.shape[0]
+    # threshold estimate from a strided subsample (small hipBLASLt GEMM)
+    stride = max(1, V // 32768)
+    sample_items = item_emb[::stride].contiguous()
+    sample = (query_emb @ sample_items.to(query_emb.dtype).T).float()
+    qr = sample.shape[1] / V
+    j = max(1, math.ceil(k * qr + 3.0 * math.sqrt(max(k * qr, 1e-9)) + 2))
+    if seen is not None:
+        j += math.ceil(seen.shape[1] * qr) + 1
+    j = min(j, sample.shape[1])
+    thresholds = sample.topk(j, dim=1).values[:, -1]
+    capacity = max(4 * k, int(2.5 * j / qr))
+    vals, idx, counts = ext.scored_topk_gemm(query_emb.contiguous(), item_emb.contiguous(), thresholds, capacity)
+    if seen is not None:
+        gid = idx.long() + item_offset
+        written = torch.isfinite(vals)
+        sorted_seen, _ = seen.sort(dim=1)
+        pos = torch.searchsorted(sorted_seen, gid).clamp(max=sorted_seen.shape[1] - 1)
+        hit = (sorted_seen.gather(1, pos) == gid) & written
+        vals = vals.masked_fill(hit, float("-inf"))
+        survivors = counts.clamp(max=capacity) - hit.sum(-1, dtype=counts.dtype)
+        bad = (survivors < k) | (counts > capacity)
+    else:
+        bad = (counts < k) | (counts > capacity)
+    kk = min(k, capacity)
+    top_s, top_pos = torch.topk(vals, kk, dim=1)
+    top_i = idx.gather(1, top_pos).long() + item_offset
+    if bool(bad.any()):
+        rows = torch.nonzero(bad).squeeze(-1)
+        sub_s, sub_i = catalog_topk(
+            query_emb[rows], item_emb, kk,
+            seen=seen[rows] if seen is not None else None,
+            item_offset=item_offset, _allow_fused=False,
+        )
+        top_s[rows], top_i[rows] = sub_s.float(), sub_i
+    return top_s.to(query_emb.dtype), top_i
+
+
+def _can_fuse(query_emb, item_emb) -> bool:
+    from replay_amd.ops import hip_ext
+
+    ext = hip_ext()
+    return (
+        query_emb.is_cuda
+        and ext is not None
+        and hasattr(ext, "scored_topk_gemm")
+        and query_emb.dtype == torch.bfloat16
+        and item_emb.dtype == torch.bfloat16
+        and query_emb.shape[1] in (64, 128, 256)
+        and item_emb.shape[0] >= 65536
+    )
+
+
 def catalog_topk(
     query_emb: torch.Tensor,  # [B, E]
     item_emb: torch.Tensor,  # [V, E]
@@ -104,12 +174,15 @@ def catalog_topk(
     seen: Optional[torch.Tensor] = None,  # [B, S] global item ids, -1 padded
     chunk_items: int = 2**21,
     item_offset: int = 0,
+    _allow_fused: bool = True,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Returns (scores [B, k], item_ids [B, k]) over the item table, with
     seen items filtered to -inf before selection (exact filter_seen parity
     with the reference anti-join semantics, base_rec.py:152-201)."""
     V = item_emb.shape[0]
     k = min(k, V)
+    if _allow_fused and k <= 512 and _can_fuse(query_emb, item_emb):
+        return fused_catalog_topk(query_emb, item_emb, k, seen, item_offset)
     run_scores: Optional[torch.Tensor] = None
     run_ids: Optional[torch.Tensor] = None
     for lo in range(0, V, chunk_items):
