@@ -7,13 +7,15 @@
 // pipeline writes + re-reads 40 GB of scores per batch; this kernel's HBM
 // traffic is the item table itself (V x E bf16, streamed once).
 //
-// Geometry: one workgroup = 4 waves = a 64-query M-tile; each wave owns 16
-// query rows, keeps their A-fragments (16 x E bf16) RESIDENT in VGPRs for
-// the whole launch (E <= 256 -> 32 VGPRs), and streams 64-item B-tiles with
-// direct 16-B global loads (the 4 waves of a WG read the same W lines ->
-// L1/L2 reuse).  Candidates (score >= per-query threshold) are rare by
+// Geometry: one workgroup = 4 waves = a 256-query M-tile; each wave owns 64
+// query rows as 4 row-fragments whose A-fragments (64 x E bf16) stay
+// RESIDENT in VGPRs for the whole launch (E = 256 -> 128 VGPRs/lane), and
+// streams 64-item B-tiles with direct 16-B global loads shared by all 4
+// row-fragments (16 MFMAs per 4 B loads per K-step).  The wide M-tile is
+// the W-traffic lever: item-table bytes scale with ceil(M/256), not
+// ceil(M/64).  Candidates (score >= per-query threshold) are rare by
 // construction (threshold ~ kth value from a host-side subsample), so the
-// epilogue's common path is 16 VALU compares per tile; hits append
+// epilogue's common path is 64 VALU compares per tile; hits append
 // (value, index) to per-query buffers via one global atomicAdd each.
 //
 // Fragment maps (verified by the asymmetric-B GPU parity test, guide G9):
@@ -30,8 +32,11 @@ namespace {
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
-template <int E>  // embedding dim, multiple of 32, <= 256
-__global__ __launch_bounds__(256) void scored_topk_gemm_kernel(
+// RESIDENT: keep the full A (64 rows x E) in VGPRs across tiles (E <= 128);
+// at E = 256 that spills (needs 128 VGPRs for A alone), so A-fragments are
+// re-read per K-step from L2 instead (Q is ~0.5 MB, fully L2-resident).
+template <int E, bool RESIDENT>
+__global__ __launch_bounds__(256, 2) void scored_topk_gemm_kernel(
     const __hip_bfloat16* __restrict__ q,  // [M, E]
     const __hip_bfloat16* __restrict__ w,  // [V, E]
     const float* __restrict__ thresholds,  // [M]
@@ -40,39 +45,54 @@ __global__ __launch_bounds__(256) void scored_topk_gemm_kernel(
     int* __restrict__ counts,              // [M]
     int M, int64_t V, int cap) {
   constexpr int KSTEPS = E / 32;
+  constexpr int MF = 4;  // row-fragments per wave (wave covers 64 rows)
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
-  const int m0 = blockIdx.x * 64 + wave * 16;  // this wave's first query row
+  const int m0 = blockIdx.x * 256 + wave * 64;  // this wave's first query row
 
-  // ---- load this wave's A fragments (Q rows), resident for the launch ----
-  bf16x8 a_frag[KSTEPS];
-  {
-    const int row = m0 + (lane & 15);
+  // ---- A fragments (64 Q rows): resident across tiles when they fit ----
+  constexpr int A_KS = RESIDENT ? KSTEPS : 1;
+  bf16x8 a_frag[MF][A_KS];
+  const __hip_bfloat16* a_rows[MF];
+  bool a_row_valid[MF];
+#pragma unroll
+  for (int mf = 0; mf < MF; ++mf) {
+    const int row = m0 + mf * 16 + (lane & 15);
     const int k0 = (lane >> 4) * 8;
-    const __hip_bfloat16* qr = q + (size_t)min(row, M - 1) * E;
+    a_rows[mf] = q + (size_t)min(row, M - 1) * E + k0;
+    a_row_valid[mf] = row < M;
+    if constexpr (RESIDENT) {
 #pragma unroll
-    for (int ks = 0; ks < KSTEPS; ++ks) {
-      a_frag[ks] = *reinterpret_cast<const bf16x8*>(qr + ks * 32 + k0);
-    }
-    if (row >= M) {
+      for (int ks = 0; ks < KSTEPS; ++ks) {
+        a_frag[mf][ks] = *reinterpret_cast<const bf16x8*>(a_rows[mf] + ks * 32);
+      }
+      if (!a_row_valid[mf]) {
 #pragma unroll
-      for (int ks = 0; ks < KSTEPS; ++ks) a_frag[ks] = bf16x8{0};
+        for (int ks = 0; ks < KSTEPS; ++ks) a_frag[mf][ks] = bf16x8{0};
+      }
     }
   }
-  // per-lane thresholds of its 4 C rows
-  float t_reg[4];
+  // per-lane thresholds of its MF x 4 C rows
+  float t_reg[MF][4];
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    const int row = m0 + (lane >> 4) * 4 + r;
-    t_reg[r] = (row < M) ? thresholds[row] : INFINITY;
+  for (int mf = 0; mf < MF; ++mf) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = m0 + mf * 16 + (lane >> 4) * 4 + r;
+      t_reg[mf][r] = (row < M) ? thresholds[row] : INFINITY;
+    }
   }
 
   const int64_t n_tiles = (V + 63) >> 6;
   for (int64_t tile = blockIdx.y; tile < n_tiles; tile += gridDim.y) {
     const int64_t n0 = tile << 6;  // first item of this 64-item tile
-    f32x4 acc[4] = {f32x4{0.f, 0.f, 0.f, 0.f}, f32x4{0.f, 0.f, 0.f, 0.f},
-                    f32x4{0.f, 0.f, 0.f, 0.f}, f32x4{0.f, 0.f, 0.f, 0.f}};
-    // B loads: lane l reads 8 consecutive k of item (n0 + f*16 + (l&15))
+    f32x4 acc[MF][4];
+#pragma unroll
+    for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+      for (int f = 0; f < 4; ++f) acc[mf][f] = f32x4{0.f, 0.f, 0.f, 0.f};
+    // B loads: lane l reads 8 consecutive k of item (n0 + f*16 + (l&15));
+    // each B fragment feeds MF MFMAs (the wide-M lever)
     const int64_t item_base = n0 + (lane & 15);
     const int bk0 = (lane >> 4) * 8;
 #pragma unroll
@@ -84,37 +104,54 @@ __global__ __launch_bounds__(256) void scored_topk_gemm_kernel(
         const __hip_bfloat16* wr = w + (size_t)min(item, V - 1) * E;
         b_frag[f] = *reinterpret_cast<const bf16x8*>(wr + ks * 32 + bk0);
       }
+      if constexpr (!RESIDENT) {
+#pragma unroll
+        for (int mf = 0; mf < MF; ++mf) {
+          a_frag[mf][0] = a_row_valid[mf]
+                              ? *reinterpret_cast<const bf16x8*>(a_rows[mf] + ks * 32)
+                              : bf16x8{0};
+        }
+      }
 #pragma unroll
       for (int f = 0; f < 4; ++f) {
-        acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a_frag[ks], b_frag[f], acc[f], 0, 0, 0);
+#pragma unroll
+        for (int mf = 0; mf < MF; ++mf) {
+          acc[mf][f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mf][RESIDENT ? ks : 0], b_frag[f], acc[mf][f], 0, 0, 0);
+        }
       }
     }
     // ---- epilogue: threshold test on accumulators (common path: no hit) ----
     bool any_hit = false;
 #pragma unroll
-    for (int f = 0; f < 4; ++f) {
-      const int64_t item = n0 + f * 16 + (lane & 15);
-      if (item >= V) continue;
+    for (int mf = 0; mf < MF; ++mf) {
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        any_hit |= (acc[f][r] >= t_reg[r]);
+      for (int f = 0; f < 4; ++f) {
+        const int64_t item = n0 + f * 16 + (lane & 15);
+        if (item >= V) continue;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          any_hit |= (acc[mf][f][r] >= t_reg[mf][r]);
+        }
       }
     }
     if (__builtin_amdgcn_ballot_w64(any_hit) == 0) continue;  // fast path
 #pragma unroll
-    for (int f = 0; f < 4; ++f) {
-      const int64_t item = n0 + f * 16 + (lane & 15);
-      if (item >= V) continue;
+    for (int mf = 0; mf < MF; ++mf) {
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const float v = acc[f][r];
-        if (v >= t_reg[r]) {
-          const int row = m0 + (lane >> 4) * 4 + r;
-          const int pos = atomicAdd(&counts[row], 1);
-          if (pos < cap) {
-            out_vals[(size_t)row * cap + pos] = v;
-            out_idx[(size_t)row * cap + pos] = (int)item;
+      for (int f = 0; f < 4; ++f) {
+        const int64_t item = n0 + f * 16 + (lane & 15);
+        if (item >= V) continue;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const float v = acc[mf][f][r];
+          if (v >= t_reg[mf][r]) {
+            const int row = m0 + mf * 16 + (lane >> 4) * 4 + r;
+            const int pos = atomicAdd(&counts[row], 1);
+            if (pos < cap) {
+              out_vals[(size_t)row * cap + pos] = v;
+              out_idx[(size_t)row * cap + pos] = (int)item;
+            }
           }
         }
       }
@@ -139,13 +176,14 @@ std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
   auto out_idx = torch::zeros({M, capacity}, opts_i);
   auto counts = torch::zeros({M}, opts_i);
   auto thr = thresholds.to(torch::kFloat32).contiguous();
-  const int m_tiles = (M + 63) / 64;
+  const int m_tiles = (M + 255) / 256;
   // fill 256 CUs x ~4 blocks with >> WGs (guide §1); stripes over item tiles
   int stripes = (int)std::min<int64_t>((V + 63) / 64, std::max(1, 4096 / m_tiles));
   dim3 grid(m_tiles, stripes);
   auto stream = at::cuda::getCurrentHIPStream();
 #define LAUNCH_STG(EE)                                                                   \
-  hipLaunchKernelGGL((scored_topk_gemm_kernel<EE>), grid, dim3(256), 0, stream,          \
+  hipLaunchKernelGGL((scored_topk_gemm_kernel<EE, (EE <= 128)>), grid, dim3(256), 0,     \
+                     stream,                                                             \
                      reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),              \
                      reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),              \
                      thr.data_ptr<float>(), out_vals.data_ptr<float>(),                  \
