@@ -33,8 +33,10 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 // RESIDENT: keep the full A (64 rows x E) in VGPRs across tiles (E <= 128);
-// at E = 256 that spills (needs 128 VGPRs for A alone), so A-fragments are
-// re-read per K-step from L2 instead (Q is ~0.5 MB, fully L2-resident).
+// at E = 256 that spills (needs 128 VGPRs for A alone), so the WG's whole
+// 256-row Q tile lives in LDS (128 KiB, XOR-swizzled so the 16-lane
+// fragment-read groups are conflict-free, guide G4) and A-fragments are
+// re-read per K-step at LDS speed.
 template <int E, bool RESIDENT>
 __global__ __launch_bounds__(256, 2) void scored_topk_gemm_kernel(
     const __hip_bfloat16* __restrict__ q,  // [M, E]
@@ -53,20 +55,37 @@ __global__ __launch_bounds__(256, 2) void scored_topk_gemm_kernel(
   // ---- A fragments (64 Q rows): resident across tiles when they fit ----
   constexpr int A_KS = RESIDENT ? KSTEPS : 1;
   bf16x8 a_frag[MF][A_KS];
-  const __hip_bfloat16* a_rows[MF];
-  bool a_row_valid[MF];
+  // non-resident path: the WG's 256xE Q tile in LDS, XOR-swizzled
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  __hip_bfloat16* q_lds = reinterpret_cast<__hip_bfloat16*>(smem);
+  auto lds_off = [&](int row, int k_byte) {
+    return row * (E * 2) + (k_byte ^ ((row & 7) << 4));
+  };
+  if constexpr (!RESIDENT) {
+    // cooperative staged load: thread t copies 16B pieces
+    const int row_q0 = blockIdx.x * 256;
+    for (int i = threadIdx.x; i < 256 * (E * 2 / 16); i += blockDim.x) {
+      const int row = i / (E * 2 / 16);
+      const int k_byte = (i % (E * 2 / 16)) * 16;
+      const int src_row = min(row_q0 + row, M - 1);
+      uint4 v = *reinterpret_cast<const uint4*>(
+          reinterpret_cast<const char*>(q + (size_t)src_row * E) + k_byte);
+      if (row_q0 + row >= M) v = uint4{0, 0, 0, 0};
+      *reinterpret_cast<uint4*>(reinterpret_cast<char*>(q_lds) + lds_off(row, k_byte)) = v;
+    }
+    __syncthreads();
+  }
 #pragma unroll
   for (int mf = 0; mf < MF; ++mf) {
     const int row = m0 + mf * 16 + (lane & 15);
     const int k0 = (lane >> 4) * 8;
-    a_rows[mf] = q + (size_t)min(row, M - 1) * E + k0;
-    a_row_valid[mf] = row < M;
     if constexpr (RESIDENT) {
+      const __hip_bfloat16* qr = q + (size_t)min(row, M - 1) * E + k0;
 #pragma unroll
       for (int ks = 0; ks < KSTEPS; ++ks) {
-        a_frag[mf][ks] = *reinterpret_cast<const bf16x8*>(a_rows[mf] + ks * 32);
+        a_frag[mf][ks] = *reinterpret_cast<const bf16x8*>(qr + ks * 32);
       }
-      if (!a_row_valid[mf]) {
+      if (row >= M) {
 #pragma unroll
         for (int ks = 0; ks < KSTEPS; ++ks) a_frag[mf][ks] = bf16x8{0};
       }
@@ -105,11 +124,14 @@ __global__ __launch_bounds__(256, 2) void scored_topk_gemm_kernel(
         b_frag[f] = *reinterpret_cast<const bf16x8*>(wr + ks * 32 + bk0);
       }
       if constexpr (!RESIDENT) {
+        // LDS fragment read: lane (g = l>>4, r = l&15) reads row (mf*16+r) of
+        // the WG tile at k byte (ks*64 + g*16), XOR-swizzled as written
+        const int lrow_base = wave * 64 + (lane & 15);
+        const int k_byte = ks * 64 + (lane >> 4) * 16;
 #pragma unroll
         for (int mf = 0; mf < MF; ++mf) {
-          a_frag[mf][0] = a_row_valid[mf]
-                              ? *reinterpret_cast<const bf16x8*>(a_rows[mf] + ks * 32)
-                              : bf16x8{0};
+          a_frag[mf][0] = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<const char*>(q_lds) + lds_off(lrow_base + mf * 16, k_byte));
         }
       }
 #pragma unroll
@@ -182,8 +204,8 @@ std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
   dim3 grid(m_tiles, stripes);
   auto stream = at::cuda::getCurrentHIPStream();
 #define LAUNCH_STG(EE)                                                                   \
-  hipLaunchKernelGGL((scored_topk_gemm_kernel<EE, (EE <= 128)>), grid, dim3(256), 0,     \
-                     stream,                                                             \
+  hipLaunchKernelGGL((scored_topk_gemm_kernel<EE, (EE <= 128)>), grid, dim3(256),        \
+                     (EE <= 128) ? 0 : (size_t)256 * EE * 2, stream,                     \
                      reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),              \
                      reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),              \
                      thr.data_ptr<float>(), out_vals.data_ptr<float>(),                  \
