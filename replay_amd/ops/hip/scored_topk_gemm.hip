@@ -103,26 +103,58 @@ __global__ __launch_bounds__(256, 2) void scored_topk_gemm_kernel(
   }
 
   const int64_t n_tiles = (V + 63) >> 6;
-  for (int64_t tile = blockIdx.y; tile < n_tiles; tile += gridDim.y) {
-    const int64_t n0 = tile << 6;  // first item of this 64-item tile
+  const int bk0 = (lane >> 4) * 8;
+  // ---- software-pipelined B stream: depth-4 ring of 4-fragment groups ----
+  // Each "group" g = (tile, ks) is 4 x 16 B loads; groups are prefetched
+  // PIPE_D sections ahead (across tile boundaries) so ~1000 cycles of HBM
+  // latency hide under the 16-MFMA sections (MFMA busy measured 9% without
+  // this, waves parked in s_waitcnt 83%).
+  constexpr int PIPE_D = RESIDENT ? 2 : 4;  // resident-A variants are register-tight
+  static_assert(KSTEPS % PIPE_D == 0 || KSTEPS < PIPE_D, "ring alignment");
+  bf16x8 b_ring[PIPE_D][4];
+  auto load_group = [&](bf16x8 (&dst)[4], int64_t tile, int ks) {
+    const int64_t n0 = tile << 6;
+    const int64_t item_base = n0 + (lane & 15);
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      const int64_t item = item_base + f * 16;
+      const __hip_bfloat16* wr = w + (size_t)min(item, V - 1) * E;
+      dst[f] = *reinterpret_cast<const bf16x8*>(wr + ks * 32 + bk0);
+    }
+  };
+  auto gidx_tile = [&](int64_t tile, int ks, int ahead) {
+    // (tile, ks) advanced by `ahead` K-sections in this WG's walk
+    int nks = ks + ahead;
+    return tile + (int64_t)(nks / KSTEPS) * gridDim.y;
+  };
+  const int64_t tile0 = blockIdx.y;
+  if (tile0 < n_tiles) {
+#pragma unroll
+    for (int d = 0; d < PIPE_D && d < KSTEPS; ++d) {
+      load_group(b_ring[d], gidx_tile(tile0, 0, d), d % KSTEPS);
+    }
+    if constexpr (KSTEPS < PIPE_D) {
+#pragma unroll
+      for (int d = KSTEPS; d < PIPE_D; ++d) {
+        load_group(b_ring[d], gidx_tile(tile0, 0, d), d % KSTEPS);
+      }
+    }
+  }
+  for (int64_t tile = tile0; tile < n_tiles; tile += gridDim.y) {
     f32x4 acc[MF][4];
 #pragma unroll
     for (int mf = 0; mf < MF; ++mf)
 #pragma unroll
       for (int f = 0; f < 4; ++f) acc[mf][f] = f32x4{0.f, 0.f, 0.f, 0.f};
-    // B loads: lane l reads 8 consecutive k of item (n0 + f*16 + (l&15));
-    // each B fragment feeds MF MFMAs (the wide-M lever)
-    const int64_t item_base = n0 + (lane & 15);
-    const int bk0 = (lane >> 4) * 8;
+    const int64_t n0 = tile << 6;
 #pragma unroll
     for (int ks = 0; ks < KSTEPS; ++ks) {
+          const int slot = ks % PIPE_D;  // compile-time after full unroll
       bf16x8 b_frag[4];
 #pragma unroll
-      for (int f = 0; f < 4; ++f) {
-        const int64_t item = item_base + f * 16;
-        const __hip_bfloat16* wr = w + (size_t)min(item, V - 1) * E;
-        b_frag[f] = *reinterpret_cast<const bf16x8*>(wr + ks * 32 + bk0);
-      }
+      for (int f = 0; f < 4; ++f) b_frag[f] = b_ring[slot][f];
+      // prefetch the group PIPE_D sections ahead into the freed slot
+      load_group(b_ring[slot], gidx_tile(tile, ks, PIPE_D), (ks + PIPE_D) % KSTEPS);
       if constexpr (!RESIDENT) {
         // LDS fragment read: lane (g = l>>4, r = l&15) reads row (mf*16+r) of
         // the WG tile at k byte (ks*64 + g*16), XOR-swizzled as written
