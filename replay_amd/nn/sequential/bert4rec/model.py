@@ -90,20 +90,26 @@ class Bert4Rec(torch.nn.Module):
         batch: Dict[str, torch.Tensor],
         candidates_to_score: Optional[torch.Tensor] = None,
     ) -> torch.Tensor:
-        """Inference: append-a-mask convention — the last valid position's id
-        is replaced by the mask token and its hidden state scores the catalog
-        (reference bert4rec predict flow)."""
-        from replay_amd.nn.utils import last_valid_index
-
+        """Inference: append-a-mask convention (reference bert4rec predict
+        flow) — the sequence is left-aligned, shifted by one, and a [MASK]
+        token appended; its hidden state scores the catalog (predicting the
+        NEXT item, not reconstructing the last)."""
         padding_mask = batch["padding_mask"]
         feats = dict(self._features_of(batch))
-        items = feats[self.item_feature_name].clone()
-        last_idx = last_valid_index(padding_mask)
-        rows = torch.arange(items.shape[0], device=items.device)
-        items[rows, last_idx] = self.mask_token
-        feats[self.item_feature_name] = items
-        hidden = self.body(feats, padding_mask)
-        last = hidden[rows, last_idx]
+        items = feats[self.item_feature_name]
+        B, L = items.shape
+        # left-align (pads first), stable so the event order is kept
+        perm = torch.argsort(padding_mask.long(), dim=1, stable=True)
+        items_l = items.gather(1, perm)
+        mask_l = padding_mask.gather(1, perm)
+        mask_col = torch.full((B, 1), self.mask_token, dtype=items.dtype, device=items.device)
+        items2 = torch.cat([items_l[:, 1:], mask_col], dim=1)
+        mask2 = torch.cat(
+            [mask_l[:, 1:], torch.ones(B, 1, dtype=torch.bool, device=items.device)], dim=1
+        )
+        feats[self.item_feature_name] = items2
+        hidden = self.body(feats, mask2)
+        last = hidden[:, -1]
         return self.head(last, candidates_to_score)
 
     predict = forward_inference
