@@ -66,6 +66,8 @@ def test_bert4rec_learns_cyclic_pattern():
     masker = TokenMaskTransform(mask_prob=0.3, generator_seed=0)
     for _ in range(200):
         batch = masker(_cyclic_batch(rng))
+        batch.pop("labels")  # BERT objective: reconstruct the masked item
+        batch.pop("labels_padding_mask")
         loss = model(batch)
         opt.zero_grad()
         loss.backward()
