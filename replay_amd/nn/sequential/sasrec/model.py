@@ -49,6 +49,10 @@ class SasRecBody(torch.nn.Module):
     def forward(self, feature_tensors: Dict[str, torch.Tensor], padding_mask: torch.Tensor) -> torch.Tensor:
         embeddings = self.embedder(feature_tensors)
         x = self.aggregator(embeddings, padding_mask)
+        if x.is_cuda and torch.is_autocast_enabled("cuda"):
+            # keep the residual stream in the autocast dtype: halves LN /
+            # residual-add traffic and removes per-block bf16<->fp32 casts
+            x = x.to(torch.get_autocast_dtype("cuda"))
         attn_mask = self.attention_mask(padding_mask)
         hidden = self.encoder(x, attn_mask=attn_mask, padding_mask=padding_mask)
         return self.output_normalization(hidden)
