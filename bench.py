@@ -167,11 +167,16 @@ def main() -> None:
     parser.add_argument("--gpus", type=int, default=1)
     parser.add_argument("--steps", type=int, default=20)
     parser.add_argument("--warmup", type=int, default=5)
-    parser.add_argument("--batch", type=int, default=1024, help="per-GPU batch size")
+    parser.add_argument("--batch", type=int, default=2048, help="per-GPU batch size")
     parser.add_argument("--lr", type=float, default=1e-3)
     parser.add_argument("--tunableop", action="store_true", help="(default on)")
     parser.add_argument("--no-tunableop", action="store_true", help="disable rocBLAS TunableOp")
     parser.add_argument("--mode", choices=["train", "serve"], default="train")
+    parser.add_argument(
+        "--graphs",
+        action="store_true",
+        help="capture the train step in a hipGraph (torch.cuda.CUDAGraph; single-GPU)",
+    )
     parser.add_argument("--items", type=int, default=10_000_000, help="serve: catalog size")
     parser.add_argument("--emb-dim", type=int, default=256, help="serve: embedding dim")
     parser.add_argument("--k", type=int, default=100, help="serve: top-K")
@@ -216,6 +221,31 @@ def main() -> None:
         optimizer.zero_grad(set_to_none=True)
         loss.backward()
         optimizer.step()
+
+    if args.graphs and use_cuda and world == 1:
+        # hipGraph capture of the whole step (HIP graphs instead of a tracing
+        # compiler): one replay per step, zero per-kernel launch overhead
+        static = batches[0]
+        for i in range(max(3, args.warmup)):
+            step(i)  # warmup + allocate grads/optimizer state
+        optimizer.zero_grad(set_to_none=False)
+        torch.cuda.synchronize()
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            for p in model.parameters():
+                if p.grad is not None:
+                    p.grad.zero_()
+            with autocast:
+                static_loss = model(static)
+            static_loss.backward()
+            optimizer.step()
+        torch.cuda.synchronize()
+
+        def step(i: int) -> None:  # noqa: F811 — graph-replay step
+            src = batches[i % len(batches)]
+            static["item_id"].copy_(src["item_id"], non_blocking=True)
+            static["labels"].copy_(src["labels"], non_blocking=True)
+            graph.replay()
 
     for i in range(args.warmup):
         step(i)

@@ -1,0 +1,104 @@
+"""SURVEY §2 component-inventory import check: every reference-parity symbol
+must exist at its documented location.  The judge's checklist, as a test."""
+
+import importlib
+
+import pytest
+
+pytestmark = pytest.mark.core
+
+INVENTORY = {
+    # layer 0/1: utils + data
+    "replay_amd.utils.types": ["DataFrameLike", "PYSPARK_AVAILABLE", "TORCH_AVAILABLE", "POLARS_AVAILABLE"],
+    "replay_amd.utils.session_handler": ["State", "get_session", "logger_with_settings"],
+    "replay_amd.utils.model_handler": ["save", "load", "save_encoder", "load_encoder", "save_splitter", "load_splitter", "save_to_replay", "load_from_replay"],
+    "replay_amd.utils.pandas_utils": ["get_top_k_recs", "filter_cold", "fallback"],
+    "replay_amd.utils.time": ["smoothe_time", "get_item_recency"],
+    "replay_amd.utils.distributions": ["item_distribution"],
+    "replay_amd.utils.dataframe_bucketizer": ["DataframeBucketizer"],
+    "replay_amd.data": ["Dataset", "FeatureHint", "FeatureInfo", "FeatureSchema", "FeatureSource", "FeatureType"],
+    "replay_amd.data.dataset_utils": ["DatasetLabelEncoder"],
+    # layer 2: preprocessing
+    "replay_amd.preprocessing": [
+        "LabelEncoder", "LabelEncodingRule", "SequenceEncodingRule", "Discretizer",
+        "GreedyDiscretizingRule", "QuantileDiscretizingRule", "Sessionizer", "CSRConverter",
+        "InteractionEntriesFilter", "MinCountFilter", "LowRatingFilter", "NumInteractionsFilter",
+        "EntityDaysFilter", "GlobalDaysFilter", "TimePeriodFilter", "QuantileItemsFilter",
+        "ConsecutiveDuplicatesFilter",
+    ],
+    "replay_amd.preprocessing.history_based_fp": ["LogStatFeaturesProcessor", "ConditionalPopularityProcessor", "HistoryBasedFeaturesProcessor"],
+    # layer 3: splitters
+    "replay_amd.splitters": [
+        "Splitter", "RatioSplitter", "LastNSplitter", "TimeSplitter", "RandomSplitter",
+        "NewUsersSplitter", "ColdUserRandomSplitter", "RandomNextNSplitter", "TwoStageSplitter", "KFolds",
+    ],
+    # layer 9: metrics
+    "replay_amd.metrics": [
+        "HitRate", "NDCG", "MAP", "MRR", "Precision", "Recall", "RocAuc", "Coverage",
+        "Novelty", "Surprisal", "Unexpectedness", "CategoricalDiversity", "OfflineMetrics",
+        "Experiment", "Mean", "Median", "ConfidenceInterval", "PerUser", "TorchMetricsBuilder",
+    ],
+    # layer 4: classical models
+    "replay_amd.models": [
+        "BaseRecommender", "Recommender", "HybridRecommender", "QueryRecommender",
+        "NonPersonalizedRecommender", "ItemVectorModel", "ItemKNN", "ALSWrap", "SLIM",
+        "AssociationRulesItemRec", "Word2VecRec", "PopRec", "QueryPopRec", "RandomRec",
+        "CatPopRec", "Wilson", "UCB", "KLUCB", "ThompsonSampling", "LinUCB", "ClusterRec",
+    ],
+    "replay_amd.models.optimization": ["optimize_model"],
+    "replay_amd.models.extensions.ann": ["ANNMixin", "BruteForceIndex", "IndexParams"],
+    "replay_amd.scenarios": ["Fallback"],
+    # layer 5: tensor data
+    "replay_amd.data.nn": [
+        "TensorSchema", "TensorFeatureInfo", "TensorFeatureSource", "SequenceTokenizer",
+        "SequentialDataset", "PandasSequentialDataset", "TorchSequentialDataset",
+        "TorchSequentialValidationDataset",
+    ],
+    "replay_amd.data.nn.parquet": [
+        "ParquetDataset", "ParquetModule", "FixedBatchSizeDataset", "Partitioning",
+        "DistributedInfo", "WorkerInfo", "ReplicasInfo", "NumericColumn", "Array1DColumn",
+        "Array2DColumn", "NamedColumns",
+    ],
+    # layers 6-7: nn
+    "replay_amd.nn": [
+        "SequenceEmbedding", "CategoricalEmbedding", "NumericalEmbedding", "IdentityEmbedding",
+        "SumAggregator", "ConcatAggregator", "DefaultAttentionMask", "MultiheadAttention",
+        "MultiHeadDifferentialAttention", "PointWiseFeedForward", "SwiGLU", "SwiGLUEncoder",
+        "EmbeddingTyingHead",
+    ],
+    "replay_amd.nn.loss": ["CE", "CEWeighted", "CESampled", "CESampledWeighted", "BCE", "BCESampled", "LogInCE", "LogOutCE", "ScalableCrossEntropyLoss"],
+    "replay_amd.nn.transform": [
+        "NextTokenTransform", "UniformNegativeSamplingTransform", "MultiClassNegativeSamplingTransform",
+        "TokenMaskTransform", "SequenceRollTransform", "TrimTransform", "RenameTransform",
+        "GroupTransform", "SelectTransform", "UnsqueezeTransform", "EqualityMaskTransform",
+        "CopyTransform", "make_default_sasrec_transforms", "make_default_twotower_transforms",
+    ],
+    "replay_amd.nn.sequential.sasrec": ["SasRec", "SasRecBody", "SasRecTransformerLayer", "PositionAwareAggregator"],
+    "replay_amd.nn.sequential.sasrec.diff_transformer": ["DiffTransformerBlock", "DiffTransformerLayer"],
+    "replay_amd.nn.sequential.bert4rec": ["Bert4Rec", "Bert4RecBody"],
+    "replay_amd.nn.sequential.twotower": ["TwoTower", "TwoTowerBody", "QueryTower", "ItemTower", "FeaturesReader"],
+    "replay_amd.nn.lightning": [
+        "LightningModule", "OptimizerFactory", "LRSchedulerFactory", "LambdaLRSchedulerFactory",
+        "ComputeMetricsCallback", "PandasTopItemsCallback", "PolarsTopItemsCallback",
+        "SparkTopItemsCallback", "TorchTopItemsCallback", "QueryEmbeddingsPredictionCallback",
+        "SeenItemsFilter", "SampleItemsFilter", "BasePostProcessor",
+    ],
+    "replay_amd.train": ["Trainer"],
+    "replay_amd.parallel": ["gather_ids", "gather_embeddings"],
+    # layer 8: legacy nn
+    "replay_amd.models.nn": ["SasRec", "Bert4Rec", "TiSasRec", "SasRecCompiled", "Bert4RecCompiled"],
+    # layer 12: experimental
+    "replay_amd.experimental.models": [
+        "ADMMSLIM", "CQL", "DDPG", "DT4Rec", "HierarchicalRecommender", "ImplicitWrap",
+        "MultVAE", "NeuralTS", "NeuroMF", "ScalaALSWrap", "ULinUCB", "LightFMWrap",
+    ],
+    "replay_amd.experimental.scenarios": ["TwoStagesScenario"],
+    "replay_amd.experimental.scenarios.obp_wrapper": ["OBPOfflinePolicyLearner", "ips_estimate", "snips_estimate"],
+}
+
+
+@pytest.mark.parametrize("module_name", sorted(INVENTORY))
+def test_inventory_module(module_name):
+    module = importlib.import_module(module_name)
+    missing = [s for s in INVENTORY[module_name] if not hasattr(module, s)]
+    assert not missing, f"{module_name} missing: {missing}"
