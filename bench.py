@@ -208,7 +208,9 @@ def main() -> None:
             bucket_cap_mb=64,
             gradient_as_bucket_view=True,
         )
-    optimizer = torch.optim.Adam(model.parameters(), lr=args.lr)
+    optimizer = torch.optim.Adam(
+        model.parameters(), lr=args.lr, capturable=args.graphs and use_cuda and world == 1
+    )
     batches = make_batches(4, args.batch, device, seed=1000 + rank)
 
     amp_dtype = torch.bfloat16
