@@ -48,7 +48,7 @@ class CE(LossBase):
                     hidden2d = hidden2d.to(torch.get_autocast_dtype("cuda"))
                 weight = head.get_item_weights()
                 n_elems = hidden2d.shape[0] * weight.shape[0]
-                if n_elems * hidden2d.element_size() > 8 * 2**30:
+                if n_elems * hidden2d.element_size() > 24 * 2**30:
                     # huge catalogs (e.g. 10M items): chunked CE, logits are
                     # recomputed in backward and never fully materialized
                     from replay_amd.ops.fused_ce import chunked_fused_ce
