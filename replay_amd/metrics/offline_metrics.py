@@ -41,6 +41,12 @@ class OfflineMetrics:
         self.query_column = query_column
         self.item_column = item_column
         self.rating_column = rating_column
+        # OfflineMetrics owns the column naming (reference offline_metrics.py)
+        for metric in metrics:
+            metric.query_column = query_column
+            metric.rating_column = rating_column
+            if not isinstance(metric, CategoricalDiversity):
+                metric.item_column = item_column
 
     def __call__(
         self,

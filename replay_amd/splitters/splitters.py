@@ -28,7 +28,7 @@ class RatioSplitter(Splitter):
     def __init__(
         self,
         test_size: float = 0.2,
-        divide_column: str = "query_id",
+        divide_column: Optional[str] = None,
         min_interactions_per_group: Optional[int] = None,
         split_by_fractions: bool = True,
         **kwargs,
@@ -37,7 +37,7 @@ class RatioSplitter(Splitter):
         if not 0 < test_size < 1:
             raise ValueError("test_size must be in (0, 1)")
         self.test_size = test_size
-        self.divide_column = divide_column
+        self.divide_column = divide_column or self.query_column
         self.min_interactions_per_group = min_interactions_per_group
         self.split_by_fractions = split_by_fractions
 
@@ -65,7 +65,7 @@ class LastNSplitter(Splitter):
     def __init__(
         self,
         N: int = 1,
-        divide_column: str = "query_id",
+        divide_column: Optional[str] = None,
         strategy: str = "interactions",
         **kwargs,
     ) -> None:
@@ -73,7 +73,7 @@ class LastNSplitter(Splitter):
         if strategy not in ("interactions", "timedelta"):
             raise ValueError("strategy must be 'interactions' or 'timedelta'")
         self.N = N
-        self.divide_column = divide_column
+        self.divide_column = divide_column or self.query_column
         self.strategy = strategy
 
     def _core_split(self, interactions: pd.DataFrame) -> SplitterReturnType:
@@ -186,11 +186,11 @@ class RandomNextNSplitter(Splitter):
 
     _init_arg_names = Splitter._init_arg_names + ("N", "seed", "divide_column")
 
-    def __init__(self, N: int = 1, seed: Optional[int] = None, divide_column: str = "query_id", **kwargs) -> None:
+    def __init__(self, N: int = 1, seed: Optional[int] = None, divide_column: Optional[str] = None, **kwargs) -> None:
         super().__init__(**kwargs)
         self.N = N
         self.seed = seed
-        self.divide_column = divide_column
+        self.divide_column = divide_column or self.query_column
 
     def _core_split(self, interactions: pd.DataFrame) -> SplitterReturnType:
         df = interactions.sort_values([self.divide_column, self.timestamp_column], kind="stable")
