@@ -212,6 +212,9 @@ class Trainer:
             if val_dataloaders is not None:
                 self._run_eval(module, val_dataloaders, stage="validate")
             self._reduce_sync_metrics()
+            # Lightning convention: current_epoch counts COMPLETED epochs, so
+            # a checkpoint saved after epoch e resumes at e+1
+            self.current_epoch = epoch + 1
             if stop:
                 break
         self._call_callbacks("on_fit_end")
