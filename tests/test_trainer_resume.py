@@ -52,8 +52,8 @@ def test_resume_from_checkpoint(tmp_path):
     module2 = _module()
     trainer2 = Trainer(max_epochs=2, accelerator="cpu", precision="32")
     trainer2.fit(module2, loader, ckpt_path=str(ckpt))
-    assert trainer2.current_epoch == 1
-    assert trainer2.global_step == 2 * len(loader)
+    assert trainer2.current_epoch == 2  # completed epochs (Lightning convention)
+    assert trainer2.global_step == 2 * len(loader)  # 6 before resume + 6 after
     # optimizer momentum actually restored (exp_avg nonzero from epoch 1)
     state = trainer2._optimizer.state_dict()["state"]
     assert len(state) > 0
