@@ -114,10 +114,21 @@ def serve_bench(args, device, rank, world) -> None:
     queries = torch.randn(B, emb_dim, device=device, dtype=dtype)
     seen = torch.randint(0, n_items, (B, 64), device=device)
 
-    def step():
-        if world > 1:
-            return sharded_catalog_topk(queries, items_shard, args.k, lo, seen)
-        return catalog_topk(queries, items_shard, args.k, seen)
+    if args.fp8 and device.type == "cuda":
+        from replay_amd.ops.topk import catalog_topk_fp8, quantize_fp8
+
+        w8, sw = quantize_fp8(items_shard)
+        q8, sq = quantize_fp8(queries)
+
+        def step():
+            return catalog_topk_fp8(q8, sq, w8, sw, args.k, seen)
+
+    else:
+
+        def step():
+            if world > 1:
+                return sharded_catalog_topk(queries, items_shard, args.k, lo, seen)
+            return catalog_topk(queries, items_shard, args.k, seen)
 
     for _ in range(args.warmup):
         step()
@@ -152,7 +163,7 @@ def serve_bench(args, device, rank, world) -> None:
                     "higher_is_better": True,
                     "scaling": "strong" if world > 1 else "weak",
                     "vs_baseline": None,
-                    "dtype": "bf16" if device.type == "cuda" else "fp32",
+                    "dtype": ("fp8-e4m3" if args.fp8 else "bf16") if device.type == "cuda" else "fp32",
                     "data": "synthetic",
                     "config": {
                         "model": f"catalog_topk_d{emb_dim}",
@@ -184,6 +195,7 @@ def main() -> None:
     parser.add_argument("--items", type=int, default=10_000_000, help="serve: catalog size")
     parser.add_argument("--emb-dim", type=int, default=256, help="serve: embedding dim")
     parser.add_argument("--k", type=int, default=100, help="serve: top-K")
+    parser.add_argument("--fp8", action="store_true", help="serve: fp8 (e4m3) score GEMM")
     args = parser.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))

@@ -232,3 +232,53 @@ def sharded_catalog_topk(
     all_i = torch.cat(gathered_i, dim=1)
     sel_s, sel_pos = torch.topk(all_s, k, dim=1)
     return sel_s, all_i.gather(1, sel_pos)
+
+
+# ---------------------------------------------------------------------------
+# fp8 scoring (BASELINE config 5: "fp8 MFMA score GEMM")
+# ---------------------------------------------------------------------------
+def quantize_fp8(t: torch.Tensor, rowwise: bool = False):
+    """Quantize to OCP e4m3 (gfx950's native fp8 — NOT the MI300X fnuz
+    variant) with an amax-based scale; returns (fp8_tensor, scale_f32)."""
+    finfo = torch.finfo(torch.float8_e4m3fn)
+    if rowwise:
+        amax = t.float().abs().amax(dim=1, keepdim=True).clamp(min=1e-12)
+    else:
+        amax = t.float().abs().amax().clamp(min=1e-12)
+    scale = amax / finfo.max
+    q = (t.float() / scale).clamp(finfo.min, finfo.max).to(torch.float8_e4m3fn)
+    return q, scale.to(torch.float32)
+
+
+def catalog_topk_fp8(
+    q8: torch.Tensor,  # [B, E] float8_e4m3fn
+    scale_q: torch.Tensor,
+    w8: torch.Tensor,  # [V, E] float8_e4m3fn
+    scale_w: torch.Tensor,
+    k: int,
+    seen: Optional[torch.Tensor] = None,
+    chunk_items: int = 2**21,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Full-catalog top-K with the score GEMM on the fp8 MFMA pipes
+    (hipBLASLt _scaled_mm; e4m3 inputs, bf16 out).  Selection via
+    fast_row_topk; ranking is by fp8-precision scores (the config-5
+    contract — half the item-table bytes of bf16)."""
+    V = w8.shape[0]
+    k = min(k, V)
+    run_scores = run_ids = None
+    for lo in range(0, V, chunk_items):
+        hi = min(lo + chunk_items, V)
+        scores = torch._scaled_mm(
+            q8, w8[lo:hi].t(), scale_a=scale_q, scale_b=scale_w, out_dtype=torch.bfloat16
+        )
+        kk = min(k, hi - lo)
+        top_s, top_i = fast_row_topk(scores, kk, seen=seen, col_offset=lo)
+        top_i = top_i + lo
+        if run_scores is None:
+            run_scores, run_ids = top_s, top_i
+        else:
+            merged_s = torch.cat([run_scores, top_s], dim=1)
+            merged_i = torch.cat([run_ids, top_i], dim=1)
+            sel_s, sel_pos = torch.topk(merged_s, min(k, merged_s.shape[1]), dim=1)
+            run_scores, run_ids = sel_s, merged_i.gather(1, sel_pos)
+    return run_scores, run_ids

@@ -319,3 +319,23 @@ class TestModelOnGPU:
         out = eager_attention(q, k, v, mask)
         ref = torch.nn.functional.scaled_dot_product_attention(q, k, v, attn_mask=mask)
         torch.testing.assert_close(out, ref, atol=1e-4, rtol=1e-4)
+
+
+@requires_gpu
+class TestFp8Scoring:
+    def test_fp8_topk_close_to_bf16(self):
+        from replay_amd.ops.topk import catalog_topk, catalog_topk_fp8, quantize_fp8
+
+        torch.manual_seed(0)
+        B, E, V, K = 64, 256, 200_000, 100
+        q = torch.randn(B, E, device="cuda", dtype=torch.bfloat16)
+        w = torch.randn(V, E, device="cuda", dtype=torch.bfloat16)
+        q8, sq = quantize_fp8(q)
+        w8, sw = quantize_fp8(w)
+        s8, i8 = catalog_topk_fp8(q8, sq, w8, sw, K)
+        s16, i16 = catalog_topk(q, w, K, _allow_fused=False)
+        # fp8 quantization reorders near-ties; demand strong top-K overlap
+        overlaps = [len(set(i8[b].tolist()) & set(i16[b].tolist())) / K for b in range(B)]
+        assert sum(overlaps) / B > 0.85, f"mean overlap {sum(overlaps)/B:.2f}"
+        # scores agree to fp8 precision
+        torch.testing.assert_close(s8.float().mean(), s16.float().mean(), atol=0.5, rtol=0.05)
