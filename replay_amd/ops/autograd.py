@@ -60,7 +60,15 @@ class FlashAttentionFunction(torch.autograd.Function):
         ext = hip_ext()
         scale = 1.0 / (q.shape[-1] ** 0.5)
         need_lse = q.requires_grad or k.requires_grad or v.requires_grad
-        out, lse = ext.attention_fwd(q, k, v, padding_mask, scale, causal, need_lse)
+        if (
+            q.dtype == torch.bfloat16
+            and q.shape[-1] in (32, 64)
+            and q.shape[2] <= 256
+            and hasattr(ext, "attention_fwd_mfma")
+        ):
+            out, lse = ext.attention_fwd_mfma(q, k, v, padding_mask, scale, causal, need_lse)
+        else:
+            out, lse = ext.attention_fwd(q, k, v, padding_mask, scale, causal, need_lse)
         ctx.save_for_backward(q, k, v, out, lse, padding_mask)
         ctx.causal = causal
         ctx.scale = scale

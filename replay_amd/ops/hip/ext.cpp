@@ -25,6 +25,9 @@ std::vector<torch::Tensor> threshold_compact(torch::Tensor scores, torch::Tensor
                                              int64_t col_offset);
 std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
                                             torch::Tensor thresholds, int64_t capacity);
+std::vector<torch::Tensor> attention_fwd_mfma(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                              c10::optional<torch::Tensor> valid, double scale,
+                                              bool causal, bool need_lse);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layer_norm_fwd", &layer_norm_fwd, "fused LayerNorm forward (gfx950)");
@@ -36,4 +39,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("threshold_compact", &threshold_compact, "top-k threshold compaction (gfx950)");
   m.def("scored_topk_gemm", &scored_topk_gemm,
         "fused MFMA score-GEMM + top-k candidate selection (gfx950)");
+  m.def("attention_fwd_mfma", &attention_fwd_mfma, "MFMA attention forward (gfx950)");
 }
