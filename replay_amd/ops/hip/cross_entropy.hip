@@ -76,7 +76,28 @@ __global__ void ce_fwd_kernel(const T* __restrict__ logits,
         if (x > m) { s *= __expf(m - x); m = x; }
         s += __expf(x - m);
       }
-      for (int64_t c = head + lane * 8; c + 7 < n_vec; c += WAVE * 8) {
+      int64_t c = head + lane * 8;
+      // 2x unrolled with both loads issued before the transcendental work:
+      // the single-chunk loop measured 84.7% SQ_WAIT_ANY (load-latency bound)
+      for (; c + WAVE * 8 + 7 < n_vec; c += WAVE * 16) {
+        float v2[8];
+        load8_bf16(reinterpret_cast<const __hip_bfloat16*>(lr) + c, v);
+        load8_bf16(reinterpret_cast<const __hip_bfloat16*>(lr) + c + WAVE * 8, v2);
+        float m8 = v[0];
+#pragma unroll
+        for (int i = 1; i < 8; ++i) m8 = fmaxf(m8, v[i]);
+#pragma unroll
+        for (int i = 0; i < 8; ++i) m8 = fmaxf(m8, v2[i]);
+        if (m8 > m) {
+          s *= __expf(m - m8);
+          m = m8;
+        }
+#pragma unroll
+        for (int i = 0; i < 8; ++i) s += __expf(v[i] - m);
+#pragma unroll
+        for (int i = 0; i < 8; ++i) s += __expf(v2[i] - m);
+      }
+      for (; c + 7 < n_vec; c += WAVE * 8) {
         load8_bf16(reinterpret_cast<const __hip_bfloat16*>(lr) + c, v);
         float m8 = v[0];
 #pragma unroll
@@ -150,7 +171,30 @@ __global__ void ce_bwd_kernel(const T* __restrict__ logits,
           dr[c] = from_f32<T>((p - (c == label ? 1.f : 0.f)) * scale);
         }
       }
-      for (int64_t c = head + lane * 8; c + 7 < n_vec; c += WAVE * 8) {
+      int64_t c = head + lane * 8;
+      for (; c + WAVE * 8 + 7 < n_vec; c += WAVE * 16) {
+        float v2[8];
+        if (ignored) {
+#pragma unroll
+          for (int i = 0; i < 8; ++i) v[i] = v2[i] = 0.f;
+        } else {
+          load8_bf16(reinterpret_cast<const __hip_bfloat16*>(lr) + c, v);
+          load8_bf16(reinterpret_cast<const __hip_bfloat16*>(lr) + c + WAVE * 8, v2);
+#pragma unroll
+          for (int i = 0; i < 8; ++i) {
+            float p = __expf(v[i] - l);
+            v[i] = (p - ((c + i) == label ? 1.f : 0.f)) * scale;
+          }
+#pragma unroll
+          for (int i = 0; i < 8; ++i) {
+            float p = __expf(v2[i] - l);
+            v2[i] = (p - ((c + WAVE * 8 + i) == label ? 1.f : 0.f)) * scale;
+          }
+        }
+        store8_bf16(reinterpret_cast<__hip_bfloat16*>(dr) + c, v);
+        store8_bf16(reinterpret_cast<__hip_bfloat16*>(dr) + c + WAVE * 8, v2);
+      }
+      for (; c + 7 < n_vec; c += WAVE * 8) {
         if (ignored) {
 #pragma unroll
           for (int i = 0; i < 8; ++i) v[i] = 0.f;
