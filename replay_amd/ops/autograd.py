@@ -78,7 +78,17 @@ class FlashAttentionFunction(torch.autograd.Function):
     def backward(ctx, dout):
         ext = hip_ext()
         q, k, v, out, lse, padding_mask = ctx.saved_tensors
-        dq, dk, dv = ext.attention_bwd(
-            q, k, v, out, dout, lse, padding_mask, ctx.scale, ctx.causal
-        )
+        if (
+            q.dtype == torch.bfloat16
+            and q.shape[-1] in (32, 64)
+            and q.shape[2] <= 256
+            and hasattr(ext, "attention_bwd_mfma")
+        ):
+            dq, dk, dv = ext.attention_bwd_mfma(
+                q, k, v, out, dout, lse, padding_mask, ctx.scale, ctx.causal
+            )
+        else:
+            dq, dk, dv = ext.attention_bwd(
+                q, k, v, out, dout, lse, padding_mask, ctx.scale, ctx.causal
+            )
         return dq, dk, dv, None, None

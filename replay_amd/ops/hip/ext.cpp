@@ -28,6 +28,11 @@ std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
 std::vector<torch::Tensor> attention_fwd_mfma(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                                               c10::optional<torch::Tensor> valid, double scale,
                                               bool causal, bool need_lse);
+std::vector<torch::Tensor> attention_bwd_mfma(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                              torch::Tensor out, torch::Tensor dout,
+                                              torch::Tensor lse,
+                                              c10::optional<torch::Tensor> valid, double scale,
+                                              bool causal);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layer_norm_fwd", &layer_norm_fwd, "fused LayerNorm forward (gfx950)");
@@ -40,4 +45,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("scored_topk_gemm", &scored_topk_gemm,
         "fused MFMA score-GEMM + top-k candidate selection (gfx950)");
   m.def("attention_fwd_mfma", &attention_fwd_mfma, "MFMA attention forward (gfx950)");
+  m.def("attention_bwd_mfma", &attention_bwd_mfma, "MFMA attention backward (gfx950)");
 }

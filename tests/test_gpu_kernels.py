@@ -395,3 +395,33 @@ class TestMfmaAttention:
         ref.sum().backward()
         torch.testing.assert_close(q.grad.float(), q2.grad, atol=6e-2, rtol=6e-2)
         torch.testing.assert_close(v.grad.float(), v2.grad, atol=6e-2, rtol=6e-2)
+
+
+@requires_gpu
+class TestMfmaAttentionBwd:
+    @pytest.mark.parametrize("causal", [True, False])
+    @pytest.mark.parametrize("shape", [(3, 2, 50, 32), (2, 2, 200, 64), (2, 1, 33, 32)])
+    def test_bwd_matches_fp32_reference(self, causal, shape):
+        from replay_amd.ops import hip_ext
+
+        ext = hip_ext()
+        torch.manual_seed(3)
+        B, H, L, D = shape
+        q = torch.randn(B, H, L, D, device="cuda", dtype=torch.bfloat16)
+        k = torch.randn(B, H, L, D, device="cuda", dtype=torch.bfloat16)
+        v = torch.randn(B, H, L, D, device="cuda", dtype=torch.bfloat16)
+        mask = torch.rand(B, L, device="cuda") > 0.2
+        mask[:, 0] = True
+        scale = 1.0 / D**0.5
+        out, lse = ext.attention_fwd_mfma(q, k, v, mask, scale, causal, True)
+        dout = torch.randn_like(out)
+        dq, dk, dv = ext.attention_bwd_mfma(q, k, v, out, dout, lse, mask, scale, causal)
+
+        q2 = q.detach().float().clone().requires_grad_(True)
+        k2 = k.detach().float().clone().requires_grad_(True)
+        v2 = v.detach().float().clone().requires_grad_(True)
+        ref = TestFlashAttention._eager_ref(q2, k2, v2, mask, causal)
+        ref.backward(dout.float())
+        torch.testing.assert_close(dq.float(), q2.grad, atol=8e-2, rtol=8e-2)
+        torch.testing.assert_close(dk.float(), k2.grad, atol=8e-2, rtol=8e-2)
+        torch.testing.assert_close(dv.float(), v2.grad, atol=8e-2, rtol=8e-2)
