@@ -130,8 +130,8 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_mfma_kernel(
     for (int f = 0; f < OF; ++f) dq_acc[f] = f32x4{0.f, 0.f, 0.f, 0.f};
 
     const int kt_end = causal ? ((q0 + 15) >> 5) + 1 : n_t32;
-    for (int kt = 0; kt < kt_end; ++kt) {
-      const int k0 = kt << 5;
+    for (int ktile = 0; ktile < kt_end; ++ktile) {  // NOT "kt": shadows the LDS stage ptr
+      const int k0 = ktile << 5;
       f32x4 s_acc[2] = {f32x4{0.f, 0.f, 0.f, 0.f}, f32x4{0.f, 0.f, 0.f, 0.f}};
       f32x4 dp_acc[2] = {f32x4{0.f, 0.f, 0.f, 0.f}, f32x4{0.f, 0.f, 0.f, 0.f}};
 #pragma unroll
