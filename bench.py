@@ -182,7 +182,7 @@ def main() -> None:
     parser.add_argument("--gpus", type=int, default=1)
     parser.add_argument("--steps", type=int, default=20)
     parser.add_argument("--warmup", type=int, default=5)
-    parser.add_argument("--batch", type=int, default=4096, help="per-GPU batch size")
+    parser.add_argument("--batch", type=int, default=8192, help="per-GPU batch size")
     parser.add_argument("--lr", type=float, default=1e-3)
     parser.add_argument("--tunableop", action="store_true", help="(default on)")
     parser.add_argument("--no-tunableop", action="store_true", help="disable rocBLAS TunableOp")
