@@ -54,12 +54,23 @@ class CE(LossBase):
                     from replay_amd.ops.fused_ce import chunked_fused_ce
 
                     return chunked_fused_ce(hidden2d, weight, labels.reshape(-1), -100)
+                weight = weight.to(hidden2d.dtype)
+                if (
+                    hidden2d.dtype == torch.bfloat16
+                    and hidden2d.shape[-1] in (64, 128, 256)
+                    and hasattr(hip_ext(), "ce_linear_fwd")
+                ):
+                    # fused linear+CE: logits never reach HBM on the forward
+                    # pass; backward recomputes them and fuses dhidden
+                    from replay_amd.ops.autograd import fused_linear_cross_entropy
+
+                    return fused_linear_cross_entropy(hidden2d, weight, labels.reshape(-1), -100)
                 # materialized path: ONE bf16 logits buffer, fused one-pass
                 # LSE forward + in-place dlogits backward (measured faster
                 # than chunking at V<=1e5: no recompute, full-width GEMMs)
                 from replay_amd.ops.autograd import fused_cross_entropy
 
-                logits2d = hidden2d @ weight.to(hidden2d.dtype).t()
+                logits2d = hidden2d @ weight.t()
                 return fused_cross_entropy(logits2d, labels.reshape(-1), -100)
         logits = self.logits_callback(embeddings)  # [B, L, V]
         return torch.nn.functional.cross_entropy(

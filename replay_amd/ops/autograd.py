@@ -51,6 +51,48 @@ def fused_cross_entropy(logits2d: torch.Tensor, labels: torch.Tensor, ignore_ind
     return FusedCrossEntropyFunction.apply(logits2d, labels, ignore_index)
 
 
+class FusedLinearCrossEntropyFunction(torch.autograd.Function):
+    """Linear + softmax-CE without materializing [N, V] logits on the forward
+    pass (ce_linear.hip): forward computes per-row LSE and the label logit in
+    the MFMA GEMM epilogue; backward recomputes dlogits tile-wise, fuses the
+    dhidden GEMM in-kernel (E <= 128), and materializes dlogits once for the
+    weight-gradient hipBLASLt GEMM."""
+
+    @staticmethod
+    def forward(ctx, hidden2d: torch.Tensor, weight: torch.Tensor, labels: torch.Tensor,
+                ignore_index: int = -100):
+        ext = hip_ext()
+        hidden2d = hidden2d.contiguous()
+        weight = weight.contiguous()
+        lse, lab_logit = ext.ce_linear_fwd(hidden2d, weight, labels)
+        valid = labels != ignore_index
+        count = valid.sum()
+        loss = torch.where(valid, lse - lab_logit, torch.zeros_like(lse)).sum()
+        loss = loss / count.clamp(min=1).to(loss.dtype)
+        ctx.save_for_backward(hidden2d, weight, labels, lse, valid, count)
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        ext = hip_ext()
+        hidden2d, weight, labels, lse, valid, count = ctx.saved_tensors
+        g = dloss.to(torch.float32) / count.clamp(min=1).to(torch.float32)
+        # the kernel folds |g| into adj = lse - ln|g|; the (uniform) sign of
+        # the upstream gradient travels as a scalar
+        gsign = float(torch.sign(g))
+        gscale = torch.where(valid, g.abs(), torch.zeros((), device=valid.device))
+        dlogits, dhidden = ext.ce_linear_bwd(hidden2d, weight, labels, lse, gscale, gsign)
+        if dhidden.numel() == 0:  # E > 128: host GEMM for the input gradient
+            dhidden = dlogits @ weight
+        dweight = dlogits.t() @ hidden2d
+        return dhidden, dweight.to(weight.dtype), None, None
+
+
+def fused_linear_cross_entropy(hidden2d: torch.Tensor, weight: torch.Tensor,
+                               labels: torch.Tensor, ignore_index: int = -100):
+    return FusedLinearCrossEntropyFunction.apply(hidden2d, weight, labels, ignore_index)
+
+
 class FlashAttentionFunction(torch.autograd.Function):
     """Fused attention (K1/K2): q,k,v [B,H,L,Dh]; padding_mask [B,L] bool
     (True = valid); causal flag.  Backward recomputes P from the saved LSE."""

@@ -1,0 +1,532 @@
+// Fused linear + softmax cross-entropy for gfx950 (CDNA4) — bf16, E in
+// {64, 128, 256}, arbitrary catalog V.
+//
+// K10 taken one step further than the materialized CE pair:
+// the [N, V] logits matrix NEVER exists in HBM on the forward pass.
+//
+//   fwd:  logits = hidden . W^T computed tile-wise on the MFMA cores
+//         (same resident-A / streamed-B geometry as scored_topk_gemm.hip);
+//         the epilogue folds each 64-item accumulator tile into a per-row
+//         online logsumexp and captures the label column's logit.  HBM
+//         traffic: hidden once + the item table once (L2-resident after the
+//         first workgroup sweep) + two [N] fp32 vectors — vs 2x [N, V] bf16
+//         (GEMM write + CE read) for the unfused pair.
+//   bwd:  recomputes the logit tiles, forms dlogits = (softmax - onehot) * g
+//         in registers, and (a) stores dlogits[N, V] bf16 through an LDS
+//         bounce so every global store is a coalesced 16-B chunk (the one
+//         [N, V] pass that must remain: the weight gradient GEMM
+//         dW = dlogits^T . hidden consumes it via hipBLASLt), and (b) for
+//         E <= 128 fuses dhidden = dlogits . W on the second MFMA chain with
+//         a transposed W tile staged in LDS — removing the dX GEMM's [N, V]
+//         read as well.
+//
+// Net per-step traffic for the bench shape (N = 409600, V = 27278, E = 64):
+// 134 GB (materialized pair) -> ~45 GB.
+//
+// Fragment maps (guide G9, verified by the GPU parity tests):
+//   A (16x32 bf16): lane l holds A[row = l&15][k = (l>>4)*8 + j], j=0..7
+//   B (32x16 bf16): lane l holds B[k = (l>>4)*8 + j][col = l&15]
+//   C (16x16 f32):  lane l holds C[row = (l>>4)*4 + r][col = l&15], r=0..3
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "common.h"
+
+namespace {
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+// combine two online-logsumexp states (m, s) across lanes: each of the 16
+// lanes of a C-fragment column group owns a disjoint column subset, so the
+// row's full state is the butterfly merge over lane bits 0-3.
+__device__ __forceinline__ void lse_combine16(float& m, float& s) {
+#pragma unroll
+  for (int off = 1; off < 16; off <<= 1) {
+    const float mo = __shfl_xor(m, off, WAVE);
+    const float so = __shfl_xor(s, off, WAVE);
+    const float m2 = fmaxf(m, mo);
+    s = ((m2 == -INFINITY) ? 0.f : s * __expf(m - m2)) +
+        ((m2 == -INFINITY) ? 0.f : so * __expf(mo - m2));
+    m = m2;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// forward: per-row logsumexp + label logit, no logits materialization
+// ---------------------------------------------------------------------------
+template <int E, bool RESIDENT>
+__global__ __launch_bounds__(256, 2) void ce_linear_fwd_kernel(
+    const __hip_bfloat16* __restrict__ hidden,  // [M, E]
+    const __hip_bfloat16* __restrict__ w,       // [V, E]
+    const int64_t* __restrict__ labels,         // [M]
+    float* __restrict__ lse_out,                // [M]
+    float* __restrict__ lab_out,                // [M] label logit (0 if none)
+    int M, int64_t V) {
+  constexpr int KSTEPS = E / 32;
+  constexpr int MF = 4;  // row fragments per wave: wave owns 64 rows
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int m0 = blockIdx.x * 256 + wave * 64;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // [labels: 256 i32][label logit: 256 f32][!RESIDENT: 256 x E bf16 A tile]
+  int* lab_lds = reinterpret_cast<int*>(smem);
+  float* rlab_lds = reinterpret_cast<float*>(lab_lds + 256);
+  __hip_bfloat16* q_lds = reinterpret_cast<__hip_bfloat16*>(rlab_lds + 256);
+  auto lds_off = [&](int row, int k_byte) {
+    return row * (E * 2) + (k_byte ^ ((row & 7) << 4));
+  };
+  for (int i = threadIdx.x; i < 256; i += blockDim.x) {
+    const int row = blockIdx.x * 256 + i;
+    lab_lds[i] = (row < M) ? (int)labels[row] : -1;
+    rlab_lds[i] = -INFINITY;  // unique-writer capture; 0 at store if unset
+  }
+  __syncthreads();
+  constexpr int A_KS = RESIDENT ? KSTEPS : 1;
+  bf16x8 a_frag[MF][A_KS];
+  if constexpr (!RESIDENT) {
+    const int row_q0 = blockIdx.x * 256;
+    for (int i = threadIdx.x; i < 256 * (E * 2 / 16); i += blockDim.x) {
+      const int row = i / (E * 2 / 16);
+      const int k_byte = (i % (E * 2 / 16)) * 16;
+      const int src_row = min(row_q0 + row, M - 1);
+      uint4 vv = *reinterpret_cast<const uint4*>(
+          reinterpret_cast<const char*>(hidden + (size_t)src_row * E) + k_byte);
+      if (row_q0 + row >= M) vv = uint4{0, 0, 0, 0};
+      *reinterpret_cast<uint4*>(reinterpret_cast<char*>(q_lds) + lds_off(row, k_byte)) = vv;
+    }
+    __syncthreads();
+  }
+#pragma unroll
+  for (int mf = 0; mf < MF; ++mf) {
+    if constexpr (RESIDENT) {
+      const int row = m0 + mf * 16 + (lane & 15);
+      const int k0 = (lane >> 4) * 8;
+      const __hip_bfloat16* qr = hidden + (size_t)min(row, M - 1) * E + k0;
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks) {
+        a_frag[mf][ks] = *reinterpret_cast<const bf16x8*>(qr + ks * 32);
+      }
+      if (row >= M) {
+#pragma unroll
+        for (int ks = 0; ks < KSTEPS; ++ks) a_frag[mf][ks] = bf16x8{0};
+      }
+    }
+  }
+  // per-lane online-LSE state for its MF x 4 C rows (labels live in LDS)
+  float r_max[MF][4], r_sum[MF][4];
+#pragma unroll
+  for (int mf = 0; mf < MF; ++mf) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      r_max[mf][r] = -INFINITY;
+      r_sum[mf][r] = 0.f;
+    }
+  }
+
+  const int64_t n_tiles = (V + 63) >> 6;
+  const int bk0 = (lane >> 4) * 8;
+  for (int64_t tile = 0; tile < n_tiles; ++tile) {
+    const int64_t n0 = tile << 6;
+    f32x4 acc[MF][4];
+#pragma unroll
+    for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+      for (int f = 0; f < 4; ++f) acc[mf][f] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks) {
+      bf16x8 b_frag[4];
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        const int64_t item = n0 + f * 16 + (lane & 15);
+        b_frag[f] = *reinterpret_cast<const bf16x8*>(w + (size_t)min(item, V - 1) * E +
+                                                     ks * 32 + bk0);
+      }
+      if constexpr (!RESIDENT) {
+        const int lrow_base = wave * 64 + (lane & 15);
+        const int k_byte = ks * 64 + (lane >> 4) * 16;
+#pragma unroll
+        for (int mf = 0; mf < MF; ++mf) {
+          a_frag[mf][0] = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<const char*>(q_lds) + lds_off(lrow_base + mf * 16, k_byte));
+        }
+      }
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+#pragma unroll
+        for (int mf = 0; mf < MF; ++mf) {
+          acc[mf][f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mf][RESIDENT ? ks : 0], b_frag[f], acc[mf][f], 0, 0, 0);
+        }
+      }
+    }
+    // epilogue: online LSE, branchless, one rescale per 4 values.  The VALU
+    // exp throughput (1/4 rate) is this kernel's floor — 5 exps per 4 logits
+    // beats the per-value online update (2 exps + divergent rescale each).
+    // OOB tail columns become -inf and exp to exactly 0.
+#pragma unroll
+    for (int mf = 0; mf < MF; ++mf) {
+      const int lrow0 = wave * 64 + mf * 16 + (lane >> 4) * 4;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int lab_r = lab_lds[lrow0 + r];
+        float v4[4];
+#pragma unroll
+        for (int f = 0; f < 4; ++f) {
+          const int64_t item = n0 + f * 16 + (lane & 15);
+          v4[f] = (item < V) ? acc[mf][f][r] : -INFINITY;
+          if ((int)item == lab_r && item < V) rlab_lds[lrow0 + r] = v4[f];
+        }
+        const float m4 = fmaxf(fmaxf(v4[0], v4[1]), fmaxf(v4[2], v4[3]));
+        const float nm = fmaxf(r_max[mf][r], m4);
+        r_sum[mf][r] = r_sum[mf][r] * __expf(r_max[mf][r] - nm) + __expf(v4[0] - nm) +
+                       __expf(v4[1] - nm) + __expf(v4[2] - nm) + __expf(v4[3] - nm);
+        r_max[mf][r] = nm;
+      }
+    }
+  }
+#pragma unroll
+  for (int mf = 0; mf < MF; ++mf) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      // merge the 16 per-lane column-subset states into the row state; the
+      // label logit needs no merge (exactly one lane ever wrote its LDS slot)
+      lse_combine16(r_max[mf][r], r_sum[mf][r]);
+      const int row = m0 + mf * 16 + (lane >> 4) * 4 + r;
+      if ((lane & 15) == 0 && row < M) {
+        const float rl = rlab_lds[wave * 64 + mf * 16 + (lane >> 4) * 4 + r];
+        lse_out[row] = r_max[mf][r] + __logf(r_sum[mf][r]);
+        lab_out[row] = (rl == -INFINITY) ? 0.f : rl;
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// backward: recompute logits, emit dlogits (coalesced via LDS bounce) and,
+// when FUSE_DH, dhidden on a second MFMA chain (W^T staged in LDS)
+// ---------------------------------------------------------------------------
+// Register budget note: a naive all-MF formulation (64 score accs + 64
+// dhidden accs + 32 resident A + 96 per-row state VGPRs) spilled 87 VGPRs at
+// occupancy 2 and ran 25 ms/call.  Three cuts bring it under 256:
+//   - per-row state folded to adj = lse - ln|g| (dl = sign * exp(acc - adj),
+//     the |g| for the rare label-column subtraction lives in LDS): 96 -> 32
+//   - labels as int32 (catalogs < 2^31)
+//   - the score MFMA runs in two MF halves (B fragments re-read from L1),
+//     halving the live score-accumulator set: 64 -> 32
+template <int E, bool RESIDENT, bool FUSE_DH>
+__global__ __launch_bounds__(256, 2) void ce_linear_bwd_kernel(
+    const __hip_bfloat16* __restrict__ hidden,  // [M, E]
+    const __hip_bfloat16* __restrict__ w,       // [V, E]
+    const int64_t* __restrict__ labels,         // [M]
+    const float* __restrict__ lse,              // [M]
+    const float* __restrict__ gscale,           // [M] |dloss|/count or 0
+    float gsign,                                // sign(dloss), uniform
+    __hip_bfloat16* __restrict__ dlogits,       // [M, V]
+    __hip_bfloat16* __restrict__ dhidden,       // [M, E] (FUSE_DH only)
+    int M, int64_t V) {
+  constexpr int KSTEPS = E / 32;
+  constexpr int MF = 4;
+  constexpr int OF = E / 16;
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int m0 = blockIdx.x * 256 + wave * 64;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // layout: [wt stage (FUSE_DH): E x 64 bf16] [4 waves x dl bounce 16x64 bf16]
+  //         [|g| per WG row: 256 f32] [!RESIDENT: 256 x E bf16 A tile]
+  __hip_bfloat16* wt_lds = reinterpret_cast<__hip_bfloat16*>(smem);
+  __hip_bfloat16* dl_lds = wt_lds + (FUSE_DH ? E * 64 : 0);
+  float* g_lds = reinterpret_cast<float*>(dl_lds + 4 * 16 * 64);
+  float* adj_lds = g_lds + 256;
+  int* lab_lds = reinterpret_cast<int*>(adj_lds + 256);
+  __hip_bfloat16* q_lds = reinterpret_cast<__hip_bfloat16*>(lab_lds + 256);
+  __hip_bfloat16* my_dl = dl_lds + wave * 16 * 64;
+  // 16-B-granular XOR swizzles (8 bf16 granules stay contiguous)
+  auto dl_off = [&](int row, int col) {  // [16][64] bounce tile, elements
+    return row * 64 + (col ^ ((row & 7) << 3));
+  };
+  auto wt_off = [&](int e, int item) {  // [E][64] transposed W tile, elements
+    return e * 64 + (item ^ ((e & 7) << 3));
+  };
+  auto lds_off = [&](int row, int k_byte) {
+    return row * (E * 2) + (k_byte ^ ((row & 7) << 4));
+  };
+
+  constexpr int A_KS = RESIDENT ? KSTEPS : 1;
+  if constexpr (!RESIDENT) {
+    const int row_q0 = blockIdx.x * 256;
+    for (int i = threadIdx.x; i < 256 * (E * 2 / 16); i += blockDim.x) {
+      const int row = i / (E * 2 / 16);
+      const int k_byte = (i % (E * 2 / 16)) * 16;
+      const int src_row = min(row_q0 + row, M - 1);
+      uint4 vv = *reinterpret_cast<const uint4*>(
+          reinterpret_cast<const char*>(hidden + (size_t)src_row * E) + k_byte);
+      if (row_q0 + row >= M) vv = uint4{0, 0, 0, 0};
+      *reinterpret_cast<uint4*>(reinterpret_cast<char*>(q_lds) + lds_off(row, k_byte)) = vv;
+    }
+    __syncthreads();
+  }
+  // per-row state folded to one float: dl = gsign * exp(acc - adj) with
+  // adj = lse - ln|g| (|g| = 0 or pad row -> adj = +inf -> dl = 0).  All of
+  // it lives in LDS (not VGPRs): 8 reads per tile per lane, hoisted out of
+  // the f-loop; the label-column |g| subtraction is one hit per row per
+  // sweep.
+  for (int i = threadIdx.x; i < 256; i += blockDim.x) {
+    const int row = blockIdx.x * 256 + i;
+    const float g = (row < M) ? gscale[row] : 0.f;
+    g_lds[i] = g;
+    adj_lds[i] = (g > 0.f && row < M) ? lse[row] - __logf(g) : INFINITY;
+    lab_lds[i] = (row < M) ? (int)labels[row] : -1;
+  }
+  __syncthreads();
+  // dhidden accumulators: MF row-fragments x OF output-column fragments
+  f32x4 dh_all[FUSE_DH ? MF : 1][FUSE_DH ? OF : 1];
+  if constexpr (FUSE_DH) {
+#pragma unroll
+    for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+      for (int f = 0; f < OF; ++f) dh_all[mf][f] = f32x4{0.f, 0.f, 0.f, 0.f};
+  }
+
+  const int64_t n_tiles = (V + 63) >> 6;
+  const int bk0 = (lane >> 4) * 8;
+  for (int64_t tile = 0; tile < n_tiles; ++tile) {
+    const int64_t n0 = tile << 6;
+    if constexpr (FUSE_DH) {
+      __syncthreads();  // all waves done with the previous W^T stage
+      for (int i = threadIdx.x; i < 64 * (E / 8); i += blockDim.x) {
+        const int item = i / (E / 8);
+        const int e0 = (i % (E / 8)) * 8;
+        bf16x8 vv = *reinterpret_cast<const bf16x8*>(w + (size_t)min(n0 + item, V - 1) * E + e0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          wt_lds[wt_off(e0 + j, item)] = ((const __hip_bfloat16*)&vv)[j];
+        }
+      }
+      __syncthreads();
+    }
+    // dlogits = (softmax - onehot) * g, bounced per (mf) through LDS so the
+    // [M, V] store is 16-B chunks; same tile feeds the dhidden MFMA A-side.
+    // The MF row-fragments run in two halves so only half the score
+    // accumulators are live at once (register budget, see header note);
+    // B fragments are re-read per half from L1.
+    const bool tail = (n0 + 64) > V;
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      bf16x8 a2[2][A_KS];
+      if constexpr (RESIDENT) {
+#pragma unroll
+        for (int m2 = 0; m2 < 2; ++m2) {
+          const int row = m0 + (half * 2 + m2) * 16 + (lane & 15);
+          const __hip_bfloat16* qr =
+              hidden + (size_t)min(row, M - 1) * E + (lane >> 4) * 8;
+#pragma unroll
+          for (int ks = 0; ks < KSTEPS; ++ks) {
+            a2[m2][ks] = *reinterpret_cast<const bf16x8*>(qr + ks * 32);
+          }
+          if (row >= M) {
+#pragma unroll
+            for (int ks = 0; ks < KSTEPS; ++ks) a2[m2][ks] = bf16x8{0};
+          }
+        }
+      }
+      f32x4 acc[2][4];
+#pragma unroll
+      for (int m2 = 0; m2 < 2; ++m2)
+#pragma unroll
+        for (int f = 0; f < 4; ++f) acc[m2][f] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks) {
+        bf16x8 b_frag[4];
+#pragma unroll
+        for (int f = 0; f < 4; ++f) {
+          const int64_t item = n0 + f * 16 + (lane & 15);
+          b_frag[f] = *reinterpret_cast<const bf16x8*>(w + (size_t)min(item, V - 1) * E +
+                                                       ks * 32 + bk0);
+        }
+        if constexpr (!RESIDENT) {
+          const int lrow_base = wave * 64 + (lane & 15);
+          const int k_byte = ks * 64 + (lane >> 4) * 16;
+#pragma unroll
+          for (int m2 = 0; m2 < 2; ++m2) {
+            a2[m2][0] = *reinterpret_cast<const bf16x8*>(
+                reinterpret_cast<const char*>(q_lds) +
+                lds_off(lrow_base + (half * 2 + m2) * 16, k_byte));
+          }
+        }
+#pragma unroll
+        for (int f = 0; f < 4; ++f) {
+#pragma unroll
+          for (int m2 = 0; m2 < 2; ++m2) {
+            acc[m2][f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a2[m2][RESIDENT ? ks : 0], b_frag[f], acc[m2][f], 0, 0, 0);
+          }
+        }
+      }
+#pragma unroll
+    for (int m2 = 0; m2 < 2; ++m2) {
+      const int mf = half * 2 + m2;
+      const int row0 = m0 + mf * 16;
+      const int lrow0 = wave * 64 + mf * 16 + (lane >> 4) * 4;
+      float adj4[4];
+      int lab4[4];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        adj4[r] = adj_lds[lrow0 + r];
+        lab4[r] = lab_lds[lrow0 + r];
+      }
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        const int64_t item = n0 + f * 16 + (lane & 15);
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          float dl = 0.f;
+          if (item < V) {
+            dl = __expf(acc[m2][f][r] - adj4[r]);
+            if ((int)item == lab4[r]) dl -= g_lds[lrow0 + r];
+          }
+          my_dl[dl_off((lane >> 4) * 4 + r, f * 16 + (lane & 15))] =
+              __float2bfloat16(dl * gsign);
+        }
+      }
+      // wave-private tile: LDS write->read is program-ordered within a wave
+      if constexpr (FUSE_DH) {
+#pragma unroll
+        for (int ks2 = 0; ks2 < 2; ++ks2) {
+          bf16x8 a_dl = *reinterpret_cast<const bf16x8*>(
+              my_dl + dl_off(lane & 15, ks2 * 32 + (lane >> 4) * 8));
+#pragma unroll
+          for (int f = 0; f < OF; ++f) {
+            bf16x8 b_w = *reinterpret_cast<const bf16x8*>(
+                wt_lds + wt_off(f * 16 + (lane & 15), ks2 * 32 + (lane >> 4) * 8));
+            dh_all[mf][f] =
+                __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_dl, b_w, dh_all[mf][f], 0, 0, 0);
+          }
+        }
+      }
+      // coalesced store: each lane writes 16-B chunks of the bounce tile
+      if (!tail) {
+        for (int i = lane; i < 16 * 8; i += WAVE) {
+          const int row = row0 + i / 8;
+          const int c0 = (i % 8) * 8;
+          if (row < M) {
+            *reinterpret_cast<bf16x8*>(dlogits + (size_t)row * V + n0 + c0) =
+                *reinterpret_cast<const bf16x8*>(my_dl + dl_off(i / 8, c0));
+          }
+        }
+      } else {
+        const int ncols = (int)(V - n0);
+        for (int i = lane; i < 16 * ncols; i += WAVE) {
+          const int row = row0 + i / ncols;
+          const int c = i % ncols;
+          if (row < M) {
+            dlogits[(size_t)row * V + n0 + c] = my_dl[dl_off(i / ncols, c)];
+          }
+        }
+      }
+    }
+    }
+  }
+  if constexpr (FUSE_DH) {
+#pragma unroll
+    for (int mf = 0; mf < MF; ++mf) {
+#pragma unroll
+      for (int f = 0; f < OF; ++f) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = m0 + mf * 16 + (lane >> 4) * 4 + r;
+          if (row < M) {
+            dhidden[(size_t)row * E + f * 16 + (lane & 15)] = __float2bfloat16(dh_all[mf][f][r]);
+          }
+        }
+      }
+    }
+  }
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> ce_linear_fwd(torch::Tensor hidden, torch::Tensor w,
+                                         torch::Tensor labels) {
+  TORCH_CHECK(hidden.is_cuda() && hidden.dim() == 2 && hidden.is_contiguous());
+  TORCH_CHECK(w.is_cuda() && w.dim() == 2 && w.is_contiguous());
+  TORCH_CHECK(hidden.scalar_type() == torch::kBFloat16 && w.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(labels.scalar_type() == torch::kInt64);
+  const int M = (int)hidden.size(0);
+  const int E = (int)hidden.size(1);
+  const int64_t V = w.size(0);
+  TORCH_CHECK(w.size(1) == E, "dim mismatch");
+  TORCH_CHECK(V < (int64_t)INT32_MAX - 64, "catalog must fit int32");
+  auto opts_f = hidden.options().dtype(torch::kFloat32);
+  auto lse = torch::empty({M}, opts_f);
+  auto lab_logit = torch::empty({M}, opts_f);
+  auto labels_c = labels.contiguous();
+  const int m_tiles = (M + 255) / 256;
+  auto stream = at::cuda::getCurrentHIPStream();
+#define LAUNCH_CLF(EE)                                                                     \
+  hipLaunchKernelGGL((ce_linear_fwd_kernel<EE, (EE <= 128)>), dim3(m_tiles), dim3(256),    \
+                     2048 + ((EE <= 128) ? 0 : (size_t)256 * EE * 2), stream,              \
+                     reinterpret_cast<const __hip_bfloat16*>(hidden.data_ptr()),           \
+                     reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),                \
+                     labels_c.data_ptr<int64_t>(), lse.data_ptr<float>(),                  \
+                     lab_logit.data_ptr<float>(), M, V)
+  if (E == 64) {
+    LAUNCH_CLF(64);
+  } else if (E == 128) {
+    LAUNCH_CLF(128);
+  } else if (E == 256) {
+    LAUNCH_CLF(256);
+  } else {
+    TORCH_CHECK(false, "ce_linear supports E in {64, 128, 256}");
+  }
+#undef LAUNCH_CLF
+  return {lse, lab_logit};
+}
+
+std::vector<torch::Tensor> ce_linear_bwd(torch::Tensor hidden, torch::Tensor w,
+                                         torch::Tensor labels, torch::Tensor lse,
+                                         torch::Tensor gscale, double gsign) {
+  TORCH_CHECK(hidden.is_cuda() && hidden.dim() == 2 && hidden.is_contiguous());
+  TORCH_CHECK(w.is_cuda() && w.dim() == 2 && w.is_contiguous());
+  const int M = (int)hidden.size(0);
+  const int E = (int)hidden.size(1);
+  const int64_t V = w.size(0);
+  TORCH_CHECK(V < (int64_t)INT32_MAX, "catalog must fit int32");
+  auto dlogits = torch::empty({(int64_t)M, V}, hidden.options());
+  const bool fuse_dh = E == 64;
+  auto dhidden = fuse_dh ? torch::empty_like(hidden)
+                         : torch::empty({0}, hidden.options());  // host GEMM fallback
+  auto labels_c = labels.contiguous();
+  auto gscale_c = gscale.to(torch::kFloat32).contiguous();
+  const int m_tiles = (M + 255) / 256;
+  auto stream = at::cuda::getCurrentHIPStream();
+#define LAUNCH_CLB(EE)                                                                       \
+  do {                                                                                       \
+    constexpr bool RES = (EE <= 128);                                                        \
+    constexpr bool FDH = (EE == 64); /* E=128 dh accumulators spill (128 VGPRs) */           \
+    size_t lds = (FDH ? (size_t)EE * 64 * 2 : 0) + 4 * 16 * 64 * 2 + 3 * 256 * 4 +           \
+                 (RES ? 0 : (size_t)256 * EE * 2);                                           \
+    hipLaunchKernelGGL((ce_linear_bwd_kernel<EE, RES, FDH>), dim3(m_tiles), dim3(256), lds,  \
+                       stream, reinterpret_cast<const __hip_bfloat16*>(hidden.data_ptr()),   \
+                       reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),                \
+                       labels_c.data_ptr<int64_t>(), lse.data_ptr<float>(),                  \
+                       gscale_c.data_ptr<float>(), (float)gsign,                           \
+                       reinterpret_cast<__hip_bfloat16*>(dlogits.data_ptr()),                \
+                       FDH ? reinterpret_cast<__hip_bfloat16*>(dhidden.data_ptr())           \
+                           : nullptr,                                                        \
+                       M, V);                                                                \
+  } while (0)
+  if (E == 64) {
+    LAUNCH_CLB(64);
+  } else if (E == 128) {
+    LAUNCH_CLB(128);
+  } else if (E == 256) {
+    LAUNCH_CLB(256);
+  } else {
+    TORCH_CHECK(false, "ce_linear supports E in {64, 128, 256}");
+  }
+#undef LAUNCH_CLB
+  return {dlogits, dhidden};
+}
