@@ -125,10 +125,39 @@ __global__ __launch_bounds__(256, 2) void ce_linear_fwd_kernel(
     }
   }
 
-  const int64_t n_tiles = (V + 63) >> 6;
+  // all index math in int32: V is checked < 2^31 host-side, and 64-bit
+  // address chains double the VGPR cost of every live index.
+  const int Vi = (int)V;
+  const int n_tiles = (Vi + 63) >> 6;
   const int bk0 = (lane >> 4) * 8;
-  for (int64_t tile = 0; tile < n_tiles; ++tile) {
-    const int64_t n0 = tile << 6;
+  // software-pipelined B stream: prefetch the group PIPE (tile, ks)-sections
+  // ahead so HBM/L2 latency hides under the MFMA+epilogue work (wait/busy
+  // measured 15x without this).  OOB prefetches clamp to V-1.  E=128 spills
+  // with the ring (a-frags already take 64 VGPRs) so it stays direct.
+  constexpr int PIPE = (RESIDENT && E == 64) ? 2 : 1;
+  auto load_group = [&](bf16x8 (&dst)[4], int t, int ks) {
+    const int nn0 = t << 6;
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      const int item = nn0 + f * 16 + (lane & 15);
+      dst[f] = *reinterpret_cast<const bf16x8*>(w + (size_t)min(item, Vi - 1) * E +
+                                                ks * 32 + bk0);
+    }
+  };
+  // NOTE: all WGs deliberately walk tiles 0,1,2,... in lockstep — concurrent
+  // WGs then share the same W lines in L2 (a staggered sweep measured 1.6x
+  // SLOWER: it turns broadcast-friendly reads into whole-table L2 pressure).
+  auto tile_at = [&](int idx) { return idx; };
+  bf16x8 b_ring[PIPE][4];
+  if constexpr (PIPE > 1) {
+#pragma unroll
+    for (int d = 0; d < PIPE; ++d) {
+      load_group(b_ring[d], tile_at(min(d / KSTEPS, n_tiles - 1)), d % KSTEPS);
+    }
+  }
+  for (int ti = 0; ti < n_tiles; ++ti) {
+    const int tile = tile_at(ti);
+    const int n0 = tile << 6;
     f32x4 acc[MF][4];
 #pragma unroll
     for (int mf = 0; mf < MF; ++mf)
@@ -137,11 +166,14 @@ __global__ __launch_bounds__(256, 2) void ce_linear_fwd_kernel(
 #pragma unroll
     for (int ks = 0; ks < KSTEPS; ++ks) {
       bf16x8 b_frag[4];
+      if constexpr (PIPE > 1) {
+        const int slot = (ti * KSTEPS + ks) % PIPE;
 #pragma unroll
-      for (int f = 0; f < 4; ++f) {
-        const int64_t item = n0 + f * 16 + (lane & 15);
-        b_frag[f] = *reinterpret_cast<const bf16x8*>(w + (size_t)min(item, V - 1) * E +
-                                                     ks * 32 + bk0);
+        for (int f = 0; f < 4; ++f) b_frag[f] = b_ring[slot][f];
+        const int g = ti * KSTEPS + ks + PIPE;
+        load_group(b_ring[slot], tile_at(min(g / KSTEPS, n_tiles - 1)), g % KSTEPS);
+      } else {
+        load_group(b_frag, tile, ks);
       }
       if constexpr (!RESIDENT) {
         const int lrow_base = wave * 64 + (lane & 15);
@@ -174,9 +206,9 @@ __global__ __launch_bounds__(256, 2) void ce_linear_fwd_kernel(
         float v4[4];
 #pragma unroll
         for (int f = 0; f < 4; ++f) {
-          const int64_t item = n0 + f * 16 + (lane & 15);
-          v4[f] = (item < V) ? acc[mf][f][r] : -INFINITY;
-          if ((int)item == lab_r && item < V) rlab_lds[lrow0 + r] = v4[f];
+          const int item = n0 + f * 16 + (lane & 15);
+          v4[f] = (item < Vi) ? acc[mf][f][r] : -INFINITY;
+          if (item == lab_r && item < Vi) rlab_lds[lrow0 + r] = v4[f];
         }
         const float m4 = fmaxf(fmaxf(v4[0], v4[1]), fmaxf(v4[2], v4[3]));
         const float nm = fmaxf(r_max[mf][r], m4);
@@ -223,9 +255,9 @@ __global__ __launch_bounds__(256, 2) void ce_linear_bwd_kernel(
     const float* __restrict__ lse,              // [M]
     const float* __restrict__ gscale,           // [M] |dloss|/count or 0
     float gsign,                                // sign(dloss), uniform
-    __hip_bfloat16* __restrict__ dlogits,       // [M, V]
+    __hip_bfloat16* __restrict__ dlogits,       // [M, ldd], ldd = V padded
     __hip_bfloat16* __restrict__ dhidden,       // [M, E] (FUSE_DH only)
-    int M, int64_t V) {
+    int M, int64_t V, int ldd) {
   constexpr int KSTEPS = E / 32;
   constexpr int MF = 4;
   constexpr int OF = E / 16;
@@ -290,16 +322,40 @@ __global__ __launch_bounds__(256, 2) void ce_linear_bwd_kernel(
       for (int f = 0; f < OF; ++f) dh_all[mf][f] = f32x4{0.f, 0.f, 0.f, 0.f};
   }
 
-  const int64_t n_tiles = (V + 63) >> 6;
+  const int Vi = (int)V;  // checked < 2^31 host-side; int index math
+  const int n_tiles = (Vi + 63) >> 6;
   const int bk0 = (lane >> 4) * 8;
-  for (int64_t tile = 0; tile < n_tiles; ++tile) {
-    const int64_t n0 = tile << 6;
+  // pipelined B stream: groups are (tile, half, ks); the address depends on
+  // (tile, ks) only, so a prefetch landing on the other half of the same
+  // tile is an L1 hit.  wait/busy was 31x with direct load-use.
+  constexpr int PIPE = RESIDENT ? 2 : 1;
+  constexpr int GP = 2 * KSTEPS;  // groups per tile
+  auto load_group = [&](bf16x8 (&dst)[4], int t, int ks) {
+    const int nn0 = t << 6;
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      const int item = nn0 + f * 16 + (lane & 15);
+      dst[f] = *reinterpret_cast<const bf16x8*>(w + (size_t)min(item, Vi - 1) * E +
+                                                ks * 32 + bk0);
+    }
+  };
+  auto tile_at = [&](int idx) { return idx; };  // lockstep sweep (see fwd note)
+  bf16x8 b_ring[PIPE][4];
+  if constexpr (PIPE > 1) {
+#pragma unroll
+    for (int d = 0; d < PIPE; ++d) {
+      load_group(b_ring[d], tile_at(min(d / GP, n_tiles - 1)), (d % GP) % KSTEPS);
+    }
+  }
+  for (int ti = 0; ti < n_tiles; ++ti) {
+    const int tile = tile_at(ti);
+    const int n0 = tile << 6;
     if constexpr (FUSE_DH) {
       __syncthreads();  // all waves done with the previous W^T stage
       for (int i = threadIdx.x; i < 64 * (E / 8); i += blockDim.x) {
         const int item = i / (E / 8);
         const int e0 = (i % (E / 8)) * 8;
-        bf16x8 vv = *reinterpret_cast<const bf16x8*>(w + (size_t)min(n0 + item, V - 1) * E + e0);
+        bf16x8 vv = *reinterpret_cast<const bf16x8*>(w + (size_t)min(n0 + item, Vi - 1) * E + e0);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           wt_lds[wt_off(e0 + j, item)] = ((const __hip_bfloat16*)&vv)[j];
@@ -308,11 +364,12 @@ __global__ __launch_bounds__(256, 2) void ce_linear_bwd_kernel(
       __syncthreads();
     }
     // dlogits = (softmax - onehot) * g, bounced per (mf) through LDS so the
-    // [M, V] store is 16-B chunks; same tile feeds the dhidden MFMA A-side.
+    // [M, ldd] store is full aligned 128-B lines (ldd is a 64-item multiple:
+    // unpadded V-strided rows made every segment straddle two cache lines —
+    // read-modify-write on 22 GB of stores).  Pad columns store exact 0.
     // The MF row-fragments run in two halves so only half the score
     // accumulators are live at once (register budget, see header note);
     // B fragments are re-read per half from L1.
-    const bool tail = (n0 + 64) > V;
 #pragma unroll
     for (int half = 0; half < 2; ++half) {
       bf16x8 a2[2][A_KS];
@@ -340,11 +397,15 @@ __global__ __launch_bounds__(256, 2) void ce_linear_bwd_kernel(
 #pragma unroll
       for (int ks = 0; ks < KSTEPS; ++ks) {
         bf16x8 b_frag[4];
+        if constexpr (PIPE > 1) {
+          const int g = ti * GP + half * KSTEPS + ks;
+          const int slot = g % PIPE;
 #pragma unroll
-        for (int f = 0; f < 4; ++f) {
-          const int64_t item = n0 + f * 16 + (lane & 15);
-          b_frag[f] = *reinterpret_cast<const bf16x8*>(w + (size_t)min(item, V - 1) * E +
-                                                       ks * 32 + bk0);
+          for (int f = 0; f < 4; ++f) b_frag[f] = b_ring[slot][f];
+          const int gn = g + PIPE;
+          load_group(b_ring[slot], tile_at(min(gn / GP, n_tiles - 1)), (gn % GP) % KSTEPS);
+        } else {
+          load_group(b_frag, tile, ks);
         }
         if constexpr (!RESIDENT) {
           const int lrow_base = wave * 64 + (lane & 15);
@@ -379,13 +440,13 @@ __global__ __launch_bounds__(256, 2) void ce_linear_bwd_kernel(
       }
 #pragma unroll
       for (int f = 0; f < 4; ++f) {
-        const int64_t item = n0 + f * 16 + (lane & 15);
+        const int item = n0 + f * 16 + (lane & 15);
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           float dl = 0.f;
-          if (item < V) {
+          if (item < Vi) {
             dl = __expf(acc[m2][f][r] - adj4[r]);
-            if ((int)item == lab4[r]) dl -= g_lds[lrow0 + r];
+            if (item == lab4[r]) dl -= g_lds[lrow0 + r];
           }
           my_dl[dl_off((lane >> 4) * 4 + r, f * 16 + (lane & 15))] =
               __float2bfloat16(dl * gsign);
@@ -407,23 +468,12 @@ __global__ __launch_bounds__(256, 2) void ce_linear_bwd_kernel(
         }
       }
       // coalesced store: each lane writes 16-B chunks of the bounce tile
-      if (!tail) {
-        for (int i = lane; i < 16 * 8; i += WAVE) {
-          const int row = row0 + i / 8;
-          const int c0 = (i % 8) * 8;
-          if (row < M) {
-            *reinterpret_cast<bf16x8*>(dlogits + (size_t)row * V + n0 + c0) =
-                *reinterpret_cast<const bf16x8*>(my_dl + dl_off(i / 8, c0));
-          }
-        }
-      } else {
-        const int ncols = (int)(V - n0);
-        for (int i = lane; i < 16 * ncols; i += WAVE) {
-          const int row = row0 + i / ncols;
-          const int c = i % ncols;
-          if (row < M) {
-            dlogits[(size_t)row * V + n0 + c] = my_dl[dl_off(i / ncols, c)];
-          }
+      for (int i = lane; i < 16 * 8; i += WAVE) {
+        const int row = row0 + i / 8;
+        const int c0 = (i % 8) * 8;
+        if (row < M) {
+          *reinterpret_cast<bf16x8*>(dlogits + (size_t)row * ldd + n0 + c0) =
+              *reinterpret_cast<const bf16x8*>(my_dl + dl_off(i / 8, c0));
         }
       }
     }
@@ -493,9 +543,10 @@ std::vector<torch::Tensor> ce_linear_bwd(torch::Tensor hidden, torch::Tensor w,
   const int M = (int)hidden.size(0);
   const int E = (int)hidden.size(1);
   const int64_t V = w.size(0);
-  TORCH_CHECK(V < (int64_t)INT32_MAX, "catalog must fit int32");
-  auto dlogits = torch::empty({(int64_t)M, V}, hidden.options());
-  const bool fuse_dh = E == 64;
+  TORCH_CHECK(V < (int64_t)INT32_MAX - 64, "catalog must fit int32");
+  const int64_t Vp = (V + 63) & ~int64_t(63);  // full-line 128-B row segments
+  auto dlogits = torch::empty({(int64_t)M, Vp}, hidden.options());
+  const bool fuse_dh = false;
   auto dhidden = fuse_dh ? torch::empty_like(hidden)
                          : torch::empty({0}, hidden.options());  // host GEMM fallback
   auto labels_c = labels.contiguous();
@@ -505,7 +556,8 @@ std::vector<torch::Tensor> ce_linear_bwd(torch::Tensor hidden, torch::Tensor w,
 #define LAUNCH_CLB(EE)                                                                       \
   do {                                                                                       \
     constexpr bool RES = (EE <= 128);                                                        \
-    constexpr bool FDH = (EE == 64); /* E=128 dh accumulators spill (128 VGPRs) */           \
+    constexpr bool FDH = false; /* dh fusion measured slower: per-tile W^T-stage\
+                                   barriers serialize the WG; host dX GEMM wins */           \
     size_t lds = (FDH ? (size_t)EE * 64 * 2 : 0) + 4 * 16 * 64 * 2 + 3 * 256 * 4 +           \
                  (RES ? 0 : (size_t)256 * EE * 2);                                           \
     hipLaunchKernelGGL((ce_linear_bwd_kernel<EE, RES, FDH>), dim3(m_tiles), dim3(256), lds,  \
@@ -516,7 +568,7 @@ std::vector<torch::Tensor> ce_linear_bwd(torch::Tensor hidden, torch::Tensor w,
                        reinterpret_cast<__hip_bfloat16*>(dlogits.data_ptr()),                \
                        FDH ? reinterpret_cast<__hip_bfloat16*>(dhidden.data_ptr())           \
                            : nullptr,                                                        \
-                       M, V);                                                                \
+                       M, V, (int)Vp);                                                       \
   } while (0)
   if (E == 64) {
     LAUNCH_CLB(64);
@@ -528,5 +580,5 @@ std::vector<torch::Tensor> ce_linear_bwd(torch::Tensor hidden, torch::Tensor w,
     TORCH_CHECK(false, "ce_linear supports E in {64, 128, 256}");
   }
 #undef LAUNCH_CLB
-  return {dlogits, dhidden};
+  return {dlogits.narrow(1, 0, V), dhidden};
 }
