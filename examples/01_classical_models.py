@@ -2,6 +2,11 @@
 synthetic ML-1M-shape log -> Dataset -> split -> encode -> ItemKNN/ALS/PopRec
 -> Experiment comparison."""
 
+import sys
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))  # repo root
+
 import numpy as np
 import pandas as pd
 

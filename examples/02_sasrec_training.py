@@ -3,6 +3,11 @@ interactions -> SequenceTokenizer -> torch datasets -> Trainer ->
 validation metrics -> top-k predictions with filter_seen.
 Runs on CPU or GPU (bf16 + HIP kernels when on MI355X)."""
 
+import sys
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))  # repo root
+
 import numpy as np
 import pandas as pd
 import torch
