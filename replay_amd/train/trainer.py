@@ -72,6 +72,7 @@ class Trainer:
 
         self.global_step = 0
         self.current_epoch = 0
+        self.should_stop = False
         self._sync_metrics: Dict[str, float] = {}
         self._module = None
         self._ddp_model = None
@@ -215,7 +216,8 @@ class Trainer:
             # Lightning convention: current_epoch counts COMPLETED epochs, so
             # a checkpoint saved after epoch e resumes at e+1
             self.current_epoch = epoch + 1
-            if stop:
+            self._call_callbacks("on_epoch_complete")
+            if stop or self.should_stop:
                 break
         self._call_callbacks("on_fit_end")
         return self

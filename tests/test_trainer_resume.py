@@ -81,3 +81,35 @@ def test_roctx_range_noop_on_cpu():
     with timer.time("phase"):
         sum(range(1000))
     assert "phase" in timer.summary()
+
+
+class TestTrainerCallbacks:
+    def test_model_checkpoint_best_only(self, tmp_path):
+        from replay_amd.train import ModelCheckpoint, Trainer
+
+        module, loader = _module(), _loader()
+        cb = ModelCheckpoint(dirpath=tmp_path, monitor="train_loss", mode="min")
+        Trainer(max_epochs=3, accelerator="cpu", callbacks=[cb]).fit(module, loader)
+        assert cb.best_model_path is not None
+        import os
+
+        assert os.path.exists(cb.best_model_path)
+        assert cb.best_model_score is not None
+        # only the single best checkpoint is kept
+        assert len(list(tmp_path.glob("*.ckpt"))) == 1
+
+    def test_early_stopping_stops(self):
+        from replay_amd.train import EarlyStopping, Trainer
+
+        module, loader = _module(), _loader()
+
+        class ConstantMetric:
+            def on_epoch_complete(self, trainer, mod):
+                trainer.logged_metrics["plateau"] = 1.0
+
+        es = EarlyStopping(monitor="plateau", patience=2, mode="min")
+        tr = Trainer(max_epochs=50, accelerator="cpu", callbacks=[ConstantMetric(), es])
+        tr.fit(module, loader)
+        # first epoch sets best, then 2 epochs of no improvement stop it
+        assert tr.current_epoch <= 4
+        assert es.stopped_epoch is not None
