@@ -94,6 +94,15 @@ INVENTORY = {
     ],
     "replay_amd.experimental.scenarios": ["TwoStagesScenario"],
     "replay_amd.experimental.scenarios.obp_wrapper": ["OBPOfflinePolicyLearner", "ips_estimate", "snips_estimate"],
+    "replay_amd.experimental.preprocessing": ["Padder", "SequenceGenerator"],
+    "replay_amd.experimental.nn.data": ["TensorSchemaBuilder"],
+    "replay_amd.experimental.utils.model_handler": ["save", "load"],
+    "replay_amd.experimental.utils.session_handler": ["State"],
+    # data utils + legacy optimizer surface
+    "replay_amd.data.utils": ["UniformBatching", "uniform_batch_count"],
+    "replay_amd.data.utils.typing": ["torch_to_numpy", "numpy_to_torch", "numpy_to_pyarrow", "pyarrow_to_numpy"],
+    "replay_amd.models.nn.optimizer_utils": ["OptimizerFactory", "LRSchedulerFactory", "FatOptimizerFactory", "FatLRSchedulerFactory"],
+    "replay_amd.train": ["Trainer", "ModelCheckpoint", "EarlyStopping"],
 }
 
 
