@@ -106,3 +106,48 @@ def test_gather_embeddings_differentiable():
 def test_ddp_gradients_identical():
     out = _spawn(_run_ddp_step, port=29523)
     assert out[0] == pytest.approx(out[1], rel=1e-6)
+
+
+def _run_sharded_topk(rank, world, port, out):
+    _init(rank, world, port)
+    from replay_amd.ops.topk import catalog_topk, sharded_catalog_topk
+
+    torch.manual_seed(0)  # identical across ranks
+    V, E, B, K = 64, 8, 5, 6
+    items = torch.randn(V, E)
+    queries = torch.randn(B, E)
+    seen = torch.randint(0, V, (B, 4))
+    shard = V // world
+    local = items[rank * shard:(rank + 1) * shard]
+    s, i = sharded_catalog_topk(queries, local, K, shard_offset=rank * shard, seen=seen)
+    ref_s, ref_i = catalog_topk(queries, items, K, seen=seen)
+    out[rank] = (
+        torch.allclose(s, ref_s, atol=1e-5),
+        bool((i == ref_i).all() or torch.allclose(s, ref_s, atol=1e-5)),  # ties may reorder ids
+        i.tolist(),
+    )
+    torch.distributed.destroy_process_group()
+
+
+def test_sharded_catalog_topk_matches_single_process():
+    out = _spawn(_run_sharded_topk, port=29524)
+    assert out[0][0] and out[1][0], "sharded scores != full-table scores"
+    assert out[0][2] == out[1][2], "ranks disagree on the merged top-K"
+
+
+def _run_trainer_sync_dist(rank, world, port, out):
+    _init(rank, world, port)
+    from replay_amd.train import Trainer
+
+    trainer = Trainer(accelerator="cpu")
+    trainer._module = None
+    trainer._log("metric", float(rank + 1), sync_dist=True)  # 1.0 and 2.0
+    trainer._reduce_sync_metrics()
+    out[rank] = trainer.logged_metrics["metric"]
+    torch.distributed.destroy_process_group()
+
+
+def test_trainer_sync_dist_mean():
+    out = _spawn(_run_trainer_sync_dist, port=29525)
+    assert out[0] == pytest.approx(1.5)
+    assert out[0] == out[1]
