@@ -15,10 +15,12 @@
 //         in registers, and (a) stores dlogits[N, V] bf16 through an LDS
 //         bounce so every global store is a coalesced 16-B chunk (the one
 //         [N, V] pass that must remain: the weight gradient GEMM
-//         dW = dlogits^T . hidden consumes it via hipBLASLt), and (b) for
-//         E <= 128 fuses dhidden = dlogits . W on the second MFMA chain with
-//         a transposed W tile staged in LDS — removing the dX GEMM's [N, V]
-//         read as well.
+//         dW = dlogits^T . hidden consumes it via hipBLASLt).  An in-kernel
+//         dhidden fusion (FUSE_DH: second MFMA chain over a transposed W
+//         tile staged in LDS) is implemented but DISABLED: its two
+//         __syncthreads per tile serialized the workgroup and measured
+//         slower than the hipBLASLt dX GEMM on the padded dlogits view;
+//         kept for a barrier-free (per-wave W^T copies) rework.
 //
 // Net per-step traffic for the bench shape (N = 409600, V = 27278, E = 64):
 // 134 GB (materialized pair) -> ~45 GB.

@@ -414,7 +414,7 @@ class TestFp8Scoring:
 @requires_gpu
 class TestMfmaAttention:
     @pytest.mark.parametrize("causal", [True, False])
-    @pytest.mark.parametrize("shape", [(4, 2, 50, 32), (2, 4, 200, 64), (3, 1, 33, 32), (2, 2, 256, 64)])
+    @pytest.mark.parametrize("shape", [(4, 2, 50, 32), (2, 4, 200, 64), (3, 1, 33, 32), (2, 2, 256, 64), (2, 2, 20, 32), (2, 1, 32, 64)])
     def test_fwd_matches_valu_kernel(self, causal, shape):
         from replay_amd.ops import hip_ext
 
@@ -470,7 +470,7 @@ class TestMfmaAttention:
 @requires_gpu
 class TestMfmaAttentionBwd:
     @pytest.mark.parametrize("causal", [True, False])
-    @pytest.mark.parametrize("shape", [(3, 2, 50, 32), (2, 2, 200, 64), (2, 1, 33, 32)])
+    @pytest.mark.parametrize("shape", [(3, 2, 50, 32), (2, 2, 200, 64), (2, 1, 33, 32), (2, 2, 20, 32), (2, 1, 32, 64)])
     def test_bwd_matches_fp32_reference(self, causal, shape):
         from replay_amd.ops import hip_ext
 
