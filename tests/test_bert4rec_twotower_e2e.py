@@ -1,0 +1,182 @@
+"""End-to-end BERT4Rec and Two-Tower integration on CPU through the Trainer
+(the reference's Lightning-loop test pattern, SURVEY §4 integration tests;
+SASRec has the same coverage in test_sasrec_e2e.py)."""
+
+import numpy as np
+import pandas as pd
+import pytest
+import torch
+
+from replay_amd.data import Dataset, FeatureHint, FeatureInfo, FeatureSchema, FeatureType
+from replay_amd.data.nn import (
+    SequenceTokenizer,
+    TensorFeatureInfo,
+    TensorSchema,
+    TorchSequentialDataset,
+    TorchSequentialValidationDataset,
+)
+from replay_amd.nn.lightning import (
+    ComputeMetricsCallback,
+    LightningModule,
+    OptimizerFactory,
+    PandasTopItemsCallback,
+    SeenItemsFilter,
+)
+from replay_amd.nn.sequential.bert4rec import Bert4Rec
+from replay_amd.nn.sequential.twotower import TwoTower
+from replay_amd.nn.transform import TokenMaskTransform
+from replay_amd.train import Trainer
+
+pytestmark = pytest.mark.torch
+
+N_ITEMS = 30
+MAX_LEN = 8
+
+
+@pytest.fixture(scope="module")
+def tensor_schema():
+    return TensorSchema(
+        [
+            TensorFeatureInfo(
+                "item_id",
+                FeatureType.CATEGORICAL,
+                is_seq=True,
+                feature_hint=FeatureHint.ITEM_ID,
+                cardinality=N_ITEMS,
+                embedding_dim=16,
+            )
+        ]
+    )
+
+
+@pytest.fixture(scope="module")
+def sequential_data(tensor_schema):
+    rng = np.random.default_rng(4)
+    rows = []
+    for q in range(12):
+        for t in range(rng.integers(4, 12)):
+            rows.append((q, rng.integers(0, N_ITEMS), t))
+    inter = pd.DataFrame(rows, columns=["query_id", "item_id", "timestamp"])
+    schema = FeatureSchema(
+        [
+            FeatureInfo("query_id", FeatureType.CATEGORICAL, FeatureHint.QUERY_ID),
+            FeatureInfo("item_id", FeatureType.CATEGORICAL, FeatureHint.ITEM_ID),
+            FeatureInfo("timestamp", FeatureType.NUMERICAL, FeatureHint.TIMESTAMP),
+        ]
+    )
+    ds = Dataset(feature_schema=schema, interactions=inter)
+    return SequenceTokenizer(tensor_schema).fit_transform(ds)
+
+
+class _Loader:
+    def __init__(self, loader, fn):
+        self.loader, self.fn = loader, fn
+
+    def __iter__(self):
+        for b in self.loader:
+            yield self.fn(dict(b))
+
+    def __len__(self):
+        return len(self.loader)
+
+
+def _bert_loader(sequential_data):
+    mask = TokenMaskTransform(generator_seed=0)
+
+    def fn(b):
+        b["labels"] = b["item_id"].clone()
+        b["labels_padding_mask"] = b["padding_mask"]
+        return mask(b)
+
+    ds = TorchSequentialDataset(sequential_data, MAX_LEN)
+    return _Loader(torch.utils.data.DataLoader(ds, batch_size=4), fn)
+
+
+def _twotower_loader(sequential_data):
+    def fn(b):
+        b["labels"] = b["item_id"].roll(-1, dims=1)
+        b["labels_padding_mask"] = b["padding_mask"]
+        return b
+
+    ds = TorchSequentialDataset(sequential_data, MAX_LEN)
+    return _Loader(torch.utils.data.DataLoader(ds, batch_size=4), fn)
+
+
+class TestBert4RecE2E:
+    def test_training_loss_improves(self, sequential_data, tensor_schema):
+        model = Bert4Rec.from_params(
+            tensor_schema, max_sequence_length=MAX_LEN, embedding_dim=16, num_blocks=1, num_heads=2
+        )
+        module = LightningModule(model, OptimizerFactory(lr=5e-3))
+        trainer = Trainer(max_epochs=1, accelerator="cpu", precision="32")
+        trainer.fit(module, _bert_loader(sequential_data))
+        first = trainer.logged_metrics["train_loss"]
+        trainer2 = Trainer(max_epochs=4, accelerator="cpu", precision="32")
+        trainer2.fit(module, _bert_loader(sequential_data))
+        assert trainer2.logged_metrics["train_loss"] < first  # masked-CE learns
+
+    def test_validation_metrics(self, sequential_data, tensor_schema):
+        model = Bert4Rec.from_params(tensor_schema, max_sequence_length=MAX_LEN, embedding_dim=16, num_blocks=1)
+        module = LightningModule(model, OptimizerFactory())
+        val_ds = TorchSequentialValidationDataset(
+            sequential_data, ground_truth=sequential_data, train=sequential_data,
+            max_sequence_length=MAX_LEN,
+        )
+        cb = ComputeMetricsCallback(metrics=["recall", "ndcg"], top_k=[5], item_count=N_ITEMS)
+        Trainer(accelerator="cpu", precision="32", callbacks=[cb]).validate(
+            module, torch.utils.data.DataLoader(val_ds, batch_size=4)
+        )
+        metrics = cb.metric_history[-1]
+        assert set(metrics) == {"recall@5", "ndcg@5"}
+
+    def test_predict_filter_seen(self, sequential_data, tensor_schema):
+        model = Bert4Rec.from_params(tensor_schema, max_sequence_length=MAX_LEN, embedding_dim=16, num_blocks=1)
+        module = LightningModule(model, OptimizerFactory())
+        ds = TorchSequentialDataset(sequential_data, MAX_LEN)
+        cb = PandasTopItemsCallback(top_k=5, postprocessors=[SeenItemsFilter()])
+        Trainer(accelerator="cpu", precision="32", callbacks=[cb]).predict(
+            module, torch.utils.data.DataLoader(ds, batch_size=4), return_predictions=False
+        )
+        recs = cb.get_result()
+        assert len(recs) > 0
+        assert set(recs.columns) >= {"query_id", "item_id"}
+
+    def test_checkpoint_roundtrip(self, tmp_path, sequential_data, tensor_schema):
+        model = Bert4Rec.from_params(tensor_schema, max_sequence_length=MAX_LEN, embedding_dim=16, num_blocks=1)
+        module = LightningModule(model, OptimizerFactory(lr=1e-2))
+        trainer = Trainer(max_epochs=1, accelerator="cpu", precision="32")
+        trainer.fit(module, _bert_loader(sequential_data))
+        path = tmp_path / "b4r.ckpt"
+        trainer.save_checkpoint(path)
+
+        model2 = Bert4Rec.from_params(tensor_schema, max_sequence_length=MAX_LEN, embedding_dim=16, num_blocks=1)
+        module2 = LightningModule(model2, OptimizerFactory())
+        module2.load_state_dict(torch.load(path, weights_only=False)["state_dict"])
+        model.eval(), model2.eval()
+        batch = next(iter(_bert_loader(sequential_data)))
+        with torch.no_grad():
+            torch.testing.assert_close(
+                model.forward_inference(dict(batch)), model2.forward_inference(dict(batch))
+            )
+
+
+class TestTwoTowerE2E:
+    def test_training_runs(self, sequential_data, tensor_schema):
+        model = TwoTower.from_params(tensor_schema, max_sequence_length=MAX_LEN, embedding_dim=16)
+        module = LightningModule(model, OptimizerFactory(lr=1e-3))
+        trainer = Trainer(max_epochs=2, accelerator="cpu", precision="32")
+        trainer.fit(module, _twotower_loader(sequential_data))
+        assert "train_loss" in trainer.logged_metrics
+
+    def test_predict_topk(self, sequential_data, tensor_schema):
+        model = TwoTower.from_params(tensor_schema, max_sequence_length=MAX_LEN, embedding_dim=16)
+        module = LightningModule(model, OptimizerFactory())
+        ds = TorchSequentialDataset(sequential_data, MAX_LEN)
+        cb = PandasTopItemsCallback(top_k=5, postprocessors=[SeenItemsFilter()])
+        Trainer(accelerator="cpu", precision="32", callbacks=[cb]).predict(
+            module, torch.utils.data.DataLoader(ds, batch_size=4), return_predictions=False
+        )
+        recs = cb.get_result()
+        assert len(recs) > 0
+        per_user = recs.groupby("query_id").size()
+        assert (per_user <= 5).all()
