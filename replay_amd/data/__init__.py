@@ -1,4 +1,5 @@
 from .dataset import Dataset
+from .frame_schema import get_schema
 from .schema import FeatureHint, FeatureInfo, FeatureSchema, FeatureSource, FeatureType
 
 __all__ = [
@@ -8,4 +9,5 @@ __all__ = [
     "FeatureSchema",
     "FeatureSource",
     "FeatureType",
+    "get_schema",
 ]

@@ -8,9 +8,13 @@ source bookkeeping.
 
 from __future__ import annotations
 
-from typing import Dict, Iterator, List, Optional, Sequence, Union
+from typing import Mapping, Dict, Iterator, List, Optional, Sequence, Union
 
 from replay_amd.data.schema import FeatureHint, FeatureSource, FeatureType
+
+
+TensorMap = Mapping[str, "torch.Tensor"]
+MutableTensorMap = Dict[str, "torch.Tensor"]
 
 
 class TensorFeatureSource:

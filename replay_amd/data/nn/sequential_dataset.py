@@ -96,3 +96,21 @@ class PandasSequentialDataset(SequentialDataset):
         return PandasSequentialDataset(
             self._schema, self._query_id_column, self._item_id_column, self._sequences[mask]
         )
+
+
+class PolarsSequentialDataset(SequentialDataset):
+    """Polars-backed sequential dataset (reference
+    sequential_dataset.py PolarsSequentialDataset).  Converts through pandas
+    when polars is installed; raises otherwise (POLARS_AVAILABLE gating,
+    utils/types.py pattern)."""
+
+    def __init__(self, tensor_schema, query_id_column, item_id_column, sequences):
+        from replay_amd.utils.types import POLARS_AVAILABLE
+
+        if not POLARS_AVAILABLE:
+            raise ImportError("polars is not installed; use PandasSequentialDataset")
+        pandas_sequences = sequences.to_pandas()
+        self._inner = PandasSequentialDataset(
+            tensor_schema, query_id_column, item_id_column, pandas_sequences
+        )
+        self.__dict__.update(self._inner.__dict__)

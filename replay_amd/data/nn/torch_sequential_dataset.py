@@ -11,7 +11,7 @@ the validation variant adds ground_truth [B, G] and train [B, T] padded -1.
 
 from __future__ import annotations
 
-from typing import Dict, Optional
+from typing import NamedTuple, Dict, Optional
 
 import numpy as np
 import torch
@@ -20,6 +20,28 @@ from .sequential_dataset import SequentialDataset
 
 GROUND_TRUTH_PAD = -1
 TRAIN_PAD = -2
+# reference names (torch_sequential_dataset.py:179-180)
+DEFAULT_GROUND_TRUTH_PADDING_VALUE = GROUND_TRUTH_PAD
+DEFAULT_TRAIN_PADDING_VALUE = TRAIN_PAD
+
+
+class TorchSequentialBatch(NamedTuple):
+    """Legacy tuple view of a training batch (reference :18, deprecated
+    there; batches are plain dicts in the current flow)."""
+
+    query_id: torch.LongTensor
+    padding_mask: torch.BoolTensor
+    features: Dict[str, torch.Tensor]
+
+
+class TorchSequentialValidationBatch(NamedTuple):
+    """Legacy tuple view of a validation batch (reference :167)."""
+
+    query_id: torch.LongTensor
+    padding_mask: torch.BoolTensor
+    features: Dict[str, torch.Tensor]
+    ground_truth: torch.LongTensor
+    train: torch.LongTensor
 
 
 class TorchSequentialDataset(torch.utils.data.Dataset):
@@ -117,7 +139,7 @@ class TorchSequentialValidationDataset(torch.utils.data.Dataset):
         tr = self._train.get_sequence_by_query_id(qid, self._label_name)
         gt_pad = np.full(self._max_gt, GROUND_TRUTH_PAD, dtype=np.int64)
         gt_pad[: len(gt)] = gt
-        tr_pad = np.full(self._max_train, GROUND_TRUTH_PAD, dtype=np.int64)
+        tr_pad = np.full(self._max_train, TRAIN_PAD, dtype=np.int64)
         tr_pad[: len(tr)] = tr
         out["ground_truth"] = torch.from_numpy(gt_pad)
         out["train"] = torch.from_numpy(tr_pad)

@@ -1,3 +1,4 @@
+from .base_torch_rec import TorchRecommender
 from .admm_slim import ADMMSLIM
 from .dt4rec import DT4Rec
 from .misc import (
@@ -14,6 +15,7 @@ from .rl import CQL, DDPG
 from .u_lin_ucb import ULinUCB
 
 __all__ = [
+    "TorchRecommender",
     "ADMMSLIM",
     "DT4Rec",
     "LIGHTFM_AVAILABLE",

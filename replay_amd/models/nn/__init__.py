@@ -5,9 +5,20 @@ the legacy classes' constructor/loss surface while reusing the single
 MI355X transformer implementation underneath.
 """
 
+from replay_amd.utils import TORCH_AVAILABLE
+
 from .sequential.bert4rec import Bert4Rec
 from .sequential.compiled import Bert4RecCompiled, SasRecCompiled
-from .sequential.sasrec import SasRec
+from .sequential.sasrec import (
+    SasRec,
+    SasRecModel,
+    SasRecPredictionBatch,
+    SasRecPredictionDataset,
+    SasRecTrainingBatch,
+    SasRecTrainingDataset,
+    SasRecValidationBatch,
+    SasRecValidationDataset,
+)
 from .sequential.tisasrec import TiSasRec
 
 __all__ = ["Bert4Rec", "Bert4RecCompiled", "SasRecCompiled", "SasRec", "TiSasRec"]

@@ -1,3 +1,5 @@
+from replay_amd.utils import TORCH_AVAILABLE
 from .optimizer_factory import FatLRSchedulerFactory, FatOptimizerFactory, LRSchedulerFactory, OptimizerFactory
 
-__all__ = ["FatLRSchedulerFactory", "FatOptimizerFactory", "LRSchedulerFactory", "OptimizerFactory"]
+__all__ = [
+    "TORCH_AVAILABLE","FatLRSchedulerFactory", "FatOptimizerFactory", "LRSchedulerFactory", "OptimizerFactory"]

@@ -22,6 +22,8 @@ from replay_amd.nn.transform import TokenMaskTransform
 
 from .sasrec import _make_loss
 
+from replay_amd.utils import TORCH_AVAILABLE  # noqa: F401  (legacy surface flag)
+
 
 class Bert4Rec(LightningModule):
     def __init__(

@@ -6,6 +6,7 @@ from replay_amd.nn.lightning.callback.metrics_callback import (
     ComputeMetricsCallback as ValidationMetricsCallback,
 )
 from replay_amd.nn.lightning.callback.predictions_callback import (
+    TopItemsCallbackBase as BasePredictionCallback,
     PandasTopItemsCallback as PandasPredictionCallback,
     PolarsTopItemsCallback as PolarsPredictionCallback,
     QueryEmbeddingsPredictionCallback,
@@ -14,6 +15,7 @@ from replay_amd.nn.lightning.callback.predictions_callback import (
 )
 
 __all__ = [
+    "BasePredictionCallback",
     "ValidationMetricsCallback",
     "PandasPredictionCallback",
     "PolarsPredictionCallback",

@@ -1,3 +1,4 @@
+from replay_amd.utils import TORCH_AVAILABLE
 """Legacy postprocessors namespace (reference replay/models/nn/sequential/
 postprocessors/postprocessors.py: RemoveSeenItems, SampleItems)."""
 
@@ -7,4 +8,5 @@ from replay_amd.nn.lightning.postprocessor.seen_items import (
     SeenItemsFilter as RemoveSeenItems,
 )
 
-__all__ = ["BasePostProcessor", "SampleItems", "RemoveSeenItems"]
+__all__ = [
+    "TORCH_AVAILABLE","BasePostProcessor", "SampleItems", "RemoveSeenItems"]

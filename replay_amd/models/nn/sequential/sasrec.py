@@ -154,3 +154,122 @@ class SasRec(LightningModule):
         self.set_item_embeddings_by_size(old_n + tensor.shape[0])
         with torch.no_grad():
             self._item_embedder.item_emb.weight[old_n : old_n + tensor.shape[0]] = tensor
+
+
+# ---------------------------------------------------------------------------
+# Legacy dataset/batch surface (reference models/nn/sequential/sasrec/
+# dataset.py): thin adapters over the new-generation TorchSequentialDataset
+# that emit the dict keys this legacy SasRec consumes.
+# ---------------------------------------------------------------------------
+from typing import Dict, NamedTuple  # noqa: E402
+
+from replay_amd.data.nn import (  # noqa: E402
+    TorchSequentialDataset,
+    TorchSequentialValidationDataset,
+)
+
+SasRecModel = _NewGenSasRec  # the raw torch module (reference model.py SasRecModel)
+
+
+class SasRecTrainingBatch(NamedTuple):
+    """Legacy tuple view of a training batch (reference dataset.py:20)."""
+
+    query_id: torch.LongTensor
+    padding_mask: torch.BoolTensor
+    features: Dict[str, torch.Tensor]
+    labels: torch.LongTensor
+    labels_padding_mask: torch.BoolTensor
+
+
+class SasRecPredictionBatch(NamedTuple):
+    """Legacy tuple view of a prediction batch (reference dataset.py:133)."""
+
+    query_id: torch.LongTensor
+    padding_mask: torch.BoolTensor
+    features: Dict[str, torch.Tensor]
+
+    def convert_to_dict(self) -> dict:
+        out = {"query_id": self.query_id, "padding_mask": self.padding_mask}
+        out.update(self.features)
+        return out
+
+
+class SasRecValidationBatch(NamedTuple):
+    """Legacy tuple view of a validation batch (reference dataset.py:195)."""
+
+    query_id: torch.LongTensor
+    padding_mask: torch.BoolTensor
+    features: Dict[str, torch.Tensor]
+    ground_truth: torch.LongTensor
+    train: torch.LongTensor
+
+
+class SasRecTrainingDataset(torch.utils.data.Dataset):
+    """Next-item training samples: the sequence shifted by ``sequence_shift``
+    becomes the labels (reference dataset.py:43)."""
+
+    def __init__(
+        self,
+        sequential,
+        max_sequence_length: int,
+        sequence_shift: int = 1,
+        sliding_window_step=None,
+        label_feature_name=None,
+    ) -> None:
+        self._shift = sequence_shift
+        self._label_name = label_feature_name or sequential.schema.item_id_feature_name
+        if label_feature_name is not None:
+            feat = sequential.schema[label_feature_name]
+            if not (feat.is_cat and feat.is_seq):
+                raise ValueError("Label feature must be a categorical sequence")
+        self._seq_names = [n for n, f in sequential.schema.items() if f.is_seq]
+        self._inner = TorchSequentialDataset(
+            sequential,
+            max_sequence_length + sequence_shift,
+            sliding_window_step=sliding_window_step,
+        )
+
+    def __len__(self) -> int:
+        return len(self._inner)
+
+    def __getitem__(self, index: int) -> dict:
+        item = dict(self._inner[index])
+        labels = item[self._label_name][self._shift :]
+        labels_padding_mask = item["padding_mask"][self._shift :]
+        for name in self._seq_names:
+            item[name] = item[name][: -self._shift]
+        item["padding_mask"] = item["padding_mask"][: -self._shift]
+        item["labels"] = labels
+        item["labels_padding_mask"] = labels_padding_mask
+        return item
+
+
+class SasRecPredictionDataset(torch.utils.data.Dataset):
+    """Inference samples: the full (left-padded) history (reference
+    dataset.py:150)."""
+
+    def __init__(self, sequential, max_sequence_length: int, padding_value=None) -> None:
+        self._inner = TorchSequentialDataset(sequential, max_sequence_length)
+
+    def __len__(self) -> int:
+        return len(self._inner)
+
+    def __getitem__(self, index: int) -> dict:
+        return dict(self._inner[index])
+
+
+class SasRecValidationDataset(torch.utils.data.Dataset):
+    """Validation samples carrying ground_truth + train ids (reference
+    dataset.py:212)."""
+
+    def __init__(self, sequential, ground_truth, train, max_sequence_length: int, label_feature_name=None) -> None:
+        self._inner = TorchSequentialValidationDataset(
+            sequential, ground_truth=ground_truth, train=train,
+            max_sequence_length=max_sequence_length, label_feature_name=label_feature_name,
+        )
+
+    def __len__(self) -> int:
+        return len(self._inner)
+
+    def __getitem__(self, index: int) -> dict:
+        return dict(self._inner[index])

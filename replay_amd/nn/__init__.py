@@ -1,3 +1,4 @@
+from replay_amd.utils import TORCH_AVAILABLE
 from .agg import ConcatAggregator, SumAggregator
 from .attention import MultiheadAttention, MultiHeadDifferentialAttention
 from .embedding import (
@@ -12,6 +13,7 @@ from .head import EmbeddingTyingHead
 from .mask import DefaultAttentionMask
 
 __all__ = [
+    "TORCH_AVAILABLE",
     "ConcatAggregator",
     "SumAggregator",
     "MultiheadAttention",

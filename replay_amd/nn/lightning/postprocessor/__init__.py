@@ -1,3 +1,5 @@
 from .seen_items import BasePostProcessor, SampleItemsFilter, SeenItemsFilter
 
-__all__ = ["BasePostProcessor", "SampleItemsFilter", "SeenItemsFilter"]
+PostprocessorBase = BasePostProcessor  # reference name
+
+__all__ = ["BasePostProcessor", "PostprocessorBase", "SampleItemsFilter", "SeenItemsFilter"]

@@ -8,11 +8,25 @@ bound to the model head (reference nn/sequential/sasrec/model.py:195).
 
 from __future__ import annotations
 
-from typing import Callable, Optional
+from typing import Callable, Optional, Protocol, runtime_checkable
 
 import torch
 
 LogitsCallback = Callable[..., torch.Tensor]
+
+
+@runtime_checkable
+class LossProto(Protocol):
+    """Structural protocol models program against (reference loss/base.py:9):
+    anything with a settable ``logits_callback`` and a loss-shaped
+    ``forward`` is a valid loss."""
+
+    @property
+    def logits_callback(self) -> LogitsCallback: ...
+
+    def set_logits_callback(self, callback: LogitsCallback) -> None: ...
+
+    def forward(self, *args, **kwargs) -> torch.Tensor: ...
 
 
 class LossBase(torch.nn.Module):

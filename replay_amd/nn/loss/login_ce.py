@@ -31,7 +31,10 @@ class LogInCE(SampledLossBase):
         weights: Optional[torch.Tensor] = None,
     ) -> torch.Tensor:
         if negative_labels is None:
-            raise ValueError("LogInCE requires negative_labels")
+            # in-batch negatives: the batch's own positives act as the
+            # negative pool (reference login_ce.py LogInCE vs LogInCESampled:
+            # the only difference is the negative set)
+            negative_labels = positive_labels[padding_mask].reshape(-1)
         pos, neg = self.get_sampled_logits(embeddings, positive_labels, negative_labels)
         logits = torch.cat([pos, neg], dim=-1).float() / self.temperature
         lse = torch.logsumexp(logits, dim=-1)
@@ -41,6 +44,17 @@ class LogInCE(SampledLossBase):
         if weights is not None:
             valid = valid * weights
         return (per_pos * valid).sum() / valid.sum().clamp(min=1e-12)
+
+
+class LogInCESampled(LogInCE):
+    """LogInCE over explicitly sampled negatives (reference login_ce.py:241;
+    ``LogInCE`` itself defaults to the in-batch pool when none are given)."""
+
+    def forward(self, *args, **kwargs) -> torch.Tensor:
+        negs = kwargs.get("negative_labels")
+        if negs is None and len(args) < 5:
+            raise ValueError("LogInCESampled requires negative_labels")
+        return super().forward(*args, **kwargs)
 
 
 class LogOutCE(SampledLossBase):

@@ -1,4 +1,7 @@
 from .model import ItemTower, QueryTower, TwoTower, TwoTowerBody, TwoTowerHead
-from .reader import FeaturesReader
+from .reader import FeaturesReader, FeaturesReaderProtocol
 
-__all__ = ["ItemTower", "QueryTower", "TwoTower", "TwoTowerBody", "TwoTowerHead", "FeaturesReader"]
+__all__ = [
+    "ItemTower", "QueryTower", "TwoTower", "TwoTowerBody", "TwoTowerHead",
+    "FeaturesReader", "FeaturesReaderProtocol",
+]

@@ -47,3 +47,16 @@ class FeaturesReader:
                     dense[ids] = values.astype(np.float32)
                 out[name] = torch.from_numpy(dense)
         return out
+
+
+from typing import Protocol, runtime_checkable
+
+
+@runtime_checkable
+class FeaturesReaderProtocol(Protocol):
+    """Structural protocol for item-feature readers (reference
+    twotower/reader.py): anything producing per-item feature tensors for the
+    item tower."""
+
+    def read(self, item_ids):  # pragma: no cover - protocol
+        ...

@@ -168,6 +168,25 @@ class TrimTransform(BatchTransform):
         return batch
 
 
+class AdaptiveTrimTransform(BatchTransform):
+    """Trim left-padded sequences to the batch's max valid length (reference
+    transform/trim.py:50) — an inference/validation speedup: the all-padding
+    prefix columns carry no information."""
+
+    def __init__(self, feature_names, padding_mask_name: str = "padding_mask") -> None:
+        super().__init__()
+        self.feature_names = [feature_names] if isinstance(feature_names, str) else list(feature_names)
+        self.padding_mask_name = padding_mask_name
+
+    def forward(self, batch: Batch) -> Batch:
+        mask = batch[self.padding_mask_name]
+        keep = int(mask.sum(-1).max().clamp(min=1))
+        for name in self.feature_names:
+            batch[name] = batch[name][:, -keep:]
+        batch[self.padding_mask_name] = mask[:, -keep:]
+        return batch
+
+
 class RenameTransform(BatchTransform):
     def __init__(self, mapping: Dict[str, str]) -> None:
         super().__init__()

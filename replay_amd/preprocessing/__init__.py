@@ -1,5 +1,10 @@
 from .converter import CSRConverter
-from .discretizer import Discretizer, GreedyDiscretizingRule, QuantileDiscretizingRule
+from .discretizer import (
+    Discretizer,
+    GreedyDiscretizingRule,
+    HandleInvalidStrategies,
+    QuantileDiscretizingRule,
+)
 from .filters import (
     ConsecutiveDuplicatesFilter,
     EntityDaysFilter,
@@ -11,8 +16,15 @@ from .filters import (
     QuantileItemsFilter,
     TimePeriodFilter,
 )
+from .history_based_fp import (
+    ConditionalPopularityProcessor,
+    EmptyFeatureProcessor,
+    HistoryBasedFeaturesProcessor,
+    LogStatFeaturesProcessor,
+)
 from .label_encoder import (
     LabelEncoder,
+    LabelEncoderPartialFitWarning,
     LabelEncoderTransformWarning,
     LabelEncodingRule,
     SequenceEncodingRule,
@@ -23,7 +35,12 @@ __all__ = [
     "CSRConverter",
     "Discretizer",
     "GreedyDiscretizingRule",
+    "HandleInvalidStrategies",
     "QuantileDiscretizingRule",
+    "ConditionalPopularityProcessor",
+    "EmptyFeatureProcessor",
+    "HistoryBasedFeaturesProcessor",
+    "LogStatFeaturesProcessor",
     "ConsecutiveDuplicatesFilter",
     "EntityDaysFilter",
     "GlobalDaysFilter",
@@ -34,6 +51,7 @@ __all__ = [
     "QuantileItemsFilter",
     "TimePeriodFilter",
     "LabelEncoder",
+    "LabelEncoderPartialFitWarning",
     "LabelEncoderTransformWarning",
     "LabelEncodingRule",
     "SequenceEncodingRule",

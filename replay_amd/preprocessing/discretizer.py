@@ -12,6 +12,9 @@ import numpy as np
 import pandas as pd
 
 
+HandleInvalidStrategies = ("error", "skip", "keep")  # reference discretizer.py:25
+
+
 class BaseDiscretizingRule:
     is_fitted = False
 

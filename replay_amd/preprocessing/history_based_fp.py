@@ -15,6 +15,17 @@ import numpy as np
 import pandas as pd
 
 
+class EmptyFeatureProcessor:
+    """No-op feature processor (reference history_based_fp.py:22): keeps the
+    two-stage pipeline shape when no feature engineering is wanted."""
+
+    def fit(self, log, features=None) -> "EmptyFeatureProcessor":
+        return self
+
+    def transform(self, log):
+        return log
+
+
 class LogStatFeaturesProcessor:
     def __init__(
         self,

@@ -19,6 +19,10 @@ import pandas as pd
 HandleUnknownStrategies = ("error", "use_default_value", "drop")
 
 
+class LabelEncoderPartialFitWarning(Warning):
+    """partial_fit saw already-known labels (reference label_encoder.py)."""
+
+
 class LabelEncoderTransformWarning(Warning):
     """Warning raised on unseen labels with non-error strategies."""
 

@@ -7,9 +7,26 @@ subsystems that exist in this stack: torch+ROCm (the compute tier), polars
 (optional tabular backend) and our HIP extension module.
 """
 
-from typing import Union
+from typing import Iterable, Union
 
 import pandas as pd
+
+
+class MissingImport:
+    """Placeholder type for gated-out backends (reference types.py:9)."""
+
+
+class FeatureUnavailableError(Exception):
+    """Raised when a gated subsystem is used without its dependency
+    (reference types.py:15)."""
+
+
+class FeatureUnavailableWarning(Warning):
+    """Warned when a gated subsystem degrades (reference types.py:19)."""
+
+
+IntOrList = Union[Iterable[int], int]
+NumType = Union[int, float]
 
 try:  # optional second tabular backend
     import polars as pl  # noqa: F401
@@ -57,3 +74,11 @@ def ROCM_AVAILABLE() -> bool:
 
 
 HIP_EXT_AVAILABLE = None  # resolved lazily by replay_amd.ops
+
+# subsystems the reference gates that are N/A or unconditionally absent in
+# the MI355X build (no JVM, no OpenVINO; ANN = our exact GPU brute force)
+ANN_AVAILABLE = TORCH_AVAILABLE  # brute-force GPU index needs torch only
+OPTUNA_AVAILABLE = False  # own search (models.optimization), no optuna dep
+OPENVINO_AVAILABLE = False  # compiled inference = torch.jit instead
+LIGHTFM_AVAILABLE = False
+OBP_AVAILABLE = False  # own IPS/SNIPS estimators (experimental.scenarios)

@@ -94,7 +94,9 @@ INVENTORY = {
     ],
     "replay_amd.experimental.scenarios": ["TwoStagesScenario"],
     "replay_amd.experimental.scenarios.obp_wrapper": ["OBPOfflinePolicyLearner", "ips_estimate", "snips_estimate"],
-    "replay_amd.experimental.preprocessing": ["Padder", "SequenceGenerator"],
+    "replay_amd.experimental.preprocessing": ["Padder", "SequenceGenerator", "DataPreparator", "Indexer"],
+    "replay_amd.experimental.metrics": ["Metric", "NCISMetric", "NCISPrecision", "HitRate", "NDCG", "MAP", "MRR", "Precision", "Recall", "RocAuc", "Coverage", "Surprisal", "Unexpectedness"],
+    "replay_amd.experimental.models": ["TorchRecommender"],
     "replay_amd.experimental.nn.data": ["TensorSchemaBuilder"],
     "replay_amd.experimental.utils.model_handler": ["save", "load"],
     "replay_amd.experimental.utils.session_handler": ["State"],
@@ -102,6 +104,18 @@ INVENTORY = {
     "replay_amd.data.utils": ["UniformBatching", "uniform_batch_count"],
     "replay_amd.data.utils.typing": ["torch_to_numpy", "numpy_to_torch", "numpy_to_pyarrow", "pyarrow_to_numpy"],
     "replay_amd.models.nn.optimizer_utils": ["OptimizerFactory", "LRSchedulerFactory", "FatOptimizerFactory", "FatLRSchedulerFactory"],
+    "replay_amd.models.nn.loss": ["SCEParams", "ScalableCrossEntropyLoss"],
+    "replay_amd.models.nn.sequential.sasrec": ["SasRec", "SasRecModel", "SasRecTrainingDataset", "SasRecTrainingBatch", "SasRecPredictionDataset", "SasRecPredictionBatch", "SasRecValidationDataset", "SasRecValidationBatch"],
+    "replay_amd.models.nn.sequential.callbacks": ["BasePredictionCallback", "PandasPredictionCallback", "TorchPredictionCallback", "QueryEmbeddingsPredictionCallback", "ValidationMetricsCallback"],
+    "replay_amd.data.nn.parquet.metadata": ["get_shape", "get_padding", "get_numeric_columns", "get_1d_array_columns", "get_2d_array_columns"],
+    "replay_amd.data": ["Dataset", "get_schema"],
+    "replay_amd.data.nn": ["TensorMap", "MutableTensorMap", "TorchSequentialBatch", "TorchSequentialValidationBatch", "PolarsSequentialDataset", "ParquetDataset", "ParquetModule", "DEFAULT_GROUND_TRUTH_PADDING_VALUE", "DEFAULT_TRAIN_PADDING_VALUE"],
+    "replay_amd.nn.loss": ["LossProto", "LogInCESampled", "LogOutCESampled", "LogOutCEWeighted"],
+    "replay_amd.nn.transform": ["AdaptiveTrimTransform"],
+    "replay_amd.nn.sequential": ["SasRec", "Bert4Rec", "TwoTower", "DiffTransformerLayer"],
+    "replay_amd.utils": ["MissingImport", "FeatureUnavailableError", "FeatureUnavailableWarning", "IntOrList", "NumType", "OPTUNA_AVAILABLE", "ANN_AVAILABLE", "OPENVINO_AVAILABLE"],
+    "replay_amd.preprocessing": ["EmptyFeatureProcessor", "HandleInvalidStrategies", "LabelEncoderPartialFitWarning", "HistoryBasedFeaturesProcessor", "ConditionalPopularityProcessor", "LogStatFeaturesProcessor"],
+    "replay_amd.models.optimization": ["IsOptimizible", "optimize_model"],
     "replay_amd.train": ["Trainer", "ModelCheckpoint", "EarlyStopping"],
 }
 
