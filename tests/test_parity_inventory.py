@@ -10,13 +10,14 @@ pytestmark = pytest.mark.core
 INVENTORY = {
     # layer 0/1: utils + data
     "replay_amd.utils.types": ["DataFrameLike", "PYSPARK_AVAILABLE", "TORCH_AVAILABLE", "POLARS_AVAILABLE"],
+    "replay_amd.utils": ["MissingImport", "FeatureUnavailableError", "FeatureUnavailableWarning", "IntOrList", "NumType", "OPTUNA_AVAILABLE", "ANN_AVAILABLE", "OPENVINO_AVAILABLE"],
     "replay_amd.utils.session_handler": ["State", "get_session", "logger_with_settings"],
     "replay_amd.utils.model_handler": ["save", "load", "save_encoder", "load_encoder", "save_splitter", "load_splitter", "save_to_replay", "load_from_replay"],
     "replay_amd.utils.pandas_utils": ["get_top_k_recs", "filter_cold", "fallback"],
     "replay_amd.utils.time": ["smoothe_time", "get_item_recency"],
     "replay_amd.utils.distributions": ["item_distribution"],
     "replay_amd.utils.dataframe_bucketizer": ["DataframeBucketizer"],
-    "replay_amd.data": ["Dataset", "FeatureHint", "FeatureInfo", "FeatureSchema", "FeatureSource", "FeatureType"],
+    "replay_amd.data": ["Dataset", "FeatureHint", "FeatureInfo", "FeatureSchema", "FeatureSource", "FeatureType", "get_schema"],
     "replay_amd.data.dataset_utils": ["DatasetLabelEncoder"],
     # layer 2: preprocessing
     "replay_amd.preprocessing": [
@@ -24,7 +25,9 @@ INVENTORY = {
         "GreedyDiscretizingRule", "QuantileDiscretizingRule", "Sessionizer", "CSRConverter",
         "InteractionEntriesFilter", "MinCountFilter", "LowRatingFilter", "NumInteractionsFilter",
         "EntityDaysFilter", "GlobalDaysFilter", "TimePeriodFilter", "QuantileItemsFilter",
-        "ConsecutiveDuplicatesFilter",
+        "ConsecutiveDuplicatesFilter", "EmptyFeatureProcessor", "HandleInvalidStrategies",
+        "LabelEncoderPartialFitWarning", "HistoryBasedFeaturesProcessor",
+        "ConditionalPopularityProcessor", "LogStatFeaturesProcessor",
     ],
     "replay_amd.preprocessing.history_based_fp": ["LogStatFeaturesProcessor", "ConditionalPopularityProcessor", "HistoryBasedFeaturesProcessor"],
     # layer 3: splitters
@@ -45,14 +48,17 @@ INVENTORY = {
         "AssociationRulesItemRec", "Word2VecRec", "PopRec", "QueryPopRec", "RandomRec",
         "CatPopRec", "Wilson", "UCB", "KLUCB", "ThompsonSampling", "LinUCB", "ClusterRec",
     ],
-    "replay_amd.models.optimization": ["optimize_model"],
+    "replay_amd.models.optimization": ["optimize_model", "IsOptimizible"],
     "replay_amd.models.extensions.ann": ["ANNMixin", "BruteForceIndex", "IndexParams"],
     "replay_amd.scenarios": ["Fallback"],
     # layer 5: tensor data
     "replay_amd.data.nn": [
         "TensorSchema", "TensorFeatureInfo", "TensorFeatureSource", "SequenceTokenizer",
         "SequentialDataset", "PandasSequentialDataset", "TorchSequentialDataset",
-        "TorchSequentialValidationDataset",
+        "TorchSequentialValidationDataset", "TensorMap", "MutableTensorMap",
+        "TorchSequentialBatch", "TorchSequentialValidationBatch", "PolarsSequentialDataset",
+        "ParquetDataset", "ParquetModule", "DEFAULT_GROUND_TRUTH_PADDING_VALUE",
+        "DEFAULT_TRAIN_PADDING_VALUE",
     ],
     "replay_amd.data.nn.parquet": [
         "ParquetDataset", "ParquetModule", "FixedBatchSizeDataset", "Partitioning",
@@ -66,12 +72,13 @@ INVENTORY = {
         "MultiHeadDifferentialAttention", "PointWiseFeedForward", "SwiGLU", "SwiGLUEncoder",
         "EmbeddingTyingHead",
     ],
-    "replay_amd.nn.loss": ["CE", "CEWeighted", "CESampled", "CESampledWeighted", "BCE", "BCESampled", "LogInCE", "LogOutCE", "ScalableCrossEntropyLoss"],
+    "replay_amd.nn.loss": ["CE", "CEWeighted", "CESampled", "CESampledWeighted", "BCE", "BCESampled", "LogInCE", "LogInCESampled", "LogOutCE", "LogOutCESampled", "LogOutCEWeighted", "LossProto", "ScalableCrossEntropyLoss"],
     "replay_amd.nn.transform": [
         "NextTokenTransform", "UniformNegativeSamplingTransform", "MultiClassNegativeSamplingTransform",
         "TokenMaskTransform", "SequenceRollTransform", "TrimTransform", "RenameTransform",
         "GroupTransform", "SelectTransform", "UnsqueezeTransform", "EqualityMaskTransform",
-        "CopyTransform", "make_default_sasrec_transforms", "make_default_twotower_transforms",
+        "CopyTransform", "AdaptiveTrimTransform",
+        "make_default_sasrec_transforms", "make_default_twotower_transforms",
     ],
     "replay_amd.nn.sequential.sasrec": ["SasRec", "SasRecBody", "SasRecTransformerLayer", "PositionAwareAggregator"],
     "replay_amd.nn.sequential.sasrec.diff_transformer": ["DiffTransformerBlock", "DiffTransformerLayer"],
@@ -83,7 +90,6 @@ INVENTORY = {
         "SparkTopItemsCallback", "TorchTopItemsCallback", "QueryEmbeddingsPredictionCallback",
         "SeenItemsFilter", "SampleItemsFilter", "BasePostProcessor",
     ],
-    "replay_amd.train": ["Trainer"],
     "replay_amd.parallel": ["gather_ids", "gather_embeddings"],
     # layer 8: legacy nn
     "replay_amd.models.nn": ["SasRec", "Bert4Rec", "TiSasRec", "SasRecCompiled", "Bert4RecCompiled"],
@@ -91,12 +97,12 @@ INVENTORY = {
     "replay_amd.experimental.models": [
         "ADMMSLIM", "CQL", "DDPG", "DT4Rec", "HierarchicalRecommender", "ImplicitWrap",
         "MultVAE", "NeuralTS", "NeuroMF", "ScalaALSWrap", "ULinUCB", "LightFMWrap",
+        "TorchRecommender",
     ],
     "replay_amd.experimental.scenarios": ["TwoStagesScenario"],
     "replay_amd.experimental.scenarios.obp_wrapper": ["OBPOfflinePolicyLearner", "ips_estimate", "snips_estimate"],
     "replay_amd.experimental.preprocessing": ["Padder", "SequenceGenerator", "DataPreparator", "Indexer"],
     "replay_amd.experimental.metrics": ["Metric", "NCISMetric", "NCISPrecision", "HitRate", "NDCG", "MAP", "MRR", "Precision", "Recall", "RocAuc", "Coverage", "Surprisal", "Unexpectedness"],
-    "replay_amd.experimental.models": ["TorchRecommender"],
     "replay_amd.experimental.nn.data": ["TensorSchemaBuilder"],
     "replay_amd.experimental.utils.model_handler": ["save", "load"],
     "replay_amd.experimental.utils.session_handler": ["State"],
@@ -108,14 +114,7 @@ INVENTORY = {
     "replay_amd.models.nn.sequential.sasrec": ["SasRec", "SasRecModel", "SasRecTrainingDataset", "SasRecTrainingBatch", "SasRecPredictionDataset", "SasRecPredictionBatch", "SasRecValidationDataset", "SasRecValidationBatch"],
     "replay_amd.models.nn.sequential.callbacks": ["BasePredictionCallback", "PandasPredictionCallback", "TorchPredictionCallback", "QueryEmbeddingsPredictionCallback", "ValidationMetricsCallback"],
     "replay_amd.data.nn.parquet.metadata": ["get_shape", "get_padding", "get_numeric_columns", "get_1d_array_columns", "get_2d_array_columns"],
-    "replay_amd.data": ["Dataset", "get_schema"],
-    "replay_amd.data.nn": ["TensorMap", "MutableTensorMap", "TorchSequentialBatch", "TorchSequentialValidationBatch", "PolarsSequentialDataset", "ParquetDataset", "ParquetModule", "DEFAULT_GROUND_TRUTH_PADDING_VALUE", "DEFAULT_TRAIN_PADDING_VALUE"],
-    "replay_amd.nn.loss": ["LossProto", "LogInCESampled", "LogOutCESampled", "LogOutCEWeighted"],
-    "replay_amd.nn.transform": ["AdaptiveTrimTransform"],
     "replay_amd.nn.sequential": ["SasRec", "Bert4Rec", "TwoTower", "DiffTransformerLayer"],
-    "replay_amd.utils": ["MissingImport", "FeatureUnavailableError", "FeatureUnavailableWarning", "IntOrList", "NumType", "OPTUNA_AVAILABLE", "ANN_AVAILABLE", "OPENVINO_AVAILABLE"],
-    "replay_amd.preprocessing": ["EmptyFeatureProcessor", "HandleInvalidStrategies", "LabelEncoderPartialFitWarning", "HistoryBasedFeaturesProcessor", "ConditionalPopularityProcessor", "LogStatFeaturesProcessor"],
-    "replay_amd.models.optimization": ["IsOptimizible", "optimize_model"],
     "replay_amd.train": ["Trainer", "ModelCheckpoint", "EarlyStopping"],
 }
 
