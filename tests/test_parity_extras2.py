@@ -183,3 +183,53 @@ class TestLossZooAdditions:
         loss = CE()
         loss.set_logits_callback(lambda *a, **k: None)  # property must resolve
         assert isinstance(loss, LossProto)
+
+
+class TestReferenceDoctestParity:
+    """Exact-value reproduction of reference docstring examples."""
+
+    def test_time_smoothing_values(self):
+        from replay_amd.utils.time import get_item_recency, smoothe_time
+
+        df = pd.DataFrame(
+            {
+                "item_idx": [1, 1, 2, 3, 3],
+                "timestamp": ["2099-03-19", "2099-03-20", "2099-03-22", "2099-03-27", "2099-03-25"],
+                "relevance": [1, 1, 1, 1, 1],
+            }
+        )
+        expected = {
+            "power": [0.6632, 0.7204, 1.0],
+            "exp": [0.8606, 0.9117, 1.0],
+            "linear": [0.8917, 0.9333, 1.0],
+        }
+        for kind, want in expected.items():
+            out = get_item_recency(df, kind=kind, item_column="item_idx").sort_values("item_idx")
+            assert [round(v, 4) for v in out["relevance"]] == want, kind
+        # smoothe_time multiplies the existing relevance
+        d2 = pd.DataFrame(
+            {"item_idx": [1, 2, 3], "timestamp": ["2099-03-19", "2099-03-20", "2099-03-22"],
+             "relevance": [10, 3, 0.1]}
+        )
+        got = [round(v, 4) for v in smoothe_time(d2).sort_values("timestamp")["relevance"]]
+        assert got == [9.3303, 2.8645, 0.1]
+
+    def test_sessionizer_partition_matches_reference(self):
+        from replay_amd.preprocessing import Sessionizer
+
+        df = pd.DataFrame(
+            {
+                "user_id": [1, 1, 1, 2, 2, 2, 3, 3, 3, 3],
+                "item_id": [3, 7, 10, 5, 8, 11, 4, 9, 2, 5],
+                "timestamp": [1, 2, 3, 3, 2, 1, 3, 12, 1, 4],
+            }
+        )
+        out = Sessionizer(session_gap=5).transform(df)
+        # session ids are opaque labels; the PARTITION must match the
+        # reference doctest: each user one session except user3's ts=12 row
+        def partition(frame, col):
+            return {tuple(sorted(g.index)) for _, g in frame.groupby(["user_id", col])}
+
+        ref_ids = pd.Series([2, 2, 2, 5, 5, 5, 9, 8, 9, 9])
+        want = {tuple(sorted(g.index)) for _, g in df.assign(s=ref_ids).groupby(["user_id", "s"])}
+        assert partition(out, "session_id") == want
