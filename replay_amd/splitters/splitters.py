@@ -44,13 +44,20 @@ class RatioSplitter(Splitter):
     def _core_split(self, interactions: pd.DataFrame) -> SplitterReturnType:
         df = interactions.sort_values([self.divide_column, self.timestamp_column], kind="stable")
         sizes = df.groupby(self.divide_column)[self.divide_column].transform("size")
-        pos = df.groupby(self.divide_column).cumcount()
+        row_num = df.groupby(self.divide_column).cumcount() + 1  # 1-based in time order
         if self.split_by_fractions:
-            n_test = np.floor(sizes * self.test_size).astype(int)
+            # reference ratio_splitter.py:215: frac = round(row/count, 3),
+            # test rows are frac > round(1 - ratio, 3)
+            frac = (row_num / sizes).round(3)
+            is_test = frac > round(1.0 - self.test_size, 3)
         else:
-            n_test = np.minimum(sizes - 1, int(np.ceil(self.test_size)))
-        is_test = pos >= (sizes - n_test)
+            # non-fraction mode: n_train = half-up-round(count * (1 - ratio));
+            # this puts one more interaction per group into train than the
+            # fraction mode (reference docstring)
+            n_train = np.floor(sizes * (1.0 - self.test_size) + 0.5)
+            is_test = row_num > n_train
         if self.min_interactions_per_group is not None:
+            # undersized groups go entirely to train (reference :214)
             is_test &= sizes >= self.min_interactions_per_group
         is_test = self._recalculate_with_session_id_column(df, is_test)
         return df[~is_test].sort_index(), df[is_test].sort_index()

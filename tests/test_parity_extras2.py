@@ -233,3 +233,53 @@ class TestReferenceDoctestParity:
         ref_ids = pd.Series([2, 2, 2, 5, 5, 5, 9, 8, 9, 9])
         want = {tuple(sorted(g.index)) for _, g in df.assign(s=ref_ids).groupby(["user_id", "s"])}
         assert partition(out, "session_id") == want
+
+    def test_filters_match_reference_doctests(self):
+        from datetime import datetime
+
+        from replay_amd.preprocessing import EntityDaysFilter, TimePeriodFilter
+
+        log = pd.DataFrame(
+            {
+                "user_id": ["u1", "u2", "u2", "u3", "u3", "u3"],
+                "item_id": ["i1", "i2", "i3", "i1", "i2", "i3"],
+                "rating": [1.0, 0.5, 3, 1, 0, 1],
+                "timestamp": pd.to_datetime(
+                    ["2020-01-01 23:59:59", "2020-02-01 00:00:00", "2020-02-01 00:00:01",
+                     "2020-01-01 00:04:15", "2020-01-02 00:04:14", "2020-01-05 23:59:59"]
+                ),
+            }
+        )
+        out = TimePeriodFilter(
+            start_date="2020-01-01 14:00:00", end_date=datetime(2020, 1, 3)
+        ).transform(log)
+        assert sorted(map(tuple, out[["user_id", "item_id"]].to_numpy())) == [("u1", "i1"), ("u3", "i2")]
+        o1 = EntityDaysFilter(1, True, entity_column="user_id").transform(log)
+        assert sorted(map(tuple, o1[["user_id", "item_id"]].to_numpy())) == [
+            ("u1", "i1"), ("u2", "i2"), ("u2", "i3"), ("u3", "i1"), ("u3", "i2")
+        ]
+        o2 = EntityDaysFilter(1, False, entity_column="item_id").transform(log)
+        assert sorted(map(tuple, o2[["user_id", "item_id"]].to_numpy())) == [
+            ("u1", "i1"), ("u2", "i2"), ("u2", "i3"), ("u3", "i1")
+        ]
+
+    def test_splitters_match_reference_doctests(self):
+        from replay_amd.splitters import LastNSplitter, RatioSplitter
+
+        data = [(1, 1, "01-01-2020"), (1, 2, "02-01-2020"), (1, 3, "03-01-2020"),
+                (1, 4, "04-01-2020"), (1, 5, "05-01-2020"), (2, 1, "06-01-2020"),
+                (2, 2, "07-01-2020"), (2, 3, "08-01-2020"), (2, 9, "09-01-2020"),
+                (2, 10, "10-01-2020"), (3, 1, "01-01-2020"), (3, 5, "02-01-2020"),
+                (3, 3, "03-01-2020"), (3, 1, "04-01-2020"), (3, 2, "05-01-2020")]
+        df = pd.DataFrame(data, columns=["query_id", "item_id", "timestamp"])
+        df["timestamp"] = pd.to_datetime(df["timestamp"], format="%d-%m-%Y")
+        tr, te = LastNSplitter(N=2, divide_column="query_id", query_column="query_id").split(df)
+        assert sorted(tr.index) == [0, 1, 2, 5, 6, 7, 10, 11, 12]
+        assert sorted(te.index) == [3, 4, 8, 9, 13, 14]
+        tr, te = RatioSplitter(test_size=0.5, divide_column="query_id", query_column="query_id").split(df)
+        assert sorted(tr.index) == [0, 1, 5, 6, 10, 11]  # fraction mode: frac > 0.5 -> test
+        assert sorted(te.index) == [2, 3, 4, 7, 8, 9, 12, 13, 14]
+        tr, te = RatioSplitter(
+            test_size=0.5, divide_column="query_id", query_column="query_id", split_by_fractions=False
+        ).split(df)
+        assert tr.groupby("query_id").size().tolist() == [3, 3, 3]  # one more per group in train
