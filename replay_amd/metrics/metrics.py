@@ -219,7 +219,9 @@ class Unexpectedness(Metric):
                 if not topk_items:
                     values.append(0.0)
                     continue
-                values.append(sum(1 for p in topk_items if p not in base_set) / len(topk_items))
+                # reference unexpectedness.py:156 divides by K (not by the
+                # number of produced recs): short lists count as unexpected
+                values.append(1.0 - len(set(topk_items) & base_set) / k)
             result[f"{self.__name__}@{k}"] = self._mode.cpu(np.asarray(values, dtype=np.float64))
         return result
 

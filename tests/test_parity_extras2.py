@@ -327,3 +327,21 @@ class TestReferenceDoctestParity:
         assert list(NDCG(2, mode=ConfidenceInterval(alpha=0.95))(recs, gt).values())[0] == pytest.approx(
             0.3508565839953337
         )
+
+    def test_unexpectedness_matches_reference_doctest(self):
+        from replay_amd.metrics import Unexpectedness
+
+        recs = pd.DataFrame(
+            {
+                "query_id": [1, 1, 1, 1, 1, 2, 2, 2, 2, 2, 3, 3, 3],
+                "item_id": [3, 7, 10, 11, 2, 5, 8, 11, 1, 3, 4, 9, 2],
+                "rating": [0.6, 0.5, 0.4, 0.3, 0.2, 0.6, 0.5, 0.4, 0.3, 0.2, 1.0, 0.5, 0.1],
+            }
+        )
+        base = pd.DataFrame(
+            {"query_id": [1, 1, 1, 2, 2, 2, 3, 3], "item_id": [3, 7, 2, 5, 8, 3, 4, 9],
+             "rating": [0.5, 0.5, 0.7, 0.6, 0.6, 0.3, 1.0, 0.5]}
+        )
+        out = Unexpectedness([2, 4])(recs, base)
+        assert out["Unexpectedness@2"] == pytest.approx(0.16666666666666666)
+        assert out["Unexpectedness@4"] == pytest.approx(0.5)  # divides by K, not len(pred)
