@@ -251,6 +251,8 @@ class CategoricalDiversity(Metric):
                 if not topk_cats:
                     values.append(0.0)
                     continue
-                values.append(len(set(topk_cats)) / len(topk_cats))
+                # reference categorical_diversity.py divides by K: fewer
+                # than K recommendations lowers diversity
+                values.append(len(set(topk_cats)) / k)
             result[f"{self.__name__}@{k}"] = self._mode.cpu(np.asarray(values, dtype=np.float64))
         return result
