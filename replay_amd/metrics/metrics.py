@@ -164,8 +164,8 @@ class Novelty(_TrainAwareMetric):
             for q, pred in recs.items():
                 seen = set(train.get(q, []))
                 topk_items = pred[:k]
-                if not topk_items:
-                    values.append(0.0)
+                if not topk_items or not seen:
+                    values.append(1.0)  # reference novelty.py:144: vacuously novel
                     continue
                 values.append(sum(1 for p in topk_items if p not in seen) / len(topk_items))
             agg = self._mode.cpu(np.asarray(values, dtype=np.float64))
