@@ -10,6 +10,7 @@ Implementation is pandas/numpy-native (no Spark tier).
 from __future__ import annotations
 
 import json
+import warnings
 from pathlib import Path
 from typing import Dict, List, Optional, Sequence, Union
 
@@ -78,7 +79,13 @@ class LabelEncodingRule:
 
     def fit(self, df: pd.DataFrame) -> "LabelEncodingRule":
         if self._mapping is None:
-            uniques = pd.unique(df[self._col])
+            # reference label_encoder.py:188: unique values are SORTED before
+            # code assignment so mappings are order-independent (mixed-type
+            # columns, which plain sorting cannot order, fall back to repr)
+            try:
+                uniques = df[self._col].sort_values().drop_duplicates().tolist()
+            except TypeError:
+                uniques = sorted(set(df[self._col].tolist()), key=repr)
             self._mapping = {value: idx for idx, value in enumerate(uniques)}
             self._rebuild_inverse()
         return self
@@ -86,12 +93,18 @@ class LabelEncodingRule:
     def partial_fit(self, df: pd.DataFrame) -> "LabelEncodingRule":
         if self._mapping is None:
             return self.fit(df)
-        uniques = pd.unique(df[self._col])
+        new_values = set(df[self._col].tolist()) - set(self._mapping)
+        if not new_values:
+            warnings.warn(
+                "partial_fit will have no effect because there are no new "
+                f"values in the incoming dataset at '{self.column}' column",
+                LabelEncoderPartialFitWarning,
+            )
+            return self
         next_code = len(self._mapping)
-        for value in uniques:
-            if value not in self._mapping:
-                self._mapping[value] = next_code
-                next_code += 1
+        for value in sorted(new_values, key=repr):
+            self._mapping[value] = next_code
+            next_code += 1
         self._rebuild_inverse()
         return self
 
@@ -189,13 +202,14 @@ class SequenceEncodingRule(LabelEncodingRule):
 
     def fit(self, df: pd.DataFrame) -> "SequenceEncodingRule":
         if self._mapping is None:
-            self._mapping = {}
-            code = 0
-            for seq in df[self._col]:
-                for value in seq:
-                    if value not in self._mapping:
-                        self._mapping[value] = code
-                        code += 1
+            # reference convention: exploded unique values sorted before
+            # code assignment (label_encoder.py:188 via fit on the explode)
+            values = {value for seq in df[self._col] for value in seq}
+            try:
+                uniques = sorted(values)
+            except TypeError:
+                uniques = sorted(values, key=repr)
+            self._mapping = {value: idx for idx, value in enumerate(uniques)}
             self._rebuild_inverse()
         return self
 

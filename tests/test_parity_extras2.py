@@ -359,3 +359,31 @@ class TestReferenceDoctestParity:
         out = CategoricalDiversity([3, 5])(recs)
         assert out["CategoricalDiversity@3"] == pytest.approx(1.0)
         assert out["CategoricalDiversity@5"] == pytest.approx(0.8666666666666667)
+
+    def test_label_encoder_matches_reference_doctest(self):
+        """Exact code assignment from the reference LabelEncoder docstring:
+        unique values are sorted before codes are assigned."""
+        import warnings as _w
+
+        from replay_amd.preprocessing import (
+            LabelEncoder,
+            LabelEncoderPartialFitWarning,
+            LabelEncodingRule,
+            SequenceEncodingRule,
+        )
+
+        df = pd.DataFrame(
+            [("u1", "item_1", [1, 2, 3]), ("u2", "item_2", [3, 4, 5]), ("u3", "item_3", [-1, -2, 4])],
+            columns=["user_id", "item_1", "list"],
+        )
+        enc = LabelEncoder(
+            [LabelEncodingRule("user_id"), LabelEncodingRule("item_1"), SequenceEncodingRule("list")]
+        )
+        out = enc.fit_transform(df)
+        assert enc.mapping["user_id"] == {"u1": 0, "u2": 1, "u3": 2}
+        assert enc.mapping["list"] == {-2: 0, -1: 1, 1: 2, 2: 3, 3: 4, 4: 5, 5: 6}
+        assert [list(x) for x in out["list"]] == [[2, 3, 4], [4, 5, 6], [1, 0, 5]]
+        back = enc.inverse_transform(out)
+        assert [list(x) for x in back["list"]] == [[1, 2, 3], [3, 4, 5], [-1, -2, 4]]
+        with pytest.warns(LabelEncoderPartialFitWarning):
+            LabelEncodingRule("user_id").fit(df).partial_fit(df)
