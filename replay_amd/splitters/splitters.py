@@ -161,11 +161,23 @@ class NewUsersSplitter(Splitter):
         self.test_size = test_size
 
     def _core_split(self, interactions: pd.DataFrame) -> SplitterReturnType:
-        first_ts = interactions.groupby(self.query_column)[self.timestamp_column].transform("min")
-        threshold = first_ts.quantile(1 - self.test_size)
-        is_test = first_ts > threshold
+        # reference new_users_splitter.py:99: test_start is the LATEST first-
+        # interaction date at which the cumulative (newest-first) user count
+        # reaches test_size of all users; users starting at/after it go to
+        # test WITH THEIR WHOLE HISTORY, and train keeps only interactions
+        # strictly BEFORE test_start (old users' later rows are dropped)
+        first_ts = interactions.groupby(self.query_column)[self.timestamp_column].min()
+        by_date = (
+            first_ts.value_counts().rename("n").sort_index(ascending=False).to_frame()
+        )
+        by_date["cum"] = by_date["n"].cumsum()
+        eligible = by_date[by_date["cum"] >= self.test_size * by_date["n"].sum()]
+        test_start = eligible.index.max()
+        new_users = set(first_ts[first_ts >= test_start].index)
+        is_test = interactions[self.query_column].isin(new_users)
+        in_train = interactions[self.timestamp_column] < test_start
         is_test = self._recalculate_with_session_id_column(interactions, is_test)
-        return interactions[~is_test], interactions[is_test]
+        return interactions[~is_test & in_train], interactions[is_test]
 
 
 class ColdUserRandomSplitter(Splitter):

@@ -387,3 +387,36 @@ class TestReferenceDoctestParity:
         assert [list(x) for x in back["list"]] == [[1, 2, 3], [3, 4, 5], [-1, -2, 4]]
         with pytest.warns(LabelEncoderPartialFitWarning):
             LabelEncodingRule("user_id").fit(df).partial_fit(df)
+
+    def test_new_users_splitter_matches_reference_doctest(self):
+        from replay_amd.splitters import NewUsersSplitter
+
+        df = pd.DataFrame(
+            {"query_id": [1, 1, 2, 2, 3, 4], "item_id": [1, 2, 3, 1, 2, 3],
+             "relevance": [1, 2, 3, 4, 5, 6], "timestamp": [20, 40, 20, 30, 10, 40]}
+        )
+        tr, te = NewUsersSplitter(test_size=0.1).split(df)
+        assert sorted(tr.index) == [0, 2, 3, 4]  # old users keep only pre-threshold rows
+        assert sorted(te.index) == [5]
+        tr, _ = NewUsersSplitter(test_size=0.3).split(df)
+        assert sorted(tr.index) == [4]
+
+    def test_csr_converter_matches_reference_doctest(self):
+        from replay_amd.preprocessing import CSRConverter
+
+        df = pd.DataFrame(
+            {"user_id": [1, 1, 1, 2, 2, 2, 3, 3, 3, 3],
+             "item_id": [3, 7, 10, 5, 8, 11, 4, 9, 2, 5],
+             "rating": [1, 2, 3, 3, 2, 1, 3, 12, 1, 4]}
+        )
+        m = CSRConverter(
+            first_dim_column="user_id", second_dim_column="item_id", data_column="rating"
+        ).transform(df)
+        want = np.array(
+            [[0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0],
+             [0, 0, 0, 1, 0, 0, 0, 2, 0, 0, 3, 0],
+             [0, 0, 0, 0, 0, 3, 0, 0, 2, 0, 0, 1],
+             [0, 0, 1, 0, 3, 4, 0, 0, 0, 12, 0, 0]]
+        )
+        assert m.shape == want.shape
+        assert (np.asarray(m.todense()) == want).all()
