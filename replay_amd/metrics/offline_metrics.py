@@ -65,7 +65,11 @@ class OfflineMetrics:
                 ) and all(isinstance(v, pd.DataFrame) for v in base_recommendations.values()):
                     for name, base in base_recommendations.items():
                         out = metric(recommendations, base)
-                        result.update({f"{key}_{name}": value for key, value in out.items()})
+                        # reference key layout: Metric_MODEL@k
+                        result.update({
+                            f"{key.split('@')[0]}_{name}@{key.split('@')[1]}": value
+                            for key, value in out.items()
+                        })
                 else:
                     result.update(metric(recommendations, base_recommendations))
             elif isinstance(metric, _TrainAwareMetric):
