@@ -177,6 +177,78 @@ def serve_bench(args, device, rank, world) -> None:
         )
 
 
+def itemknn_bench(args) -> None:
+    """BASELINE config 1: ItemKNN fit/predict on an ML-1M-shape synthetic log
+    (pandas/CPU plumbing path; no GPU involved)."""
+    import numpy as np
+    import pandas as pd
+
+    from replay_amd.data import Dataset, FeatureHint, FeatureInfo, FeatureSchema, FeatureType
+    from replay_amd.models import ItemKNN
+
+    rng = np.random.default_rng(0)
+    n_users, n_items, n_inter = 6040, 3706, 1_000_000
+    # moderate popularity skew so ~1M (user, item) pairs stay distinct
+    pop = 1.0 / (np.arange(n_items) + 30.0)
+    draw = int(n_inter * 1.7)
+    df = (
+        pd.DataFrame(
+            {
+                "query_id": rng.integers(0, n_users, draw),
+                "item_id": rng.choice(n_items, draw, p=pop / pop.sum()),
+                "rating": rng.integers(1, 6, draw).astype(float),
+                "timestamp": rng.integers(0, 10_000_000, draw),
+            }
+        )
+        .drop_duplicates(["query_id", "item_id"])
+        .head(n_inter)
+        .reset_index(drop=True)
+    )
+    schema = FeatureSchema(
+        [
+            FeatureInfo("query_id", FeatureType.CATEGORICAL, FeatureHint.QUERY_ID),
+            FeatureInfo("item_id", FeatureType.CATEGORICAL, FeatureHint.ITEM_ID),
+            FeatureInfo("rating", FeatureType.NUMERICAL, FeatureHint.RATING),
+            FeatureInfo("timestamp", FeatureType.NUMERICAL, FeatureHint.TIMESTAMP),
+        ]
+    )
+    ds = Dataset(feature_schema=schema, interactions=df, categorical_encoded=True)
+    model = ItemKNN(num_neighbours=100)
+    t0 = time.perf_counter()
+    model.fit(ds)
+    fit_s = time.perf_counter() - t0
+    t0 = time.perf_counter()
+    recs = model.predict(ds, k=10)
+    predict_s = time.perf_counter() - t0
+    n_queries = ds.interactions["query_id"].nunique()
+    print(
+        json.dumps(
+            {
+                "metric": "ItemKNN fit interactions/sec + predict recs/sec@10",
+                "value": len(df) / fit_s,
+                "unit": "interactions/sec (fit)",
+                "n_gpus": 0,
+                "steps": 1,
+                "warmup": 0,
+                "ms_per_step": fit_s * 1000,
+                "higher_is_better": True,
+                "scaling": "strong",
+                "vs_baseline": None,
+                "dtype": "fp64",
+                "data": "synthetic ML-1M shape",
+                "config": {
+                    "model": "itemknn_k100",
+                    "n_interactions": len(df),
+                    "n_queries": int(n_queries),
+                    "predict_recs_per_sec": n_queries * 10 / predict_s,
+                    "predict_s": predict_s,
+                    "n_recs": len(recs),
+                },
+            }
+        )
+    )
+
+
 def main() -> None:
     parser = argparse.ArgumentParser()
     parser.add_argument("--gpus", type=int, default=1)
@@ -186,7 +258,7 @@ def main() -> None:
     parser.add_argument("--lr", type=float, default=1e-3)
     parser.add_argument("--tunableop", action="store_true", help="(default on)")
     parser.add_argument("--no-tunableop", action="store_true", help="disable rocBLAS TunableOp")
-    parser.add_argument("--mode", choices=["train", "serve"], default="train")
+    parser.add_argument("--mode", choices=["train", "serve", "itemknn"], default="train")
     parser.add_argument(
         "--graphs",
         action="store_true",
@@ -210,6 +282,9 @@ def main() -> None:
         torch.cuda.set_device(device)
     torch.manual_seed(1234 + rank)
 
+    if args.mode == "itemknn":
+        itemknn_bench(args)
+        return
     if args.mode == "serve":
         serve_bench(args, device, rank, world)
         if world > 1:
