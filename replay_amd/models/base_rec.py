@@ -174,17 +174,41 @@ class BaseRecommender(IsSavable):
 
     def _filter_seen(self, recs: pd.DataFrame, interactions: pd.DataFrame, queries: pd.DataFrame) -> pd.DataFrame:
         """Anti-join recommendations against the query's seen items
-        (reference base_rec.py:152-201)."""
-        seen = interactions.merge(queries, on=self.query_column)[[self.query_column, self.item_column]]
-        merged = recs.merge(
-            seen.assign(__seen=True), on=[self.query_column, self.item_column], how="left"
-        )
+        (reference base_rec.py:152-201).  Integer-id frames take a numpy
+        combined-key isin (the pandas merge was the predict bottleneck for
+        dense-similarity models at ML-1M scale)."""
+        qc, ic = self.query_column, self.item_column
+        seen = interactions[interactions[qc].isin(set(queries[qc]))][[qc, ic]]
+        if (
+            pd.api.types.is_integer_dtype(recs[qc])
+            and pd.api.types.is_integer_dtype(recs[ic])
+            and pd.api.types.is_integer_dtype(seen[qc])
+            and pd.api.types.is_integer_dtype(seen[ic])
+            and len(recs)
+        ):
+            scale = int(max(recs[ic].max(), seen[ic].max() if len(seen) else 0)) + 1
+            rk = recs[qc].to_numpy(np.int64) * scale + recs[ic].to_numpy(np.int64)
+            sk = seen[qc].to_numpy(np.int64) * scale + seen[ic].to_numpy(np.int64)
+            return recs[~np.isin(rk, sk)]
+        merged = recs.merge(seen.assign(__seen=True), on=[qc, ic], how="left")
         return merged[merged["__seen"].isna()].drop(columns="__seen")
 
     @staticmethod
     def _get_top_k(recs: pd.DataFrame, query_column: str, rating_column: str, k: int) -> pd.DataFrame:
-        recs = recs.sort_values([query_column, rating_column], ascending=[True, False], kind="stable")
-        return recs.groupby(query_column, sort=False).head(k).reset_index(drop=True)
+        """Per-query top-k by rating: one stable numpy lexsort + rank-within-
+        group instead of pandas sort_values + groupby-head (several-x faster
+        on multi-million-row candidate frames)."""
+        if not len(recs):
+            return recs.reset_index(drop=True)
+        q_codes, _ = pd.factorize(recs[query_column], sort=True)
+        ratings = recs[rating_column].to_numpy(np.float64)
+        order = np.lexsort((-ratings, q_codes))  # stable: ties keep frame order
+        sorted_q = q_codes[order]
+        boundaries = np.flatnonzero(np.diff(sorted_q)) + 1
+        starts = np.r_[0, boundaries]
+        lengths = np.diff(np.r_[starts, len(sorted_q)])
+        ranks = np.arange(len(sorted_q)) - np.repeat(starts, lengths)
+        return recs.iloc[order[ranks < k]].reset_index(drop=True)
 
     def _predict_wrap(
         self,
