@@ -10,6 +10,8 @@ from __future__ import annotations
 
 from typing import Optional
 
+import os
+
 import numpy as np
 import pandas as pd
 from scipy.sparse import csr_matrix
@@ -49,7 +51,7 @@ class SLIM(NeighbourRec):
 
         alpha = self.beta + self.lambda_
         l1_ratio = self.lambda_ / alpha
-        model = ElasticNet(
+        params = dict(
             alpha=alpha,
             l1_ratio=l1_ratio,
             positive=True,
@@ -61,22 +63,44 @@ class SLIM(NeighbourRec):
             tol=1e-4,
             random_state=self.seed,
         )
-        out_rows, out_cols, out_data = [], [], []
-        for j in range(mat.shape[1]):
-            y = np.asarray(mat[:, j].todense()).ravel()
-            if not y.any():
-                continue
-            start, end = mat.indptr[j], mat.indptr[j + 1]
-            backup = mat.data[start:end].copy()
-            mat.data[start:end] = 0.0  # exclude the target column
-            model.fit(mat, y)
-            mat.data[start:end] = backup
-            coef = model.sparse_coef_.tocoo()
-            for i, v in zip(coef.col, coef.data):
-                if v > 0 and i != j:
-                    out_rows.append(i)
-                    out_cols.append(j)
-                    out_data.append(v)
+
+        def fit_columns(col_range):
+            model = ElasticNet(**params)
+            local = mat.copy()
+            rows_o, cols_o, data_o = [], [], []
+            for j in col_range:
+                start, end = local.indptr[j], local.indptr[j + 1]
+                if start == end:
+                    continue
+                y = np.zeros(local.shape[0])
+                y[local.indices[start:end]] = local.data[start:end]
+                backup = local.data[start:end].copy()
+                local.data[start:end] = 0.0  # exclude the target column
+                model.fit(local, y)
+                local.data[start:end] = backup
+                coef = model.sparse_coef_.tocoo()
+                keep = (coef.data > 0) & (coef.col != j)
+                rows_o.append(coef.col[keep].astype(np.int64))
+                cols_o.append(np.full(int(keep.sum()), j, dtype=np.int64))
+                data_o.append(coef.data[keep])
+            cat = lambda parts, dt: np.concatenate(parts) if parts else np.array([], dtype=dt)
+            return cat(rows_o, np.int64), cat(cols_o, np.int64), cat(data_o, np.float64)
+
+        # one independent ElasticNet per item column: parallelize across
+        # processes (the reference distributes the same loop over Spark
+        # executors, replay/models/slim.py)
+        try:
+            from joblib import Parallel, delayed
+
+            n_jobs = min(8, max(1, (os.cpu_count() or 1)))
+            chunks = np.array_split(np.arange(mat.shape[1]), n_jobs * 4)
+            results = Parallel(n_jobs=n_jobs)(delayed(fit_columns)(c) for c in chunks if len(c))
+        except ImportError:  # pragma: no cover
+            results = [fit_columns(np.arange(mat.shape[1]))]
         self.similarity = pd.DataFrame(
-            {"item_idx_one": out_rows, "item_idx_two": out_cols, "similarity": out_data}
+            {
+                "item_idx_one": np.concatenate([r[0] for r in results]),
+                "item_idx_two": np.concatenate([r[1] for r in results]),
+                "similarity": np.concatenate([r[2] for r in results]),
+            }
         )
