@@ -30,6 +30,12 @@ class CategoricalEmbedding(torch.nn.Module):
         self.cardinality = cardinality
         self.embedding_dim = embedding_dim
         self.item_emb = torch.nn.Embedding(cardinality + 1 + n_extra, embedding_dim, padding_idx=cardinality)
+        # reference embedding.py:199: xavier-normal table init (the torch
+        # N(0,1) default puts the initial full-softmax CE at ~5x ln(V)
+        # through the sqrt(d)-scaled tied head)
+        torch.nn.init.xavier_normal_(self.item_emb.weight.data)
+        with torch.no_grad():
+            self.item_emb.weight[cardinality].zero_()
 
     @property
     def weight(self) -> torch.Tensor:

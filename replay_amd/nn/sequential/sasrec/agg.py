@@ -19,6 +19,7 @@ class PositionAwareAggregator(torch.nn.Module):
         self._dim = embedding_dim
         self.max_sequence_length = max_sequence_length
         self.pos_embedding = torch.nn.Embedding(max_sequence_length, embedding_dim)
+        torch.nn.init.xavier_normal_(self.pos_embedding.weight.data)  # reference init
         self.dropout = torch.nn.Dropout(dropout)
 
     @property
