@@ -138,7 +138,8 @@ class Metric:
             name = f"{self.__name__}@{k}"
             agg = self._mode.cpu(values)
             if isinstance(self._mode, PerUser):
-                result[name] = pd.DataFrame({self.query_column: queries, name: agg})
+                # reference layout: "Metric-PerUser@k" -> {query: value}
+                result[f"{self.__name__}-PerUser@{k}"] = dict(zip(queries, agg.tolist()))
             else:
                 if not isinstance(self._mode, Mean):
                     name = f"{self.__name__}-{self._mode.__name__}@{k}"

@@ -134,9 +134,9 @@ def test_median_and_ci_modes(recs, ground_truth):
 
 def test_per_user_mode(recs, ground_truth):
     out = Recall([2], mode=PerUser())(recs, ground_truth)
-    df = out["Recall@2"]
-    assert isinstance(df, pd.DataFrame)
-    assert len(df) == 2
+    per_user = out["Recall-PerUser@2"]  # reference layout: {query: value}
+    assert isinstance(per_user, dict)
+    assert len(per_user) == 2
 
 
 def test_offline_metrics(recs, ground_truth, train_log):
