@@ -420,3 +420,20 @@ class TestReferenceDoctestParity:
         )
         assert m.shape == want.shape
         assert (np.asarray(m.todense()) == want).all()
+
+    def test_per_user_mode_matches_reference_doctest(self):
+        from replay_amd.metrics import NDCG, PerUser
+
+        recs = pd.DataFrame(
+            {"query_id": [1, 1, 1, 1, 1, 2, 2, 2, 2, 2, 3, 3, 3],
+             "item_id": [3, 7, 10, 11, 2, 5, 8, 11, 1, 3, 4, 9, 2],
+             "rating": [0.6, 0.5, 0.4, 0.3, 0.2, 0.6, 0.5, 0.4, 0.3, 0.2, 1.0, 0.5, 0.1]}
+        )
+        gt = pd.DataFrame(
+            {"query_id": [1] * 6 + [2] * 5 + [3] * 5,
+             "item_id": [5, 6, 7, 8, 9, 10, 6, 7, 4, 10, 11, 1, 2, 3, 4, 5]}
+        )
+        out = NDCG(2, mode=PerUser())(recs, gt)
+        assert out == {
+            "NDCG-PerUser@2": {1: 0.38685280723454163, 2: 0.0, 3: 0.6131471927654584}
+        }
