@@ -1,0 +1,147 @@
+"""Quality parity against the ACTUAL reference implementation, executed from
+the read-only checkout as a numerical oracle ("quality parity
+(NDCG@10/HitRate@10 match vs reference implementations on identical
+seeds/data)").
+
+Both models train with the same loop, optimizer, seeds and synthetic
+next-item data; we assert our redesigned SASRec reaches at least the
+reference's ranking quality.  (Logit-level equality is NOT expected: the
+reference replicates Kang's original composition — only the attention
+query is layer-normed and the residual is taken from the normalized
+query — while this framework uses a standard pre-LN transformer.)
+"""
+
+import sys
+from pathlib import Path
+
+import numpy as np
+import pytest
+import torch
+
+sys.path.insert(0, str(Path(__file__).resolve().parent))
+
+pytestmark = [pytest.mark.torch, pytest.mark.slow]
+
+V, L, D, HEADS = 60, 12, 32, 2
+N_USERS, EPOCHS, LR = 256, 60, 5e-3
+
+
+def _make_data(seed=0):
+    """Deterministic next-item structure: item[t+1] = (item[t] + step) % V
+    with a per-user step in {1, 3}."""
+    rng = np.random.default_rng(seed)
+    seqs = np.zeros((N_USERS, L + 1), dtype=np.int64)
+    for u in range(N_USERS):
+        step = 1 if u % 2 == 0 else 3
+        start = int(rng.integers(0, V))
+        seqs[u] = (start + step * np.arange(L + 1)) % V
+    return torch.from_numpy(seqs)
+
+
+def _train(model, forward_logits, seqs, seed=1):
+    """Shared loop: full-catalog CE on every next-item position."""
+    torch.manual_seed(seed)
+    opt = torch.optim.Adam(model.parameters(), lr=LR)
+    inputs, labels = seqs[:, :-1], seqs[:, 1:]
+    mask = torch.ones_like(inputs, dtype=torch.bool)
+    model.train()
+    for _ in range(EPOCHS):
+        opt.zero_grad(set_to_none=True)
+        logits = forward_logits(inputs, mask)  # [B, L, V]
+        loss = torch.nn.functional.cross_entropy(
+            logits.reshape(-1, logits.shape[-1]).float(), labels.reshape(-1)
+        )
+        loss.backward()
+        opt.step()
+    return float(loss.detach())
+
+
+def _hitrate_at1(model, forward_last_logits, seqs):
+    model.eval()
+    inputs, target = seqs[:, :-1], seqs[:, -1]
+    mask = torch.ones_like(inputs, dtype=torch.bool)
+    with torch.no_grad():
+        logits = forward_last_logits(inputs, mask)  # [B, V]
+    return float((logits.argmax(-1) == target).float().mean())
+
+
+@pytest.fixture(scope="module")
+def reference():
+    from _reference_harness import load_reference
+
+    return load_reference()
+
+
+class TestSasRecQualityParity:
+    def test_ours_matches_reference_hitrate(self, reference):
+        from replay.data import FeatureHint as RFH, FeatureType as RFT
+        from replay.data.nn import TensorFeatureInfo as RTFI, TensorSchema as RTS
+        from replay.models.nn.sequential.sasrec.model import SasRecModel
+
+        from replay_amd.data.nn import TensorFeatureInfo, TensorSchema
+        from replay_amd.data.schema import FeatureHint, FeatureType
+        from replay_amd.nn.sequential.sasrec import SasRec
+
+        seqs = _make_data()
+        holdout = _make_data(seed=99)
+
+        ref_ts = RTS([RTFI("item_id", RFT.CATEGORICAL, is_seq=True,
+                           feature_hint=RFH.ITEM_ID, cardinality=V, embedding_dim=D)])
+        torch.manual_seed(0)
+        ref_model = SasRecModel(schema=ref_ts, max_len=L, hidden_size=D,
+                                num_blocks=1, num_heads=HEADS, dropout=0.0)
+        _train(ref_model, lambda x, m: ref_model({"item_id": x}, m), seqs)
+        ref_hit = _hitrate_at1(ref_model, lambda x, m: ref_model.predict({"item_id": x}, m), holdout)
+
+        our_ts = TensorSchema([TensorFeatureInfo("item_id", FeatureType.CATEGORICAL, is_seq=True,
+                                                 feature_hint=FeatureHint.ITEM_ID,
+                                                 cardinality=V, embedding_dim=D)])
+        torch.manual_seed(0)
+        our_model = SasRec.from_params(our_ts, max_sequence_length=L, embedding_dim=D,
+                                       num_blocks=1, num_heads=HEADS, dropout=0.0)
+
+        def our_logits(x, m):
+            hidden = our_model.body({"item_id": x}, m)
+            return our_model.head(hidden)
+
+        _train(our_model, our_logits, seqs)
+        our_hit = _hitrate_at1(
+            our_model, lambda x, m: our_model.forward_inference({"item_id": x, "padding_mask": m}), holdout
+        )
+
+        # both must learn the pattern, and ours must not be worse
+        assert ref_hit > 0.8, f"oracle failed to learn (ref hit@1={ref_hit:.2f})"
+        assert our_hit >= ref_hit - 0.05, f"ours {our_hit:.2f} vs reference {ref_hit:.2f}"
+
+    def test_architectures_share_parameter_layout(self, reference):
+        """Every reference parameter maps 1:1 onto ours (same shapes), so
+        reference checkpoints are convertible."""
+        from replay.data import FeatureHint as RFH, FeatureType as RFT
+        from replay.data.nn import TensorFeatureInfo as RTFI, TensorSchema as RTS
+        from replay.models.nn.sequential.sasrec.model import SasRecModel
+
+        from replay_amd.data.nn import TensorFeatureInfo, TensorSchema
+        from replay_amd.data.schema import FeatureHint, FeatureType
+        from replay_amd.nn.sequential.sasrec import SasRec
+
+        ref_ts = RTS([RTFI("item_id", RFT.CATEGORICAL, is_seq=True,
+                           feature_hint=RFH.ITEM_ID, cardinality=V, embedding_dim=D)])
+        ref_model = SasRecModel(schema=ref_ts, max_len=L, hidden_size=D,
+                                num_blocks=2, num_heads=HEADS, dropout=0.0)
+        our_ts = TensorSchema([TensorFeatureInfo("item_id", FeatureType.CATEGORICAL, is_seq=True,
+                                                 feature_hint=FeatureHint.ITEM_ID,
+                                                 cardinality=V, embedding_dim=D)])
+        our_model = SasRec.from_params(our_ts, max_sequence_length=L, embedding_dim=D,
+                                       num_blocks=2, num_heads=HEADS, dropout=0.0)
+
+        def canon(sd):
+            # body-side only (the tied head re-registers embedder tensors
+            # differently: the reference also registers its positional table
+            # under the head); conv1d [E, E, 1] == linear [E, E]
+            return sorted(
+                tuple(v.squeeze(-1).shape)
+                for k, v in sd.items()
+                if "head" not in k.lower()
+            )
+
+        assert canon(ref_model.state_dict()) == canon(our_model.state_dict())
