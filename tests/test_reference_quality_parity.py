@@ -145,3 +145,84 @@ class TestSasRecQualityParity:
             )
 
         assert canon(ref_model.state_dict()) == canon(our_model.state_dict())
+
+
+class TestBert4RecQualityParity:
+    def test_ours_matches_reference_hitrate(self, reference):
+        from replay.data import FeatureHint as RFH, FeatureType as RFT
+        from replay.data.nn import TensorFeatureInfo as RTFI, TensorSchema as RTS
+        from replay.models.nn.sequential.bert4rec.model import Bert4RecModel
+
+        from replay_amd.data.nn import TensorFeatureInfo, TensorSchema
+        from replay_amd.data.schema import FeatureHint, FeatureType
+        from replay_amd.nn.sequential.bert4rec import Bert4Rec
+
+        seqs = _make_data()
+        holdout = _make_data(seed=99)
+
+        def last_masked_batch_ref(s):
+            # reference convention: token_mask 0 = <MASK>; mask the LAST slot
+            inputs = s[:, :-1].clone()
+            target = s[:, -1]
+            x = torch.cat([inputs[:, 1:], torch.zeros(len(s), 1, dtype=torch.long)], 1)
+            pad = torch.ones_like(x, dtype=torch.bool)
+            tok = torch.ones_like(x, dtype=torch.bool)
+            tok[:, -1] = False
+            return x, pad, tok, target
+
+        ref_ts = RTS([RTFI("item_id", RFT.CATEGORICAL, is_seq=True,
+                           feature_hint=RFH.ITEM_ID, cardinality=V, embedding_dim=D)])
+        torch.manual_seed(0)
+        ref_model = Bert4RecModel(schema=ref_ts, max_len=L, hidden_size=D,
+                                  num_blocks=1, num_heads=HEADS, dropout=0.0)
+        opt = torch.optim.Adam(ref_model.parameters(), lr=LR)
+        x, pad, tok, target = last_masked_batch_ref(seqs)
+        ref_model.train()
+        for _ in range(EPOCHS):
+            opt.zero_grad(set_to_none=True)
+            logits = ref_model.forward({"item_id": x}, pad, tok)[:, -1]
+            torch.nn.functional.cross_entropy(logits.float(), target).backward()
+            opt.step()
+        ref_model.eval()
+        hx, hpad, htok, htarget = last_masked_batch_ref(holdout)
+        with torch.no_grad():
+            ref_hit = float(
+                (ref_model.predict({"item_id": hx}, hpad, htok).argmax(-1) == htarget).float().mean()
+            )
+
+        our_ts = TensorSchema([TensorFeatureInfo("item_id", FeatureType.CATEGORICAL, is_seq=True,
+                                                 feature_hint=FeatureHint.ITEM_ID,
+                                                 cardinality=V, embedding_dim=D)])
+        torch.manual_seed(0)
+        our_model = Bert4Rec.from_params(our_ts, max_sequence_length=L, embedding_dim=D,
+                                         num_blocks=1, num_heads=HEADS, dropout=0.0)
+        opt = torch.optim.Adam(our_model.parameters(), lr=LR)
+        inputs, target = seqs[:, :-1], seqs[:, -1]
+        x = torch.cat([inputs[:, 1:], torch.zeros(len(seqs), 1, dtype=torch.long)], 1)
+        batch = {
+            "item_id": x,
+            "padding_mask": torch.ones_like(x, dtype=torch.bool),
+            "labels": torch.cat([x[:, :-1], target.unsqueeze(1)], 1),
+            "labels_padding_mask": torch.ones_like(x, dtype=torch.bool),
+            "token_mask": torch.zeros_like(x, dtype=torch.bool),
+        }
+        batch["token_mask"][:, -1] = True  # our convention: True = masked
+        our_model.train()
+        for _ in range(EPOCHS):
+            opt.zero_grad(set_to_none=True)
+            our_model(batch).backward()
+            opt.step()
+        our_model.eval()
+        # forward_inference expects max_len-padded input (one left pad here);
+        # it left-aligns, drops the oldest slot and appends the mask token
+        hist = torch.cat([torch.zeros(len(holdout), 1, dtype=torch.long), holdout[:, 1:-1]], 1)
+        hmask = torch.ones(len(holdout), L, dtype=torch.bool)
+        hmask[:, 0] = False
+        hbatch = {"item_id": hist, "padding_mask": hmask}
+        with torch.no_grad():
+            our_hit = float(
+                (our_model.forward_inference(hbatch).argmax(-1) == holdout[:, -1]).float().mean()
+            )
+
+        assert ref_hit > 0.7, f"oracle failed to learn (ref hit@1={ref_hit:.2f})"
+        assert our_hit >= ref_hit - 0.05, f"ours {our_hit:.2f} vs reference {ref_hit:.2f}"
