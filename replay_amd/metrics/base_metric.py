@@ -135,16 +135,21 @@ class Metric:
         result: MetricsReturnType = {}
         for k in self.topk:
             values = np.asarray(per_k[k], dtype=np.float64)
-            name = f"{self.__name__}@{k}"
-            agg = self._mode.cpu(values)
-            if isinstance(self._mode, PerUser):
-                # reference layout: "Metric-PerUser@k" -> {query: value}
-                result[f"{self.__name__}-PerUser@{k}"] = dict(zip(queries, agg.tolist()))
-            else:
-                if not isinstance(self._mode, Mean):
-                    name = f"{self.__name__}-{self._mode.__name__}@{k}"
-                result[name] = agg
+            result.update(self._format_result(k, values, queries))
         return result
+
+    def _format_result(self, k: int, values: np.ndarray, queries) -> "MetricsReturnType":
+        """Apply the aggregation mode with the reference's key layout
+        ("Metric@k", "Metric-Median@k", "Metric-PerUser@k" -> {query: v})."""
+        agg = self._mode.cpu(values)
+        if isinstance(self._mode, PerUser):
+            return {f"{self.__name__}-PerUser@{k}": dict(zip(queries, agg.tolist()))}
+        name = (
+            f"{self.__name__}@{k}"
+            if isinstance(self._mode, Mean)
+            else f"{self.__name__}-{self._mode.__name__}@{k}"
+        )
+        return {name: agg}
 
     @staticmethod
     def _get_metric_value_by_user(k: int, pred: Sequence, ground_truth: Sequence) -> float:  # pragma: no cover

@@ -159,17 +159,18 @@ class Novelty(_TrainAwareMetric):
 
     def _compute_with_train(self, recs: Dict, train: Dict) -> MetricsReturnType:
         result = {}
+        queries = list(train.keys())  # reference iterates TRAIN users
         for k in self.topk:
             values = []
-            for q, pred in recs.items():
+            for q in queries:
+                pred = recs.get(q, [])
                 seen = set(train.get(q, []))
                 topk_items = pred[:k]
                 if not topk_items or not seen:
                     values.append(1.0)  # reference novelty.py:144: vacuously novel
                     continue
                 values.append(sum(1 for p in topk_items if p not in seen) / len(topk_items))
-            agg = self._mode.cpu(np.asarray(values, dtype=np.float64))
-            result[f"{self.__name__}@{k}"] = agg
+            result.update(self._format_result(k, np.asarray(values, dtype=np.float64), queries))
         return result
 
 
@@ -188,13 +189,17 @@ class Surprisal(_TrainAwareMetric):
         for k in self.topk:
             values = []
             for q, pred in recs.items():
-                infos = []
+                # reference surprisal.py:188 divides the weight SUM by K
+                # (short lists lower the score); unseen items weigh 1.0
+                total = 0.0
                 for p in pred[:k]:
                     share = item_users.get(p, 0) / n_users
                     info = -np.log2(share) if share > 0 else max_info
-                    infos.append(info / max_info)
-                values.append(float(np.mean(infos)) if infos else 0.0)
-            result[f"{self.__name__}@{k}"] = self._mode.cpu(np.asarray(values, dtype=np.float64))
+                    total += info / max_info
+                values.append(total / k if pred else 0.0)
+            result.update(
+                self._format_result(k, np.asarray(values, dtype=np.float64), list(recs.keys()))
+            )
         return result
 
 
@@ -222,7 +227,9 @@ class Unexpectedness(Metric):
                 # reference unexpectedness.py:156 divides by K (not by the
                 # number of produced recs): short lists count as unexpected
                 values.append(1.0 - len(set(topk_items) & base_set) / k)
-            result[f"{self.__name__}@{k}"] = self._mode.cpu(np.asarray(values, dtype=np.float64))
+            result.update(
+                self._format_result(k, np.asarray(values, dtype=np.float64), list(recs.keys()))
+            )
         return result
 
 

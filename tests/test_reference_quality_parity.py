@@ -226,3 +226,51 @@ class TestBert4RecQualityParity:
 
         assert ref_hit > 0.7, f"oracle failed to learn (ref hit@1={ref_hit:.2f})"
         assert our_hit >= ref_hit - 0.05, f"ours {our_hit:.2f} vs reference {ref_hit:.2f}"
+
+
+class TestMetricOracleEquality:
+    """Ours == the reference metric implementations on randomized frames
+    (the reference metrics are pandas-capable, so they run directly as
+    oracles here)."""
+
+    def test_randomized_equality(self, reference):
+        import numpy as np
+        import pandas as pd
+
+        from replay.metrics import (
+            MAP as RefMAP, MRR as RefMRR, NDCG as RefNDCG,
+            Coverage as RefCoverage, HitRate as RefHitRate,
+            Novelty as RefNovelty, Precision as RefPrecision,
+            Recall as RefRecall, RocAuc as RefRocAuc, Surprisal as RefSurprisal,
+        )
+
+        from replay_amd.metrics import (
+            MAP, MRR, NDCG, Coverage, HitRate, Novelty, Precision, Recall,
+            RocAuc, Surprisal,
+        )
+
+        rng = np.random.default_rng(0)
+        gt_pairs = [(RefNDCG, NDCG), (RefMAP, MAP), (RefMRR, MRR), (RefHitRate, HitRate),
+                    (RefPrecision, Precision), (RefRecall, Recall), (RefRocAuc, RocAuc)]
+        train_pairs = [(RefCoverage, Coverage), (RefNovelty, Novelty), (RefSurprisal, Surprisal)]
+        for trial in range(15):
+            n_rec = int(rng.integers(5, 60))
+            n_gt = int(rng.integers(3, 40))
+            recs = pd.DataFrame(
+                {"query_id": rng.integers(0, 6, n_rec), "item_id": rng.integers(0, 25, n_rec),
+                 "rating": rng.random(n_rec)}
+            ).drop_duplicates(["query_id", "item_id"])
+            gt = pd.DataFrame(
+                {"query_id": rng.integers(0, 6, n_gt), "item_id": rng.integers(0, 25, n_gt)}
+            ).drop_duplicates()
+            train = pd.DataFrame(
+                {"query_id": rng.integers(0, 6, n_gt + 5), "item_id": rng.integers(0, 25, n_gt + 5)}
+            ).drop_duplicates()
+            for Ref, Ours in gt_pairs:
+                r, o = Ref([3, 7])(recs, gt), Ours([3, 7])(recs, gt)
+                for key, val in r.items():
+                    assert o[key] == pytest.approx(val, abs=1e-9), (trial, key)
+            for Ref, Ours in train_pairs:
+                r, o = Ref([3, 7])(recs, train), Ours([3, 7])(recs, train)
+                for key, val in r.items():
+                    assert o[key] == pytest.approx(val, abs=1e-9), (trial, key)
