@@ -215,18 +215,35 @@ class QuantileItemsFilter(_BaseFilter):
         self.item_column = item_column
 
     def _core_filter(self, df: pd.DataFrame) -> pd.DataFrame:
-        counts = df[self.item_column].value_counts()
-        threshold = counts.quantile(self.alpha_quantile)
-        popular = set(counts[counts > threshold].index)
-        if not popular:
+        """Exact reference algorithm (filters.py:900-922): the popularity
+        threshold is the midpoint-interpolated quantile of item counts; for
+        each over-threshold item, items_proportion * (count - max long-tail
+        count) interactions are deleted, taken from the MOST ACTIVE users
+        first (keeps the item-popularity ordering intact)."""
+        item_counts = df.groupby(self.item_column)[self.item_column].transform("size")
+        user_counts = df.groupby(self.query_column)[self.query_column].transform("size")
+        per_item = df.groupby(self.item_column).size()
+        threshold = per_item.quantile(self.alpha_quantile, interpolation="midpoint")
+        long_tail_mask = item_counts <= threshold
+        long_tail = df[long_tail_mask]
+        short = df[~long_tail_mask].copy()
+        if not len(short):
             return df
-        keep_rows = []
-        is_popular = df[self.item_column].isin(popular)
-        keep_rows.append(df[~is_popular])
-        for _, group in df[is_popular].groupby(self.item_column):
-            n_keep = max(1, int(np.ceil(len(group) * self.items_proportion)))
-            keep_rows.append(group.head(n_keep))
-        return pd.concat(keep_rows).sort_index()
+        long_tail_max = int(item_counts[long_tail_mask].max()) if long_tail_mask.any() else 0
+        short["_n_del"] = (
+            self.items_proportion * (item_counts[~long_tail_mask] - long_tail_max)
+        ).astype(int)
+        short["_u_cnt"] = user_counts[~long_tail_mask]
+        short = short.sort_values("_u_cnt", ascending=False)  # reference tie order (quicksort)
+
+        def keep_mask(x):
+            mask = np.ones(len(x), dtype=bool)
+            mask[: int(x.iloc[0])] = False
+            return pd.Series(mask, index=x.index)
+
+        mask = short.groupby(self.item_column)["_n_del"].transform(keep_mask).astype(bool)
+        kept = short.loc[mask, df.columns.tolist()]
+        return pd.concat([long_tail, kept])
 
 
 class ConsecutiveDuplicatesFilter(_BaseFilter):

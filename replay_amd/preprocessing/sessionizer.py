@@ -39,7 +39,7 @@ class Sessionizer:
             seconds = ts.astype("int64")
         new_user = df[self.user_column].ne(df[self.user_column].shift())
         gap = seconds.diff().fillna(0)
-        new_session = new_user | (gap > self.session_gap)
+        new_session = new_user | (gap >= self.session_gap)  # reference: gap >= session_gap splits
         df[self.session_column] = np.cumsum(new_session.to_numpy()).astype(np.int64) - 1
         if self.min_inter_per_session is not None or self.max_inter_per_session is not None:
             sizes = df.groupby(self.session_column)[self.session_column].transform("size")

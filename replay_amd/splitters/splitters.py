@@ -51,10 +51,14 @@ class RatioSplitter(Splitter):
             frac = (row_num / sizes).round(3)
             is_test = frac > round(1.0 - self.test_size, 3)
         else:
-            # non-fraction mode: n_train = half-up-round(count * (1 - ratio));
-            # this puts one more interaction per group into train than the
-            # fraction mode (reference docstring)
-            n_train = np.floor(sizes * (1.0 - self.test_size) + 0.5)
+            # non-fraction mode (reference ratio_splitter.py:292-309):
+            # n_train = count - int(count * ratio); when no
+            # min_interactions_per_group is set, groups too small to yield a
+            # test row naturally (0 < count*ratio < 1) give up one anyway
+            n_train = sizes - (sizes * self.test_size).astype(int)
+            if self.min_interactions_per_group is None:
+                frac = sizes * self.test_size
+                n_train = n_train.where(~((frac > 0) & (frac < 1) & (n_train > 1)), n_train - 1)
             is_test = row_num > n_train
         if self.min_interactions_per_group is not None:
             # undersized groups go entirely to train (reference :214)

@@ -1,0 +1,177 @@
+"""Randomized frame-level equality against the reference's own pandas
+implementations of splitters, filters, encoders and preprocessing, executed
+from the read-only checkout as oracles (same harness as
+test_reference_quality_parity)."""
+
+import sys
+from pathlib import Path
+
+import numpy as np
+import pandas as pd
+import pytest
+
+sys.path.insert(0, str(Path(__file__).resolve().parent))
+
+pytestmark = [pytest.mark.core, pytest.mark.slow]
+
+
+@pytest.fixture(scope="module")
+def reference():
+    from _reference_harness import load_reference
+
+    return load_reference()
+
+
+def _frames(seed):
+    rng = np.random.default_rng(seed)
+    n = int(rng.integers(30, 90))
+    return pd.DataFrame(
+        {
+            "query_id": rng.integers(0, 8, n),
+            "item_id": rng.integers(0, 20, n),
+            "rating": (rng.random(n) * 5).round(2),
+            "timestamp": rng.permutation(n).astype(np.int64),
+        }
+    )
+
+
+def _same_rows(a: pd.DataFrame, b: pd.DataFrame) -> bool:
+    if len(a) != len(b):
+        return False
+    cols = ["query_id", "item_id", "rating", "timestamp"]
+    cols = [c for c in cols if c in a.columns and c in b.columns]
+    aa = a[cols].sort_values(cols).reset_index(drop=True)
+    bb = b[cols].sort_values(cols).reset_index(drop=True)
+    return aa.equals(bb)
+
+
+class TestSplitterOracle:
+    @pytest.mark.parametrize("seed", range(6))
+    def test_deterministic_splitters_match(self, reference, seed):
+        import replay.splitters as ref_sp
+
+        import replay_amd.splitters as our_sp
+
+        df = _frames(seed)
+        cases = [
+            ("RatioSplitter", dict(test_size=0.3, divide_column="query_id", query_column="query_id")),
+            ("RatioSplitter", dict(test_size=0.4, divide_column="query_id", query_column="query_id",
+                                   split_by_fractions=False)),
+            ("LastNSplitter", dict(N=2, divide_column="query_id", query_column="query_id")),
+            ("TimeSplitter", dict(time_threshold=int(df["timestamp"].quantile(0.7)) + 1,
+                                  query_column="query_id")),
+            ("NewUsersSplitter", dict(test_size=0.25, query_column="query_id")),
+        ]
+        for name, kwargs in cases:
+            rt, re_ = getattr(ref_sp, name)(**kwargs).split(df)
+            ot, oe = getattr(our_sp, name)(**kwargs).split(df)
+            assert _same_rows(rt, ot), f"{name} {kwargs}: train differs (seed {seed})"
+            assert _same_rows(re_, oe), f"{name} {kwargs}: test differs (seed {seed})"
+
+
+class TestFilterOracle:
+    @pytest.mark.parametrize("seed", range(6))
+    def test_filters_match(self, reference, seed):
+        import replay.preprocessing.filters as ref_f
+
+        import replay_amd.preprocessing.filters as our_f
+
+        df = _frames(seed)
+        tdf = df.copy()
+        tdf["timestamp"] = pd.to_datetime(tdf["timestamp"], unit="D", origin="2020-01-01")
+        cases = [
+            ("MinCountFilter", dict(num_entries=3, groupby_column="query_id"), df),
+            ("LowRatingFilter", dict(value=2.5), df),
+            ("InteractionEntriesFilter",
+             dict(query_column="query_id", min_inter_per_user=3, max_inter_per_item=20), df),
+            ("NumInteractionsFilter",
+             dict(num_interactions=4, first=False, query_column="query_id"), df),
+            ("NumInteractionsFilter",
+             dict(num_interactions=2, first=True, query_column="query_id"), df),
+            ("EntityDaysFilter", dict(days=10, first=True, entity_column="query_id"), tdf),
+            ("GlobalDaysFilter", dict(days=15, first=False), tdf),
+            ("TimePeriodFilter", dict(start_date="2020-01-05 00:00:00",
+                                      end_date="2020-02-01 00:00:00"), tdf),
+            ("QuantileItemsFilter", dict(alpha_quantile=0.8, items_proportion=0.5,
+                                         query_column="query_id"), df),
+            ("ConsecutiveDuplicatesFilter", dict(query_column="query_id"), df),
+        ]
+        for name, kwargs, frame in cases:
+            r = getattr(ref_f, name)(**kwargs).transform(frame)
+            o = getattr(our_f, name)(**kwargs).transform(frame)
+            assert _same_rows(r, o), f"{name} {kwargs}: rows differ (seed {seed})"
+
+
+class TestEncoderOracle:
+    @pytest.mark.parametrize("seed", range(6))
+    def test_label_encoder_matches(self, reference, seed):
+        from replay.preprocessing import LabelEncoder as RefLE, LabelEncodingRule as RefRule
+
+        from replay_amd.preprocessing import LabelEncoder, LabelEncodingRule
+
+        df = _frames(seed)
+        ref = RefLE([RefRule("query_id"), RefRule("item_id")]).fit(df)
+        ours = LabelEncoder([LabelEncodingRule("query_id"), LabelEncodingRule("item_id")]).fit(df)
+        assert ref.mapping == ours.mapping
+        rt = ref.transform(df)
+        ot = ours.transform(df)
+        assert rt["query_id"].tolist() == ot["query_id"].tolist()
+        assert rt["item_id"].tolist() == ot["item_id"].tolist()
+
+    def test_sequence_rule_matches(self, reference):
+        from replay.preprocessing import SequenceEncodingRule as RefSeq
+
+        from replay_amd.preprocessing import SequenceEncodingRule
+
+        df = pd.DataFrame({"items": [[3, 1, 2], [2, 5], [9, 1, 3]]})
+        r = RefSeq("items").fit(df)
+        o = SequenceEncodingRule("items").fit(df)
+        assert getattr(r, "mapping", None) or r._mapping == o.mapping
+        rt = r.transform(df)["items"].apply(list).tolist()
+        ot = o.transform(df)["items"].apply(list).tolist()
+        assert rt == ot
+
+
+class TestPreprocessingOracle:
+    @pytest.mark.parametrize("seed", range(4))
+    def test_sessionizer_partition_matches(self, reference, seed):
+        from replay.preprocessing import Sessionizer as RefSess
+
+        from replay_amd.preprocessing import Sessionizer
+
+        df = _frames(seed)
+        r = RefSess(session_gap=5, user_column="query_id").transform(df)
+        o = Sessionizer(session_gap=5, user_column="query_id").transform(df)
+
+        def partition(frame):
+            return {
+                tuple(sorted(map(tuple, g[["query_id", "timestamp"]].to_numpy())))
+                for _, g in frame.groupby(["query_id", "session_id"])
+            }
+
+        assert partition(r) == partition(o), f"seed {seed}"
+
+    @pytest.mark.parametrize("seed", range(4))
+    def test_csr_converter_matches(self, reference, seed):
+        from replay.preprocessing import CSRConverter as RefCSR
+
+        from replay_amd.preprocessing import CSRConverter
+
+        df = _frames(seed).drop_duplicates(["query_id", "item_id"])
+        r = RefCSR(first_dim_column="query_id", second_dim_column="item_id",
+                   data_column="rating").transform(df)
+        o = CSRConverter(first_dim_column="query_id", second_dim_column="item_id",
+                         data_column="rating").transform(df)
+        assert r.shape == o.shape
+        assert np.allclose(np.asarray(r.todense()), np.asarray(o.todense()))
+
+    @pytest.mark.parametrize("seed", range(4))
+    def test_quantile_discretizer_matches(self, reference, seed):
+        from replay.preprocessing import Discretizer as RefDisc, QuantileDiscretizingRule as RefQ
+
+        from replay_amd.preprocessing import Discretizer, QuantileDiscretizingRule
+
+        df = _frames(seed)
+        r = RefDisc([RefQ("rating", n_bins=4)]).fit_transform(df)
+        o = Discretizer([QuantileDiscretizingRule("rating", n_bins=4)]).fit_transform(df)
+        assert r["rating"].tolist() == o["rating"].tolist(), f"seed {seed}"
