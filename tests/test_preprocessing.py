@@ -180,7 +180,9 @@ def test_quantile_items_filter():
         }
     )
     out = QuantileItemsFilter(alpha_quantile=0.5, items_proportion=0.5, query_column="user_id").transform(df)
-    assert (out["item_id"] == 1).sum() == 50
+    # reference formula: delete proportion * (count - long_tail_max)
+    # = int(0.5 * (100 - 1)) = 49 -> 51 kept
+    assert (out["item_id"] == 1).sum() == 51
     assert (out["item_id"] == 2).sum() == 1
 
 
