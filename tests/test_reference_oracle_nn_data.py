@@ -1,0 +1,116 @@
+"""Oracle equality for the tensor data layer and the GPU metrics builder:
+the reference's SequenceTokenizer/TorchSequentialDataset and
+TorchMetricsBuilder run directly (torch-only code) and ours must produce
+identical windows, masks and metric values."""
+
+import sys
+from pathlib import Path
+
+import numpy as np
+import pandas as pd
+import pytest
+import torch
+
+sys.path.insert(0, str(Path(__file__).resolve().parent))
+
+pytestmark = [pytest.mark.torch, pytest.mark.slow]
+
+
+@pytest.fixture(scope="module")
+def reference():
+    from _reference_harness import load_reference
+
+    return load_reference()
+
+
+class TestTokenizerAndWindowsOracle:
+    @pytest.mark.parametrize("seed", range(4))
+    @pytest.mark.parametrize("max_len", [4, 6])
+    def test_sequences_and_windows_match(self, reference, seed, max_len):
+        from replay.data import (
+            Dataset as RefDS, FeatureHint as RFH, FeatureInfo as RFI,
+            FeatureSchema as RFS, FeatureSource as RFSo, FeatureType as RFT,
+        )
+        from replay.data.nn import (
+            SequenceTokenizer as RefTok, TensorFeatureInfo as RTFI,
+            TensorFeatureSource as RTFS, TensorSchema as RTS,
+            TorchSequentialDataset as RefTSD,
+        )
+
+        from replay_amd.data import Dataset, FeatureHint, FeatureInfo, FeatureSchema, FeatureType
+        from replay_amd.data.nn import (
+            SequenceTokenizer, TensorFeatureInfo, TensorSchema, TorchSequentialDataset,
+        )
+
+        rng = np.random.default_rng(seed)
+        n = int(rng.integers(30, 60))
+        df = pd.DataFrame(
+            {
+                "query_id": rng.integers(0, 5, n),
+                "item_id": rng.integers(0, 12, n),
+                "timestamp": np.arange(n),
+            }
+        )
+
+        ref_schema = RFS([RFI("query_id", RFT.CATEGORICAL, RFH.QUERY_ID),
+                          RFI("item_id", RFT.CATEGORICAL, RFH.ITEM_ID),
+                          RFI("timestamp", RFT.NUMERICAL, RFH.TIMESTAMP)])
+        ref_ts = RTS([RTFI("item_id", RFT.CATEGORICAL, is_seq=True, feature_hint=RFH.ITEM_ID,
+                           feature_sources=[RTFS(RFSo.INTERACTIONS, "item_id")],
+                           cardinality=12, embedding_dim=8)])
+        ref_seqs = RefTok(ref_ts).fit_transform(RefDS(feature_schema=ref_schema, interactions=df))
+        ref_tsd = RefTSD(ref_seqs, max_sequence_length=max_len)
+
+        our_schema = FeatureSchema([FeatureInfo("query_id", FeatureType.CATEGORICAL, FeatureHint.QUERY_ID),
+                                    FeatureInfo("item_id", FeatureType.CATEGORICAL, FeatureHint.ITEM_ID),
+                                    FeatureInfo("timestamp", FeatureType.NUMERICAL, FeatureHint.TIMESTAMP)])
+        our_ts = TensorSchema([TensorFeatureInfo("item_id", FeatureType.CATEGORICAL, is_seq=True,
+                                                 feature_hint=FeatureHint.ITEM_ID,
+                                                 cardinality=12, embedding_dim=8)])
+        our_seqs = SequenceTokenizer(our_ts).fit_transform(
+            Dataset(feature_schema=our_schema, interactions=df)
+        )
+        our_tsd = TorchSequentialDataset(our_seqs, max_len)
+
+        assert len(ref_tsd) == len(our_tsd)
+        ref_by_q = {}
+        for i in range(len(ref_tsd)):
+            b = ref_tsd[i]
+            ref_by_q[int(b.query_id)] = (b.padding_mask.tolist(), b.features["item_id"].tolist())
+        for i in range(len(our_tsd)):
+            b = our_tsd[i]
+            q = int(b["query_id"])
+            mask = b["padding_mask"].tolist()
+            items = b["item_id"].tolist()
+            r_mask, r_items = ref_by_q[q]
+            assert mask == r_mask, (seed, max_len, q)
+            # compare only VALID positions: pad slots carry the pad value,
+            # which legitimately differs (reference 0, ours cardinality)
+            assert [x for x, m in zip(items, mask) if m] == [
+                x for x, m in zip(r_items, r_mask) if m
+            ], (seed, max_len, q)
+
+
+class TestTorchMetricsBuilderOracle:
+    @pytest.mark.parametrize("seed", range(5))
+    def test_builder_matches_reference(self, reference, seed):
+        from replay.metrics.torch_metrics_builder import TorchMetricsBuilder as RefBuilder
+
+        from replay_amd.metrics import TorchMetricsBuilder
+
+        torch.manual_seed(seed)
+        B, K, G, V = 16, 10, 5, 40
+        metrics = ["ndcg", "recall", "map", "precision", "coverage", "novelty"]
+        ref = RefBuilder(metrics=metrics, top_k=[3, 10], item_count=V)
+        ours = TorchMetricsBuilder(metrics=metrics, top_k=[3, 10], item_count=V)
+        for _ in range(3):  # multi-batch accumulation
+            preds = torch.rand(B, V).argsort(-1, descending=True)[:, :K]
+            gt = torch.randint(0, V, (B, G))
+            train = torch.randint(0, V, (B, 8))
+            ref.add_prediction(preds, gt, train)
+            ours.add_prediction(preds, gt, train)
+        r, o = ref.get_metrics(), ours.get_metrics()
+        assert set(r) == set(o)
+        for key, val in r.items():
+            # reference accumulates fp32; ours fp64 — allow fp32 epsilon
+            assert o[key] == pytest.approx(float(val), abs=1e-6), key
