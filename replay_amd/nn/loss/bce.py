@@ -29,8 +29,9 @@ class BCE(LossBase):
         target.scatter_(-1, safe.unsqueeze(-1), 1.0)
         per_elem = torch.nn.functional.binary_cross_entropy_with_logits(logits, target, reduction="none")
         valid = mask.unsqueeze(-1).to(per_elem.dtype)
-        n_valid_elems = valid.sum() * logits.shape[-1]
-        return (per_elem * valid).sum() / n_valid_elems.clamp(min=1e-12)
+        # reference bce.py: SUM over the catalog axis, MEAN over valid
+        # positions (not over positions x catalog)
+        return (per_elem * valid).sum() / valid.sum().clamp(min=1e-12)
 
 
 class BCESampled(SampledLossBase):

@@ -114,3 +114,43 @@ class TestTorchMetricsBuilderOracle:
         for key, val in r.items():
             # reference accumulates fp32; ours fp64 — allow fp32 epsilon
             assert o[key] == pytest.approx(float(val), abs=1e-6), key
+
+
+class TestLossOracle:
+    """Loss values equal the reference's own loss implementations on
+    identical tensors (CE / CESampled exact; BCE normalization was aligned
+    to the reference: catalog-summed, per-position mean)."""
+
+    @pytest.mark.parametrize("seed", range(4))
+    def test_losses_match_reference(self, reference, seed):
+        from replay.nn.loss import BCE as RefBCE, CE as RefCE, CESampled as RefCES
+
+        from replay_amd.nn.embedding import CategoricalEmbedding
+        from replay_amd.nn.head import EmbeddingTyingHead
+        from replay_amd.nn.loss import BCE, CE, CESampled
+
+        torch.manual_seed(seed)
+        B, L, E, V = 3, 5, 8, 20
+        head = EmbeddingTyingHead(CategoricalEmbedding(V, E))
+        x = torch.randn(B, L, E)
+        labels = torch.randint(0, V, (B, L))
+        mask = torch.ones(B, L, dtype=torch.bool)
+        mask[:, 0] = False
+        negs = torch.tensor([0, 3, 7, 11])
+        lab3, mask3 = labels.unsqueeze(-1), mask.unsqueeze(-1)
+
+        pairs = [
+            (RefCE(), CE(), lambda r: r(x, {}, lab3, negs, mask, mask3),
+             lambda o: o(x, labels, mask)),
+            (RefCES(), CESampled(),
+             lambda r: r(x, {}, lab3, negs.view(1, 1, -1).expand(B, L, -1), mask, mask3),
+             lambda o: o(x, labels, mask, negative_labels=negs)),
+            (RefBCE(), BCE(), lambda r: r(x, {}, lab3, negs, mask, mask3),
+             lambda o: o(x, labels, mask)),
+        ]
+        for ref_loss, our_loss, call_ref, call_ours in pairs:
+            ref_loss.logits_callback = head
+            our_loss.set_logits_callback(head)
+            r = float(call_ref(ref_loss).detach())
+            o = float(call_ours(our_loss).detach())
+            assert o == pytest.approx(r, rel=1e-5), type(our_loss).__name__
