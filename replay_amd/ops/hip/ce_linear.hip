@@ -31,6 +31,8 @@
 //   C (16x16 f32):  lane l holds C[row = (l>>4)*4 + r][col = l&15], r=0..3
 
 #include <torch/extension.h>
+
+#include <cstdlib>
 #include <ATen/hip/HIPContext.h>
 #include "common.h"
 
@@ -268,10 +270,13 @@ __global__ __launch_bounds__(256, 2) void ce_linear_bwd_kernel(
   const int m0 = blockIdx.x * 256 + wave * 64;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // layout: [wt stage (FUSE_DH): E x 64 bf16] [4 waves x dl bounce 16x64 bf16]
-  //         [|g| per WG row: 256 f32] [!RESIDENT: 256 x E bf16 A tile]
+  // layout: [FUSE_DH: 4 waves x W^T tile E x 64 bf16 (PER WAVE — barrier
+  // free: a WG-shared stage needed 2 __syncthreads per tile, which
+  // serialized the workgroup)] [4 waves x dl bounce 16x64 bf16]
+  // [per-row state: 3 x 256 words] [!RESIDENT: 256 x E bf16 A tile]
   __hip_bfloat16* wt_lds = reinterpret_cast<__hip_bfloat16*>(smem);
-  __hip_bfloat16* dl_lds = wt_lds + (FUSE_DH ? E * 64 : 0);
+  __hip_bfloat16* my_wt = wt_lds + (FUSE_DH ? (size_t)wave * E * 64 : 0);
+  __hip_bfloat16* dl_lds = wt_lds + (FUSE_DH ? 4 * E * 64 : 0);
   float* g_lds = reinterpret_cast<float*>(dl_lds + 4 * 16 * 64);
   float* adj_lds = g_lds + 256;
   int* lab_lds = reinterpret_cast<int*>(adj_lds + 256);
@@ -353,17 +358,18 @@ __global__ __launch_bounds__(256, 2) void ce_linear_bwd_kernel(
     const int tile = tile_at(ti);
     const int n0 = tile << 6;
     if constexpr (FUSE_DH) {
-      __syncthreads();  // all waves done with the previous W^T stage
-      for (int i = threadIdx.x; i < 64 * (E / 8); i += blockDim.x) {
+      // per-wave private W^T copy: no barriers, LDS write->read is
+      // program-ordered within the wave (4x the LDS write traffic of a
+      // shared stage, but the shared stage's 2 barriers/tile cost more)
+      for (int i = lane; i < 64 * (E / 8); i += WAVE) {
         const int item = i / (E / 8);
         const int e0 = (i % (E / 8)) * 8;
         bf16x8 vv = *reinterpret_cast<const bf16x8*>(w + (size_t)min(n0 + item, Vi - 1) * E + e0);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          wt_lds[wt_off(e0 + j, item)] = ((const __hip_bfloat16*)&vv)[j];
+          my_wt[wt_off(e0 + j, item)] = ((const __hip_bfloat16*)&vv)[j];
         }
       }
-      __syncthreads();
     }
     // dlogits = (softmax - onehot) * g, bounced per (mf) through LDS so the
     // [M, ldd] store is full aligned 128-B lines (ldd is a 64-item multiple:
@@ -463,7 +469,7 @@ __global__ __launch_bounds__(256, 2) void ce_linear_bwd_kernel(
 #pragma unroll
           for (int f = 0; f < OF; ++f) {
             bf16x8 b_w = *reinterpret_cast<const bf16x8*>(
-                wt_lds + wt_off(f * 16 + (lane & 15), ks2 * 32 + (lane >> 4) * 8));
+                my_wt + wt_off(f * 16 + (lane & 15), ks2 * 32 + (lane >> 4) * 8));
             dh_all[mf][f] =
                 __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_dl, b_w, dh_all[mf][f], 0, 0, 0);
           }
@@ -548,19 +554,24 @@ std::vector<torch::Tensor> ce_linear_bwd(torch::Tensor hidden, torch::Tensor w,
   TORCH_CHECK(V < (int64_t)INT32_MAX - 64, "catalog must fit int32");
   const int64_t Vp = (V + 63) & ~int64_t(63);  // full-line 128-B row segments
   auto dlogits = torch::empty({(int64_t)M, Vp}, hidden.options());
-  const bool fuse_dh = false;
+  // opt-in (round-2 experiment): barrier-free in-kernel dhidden fusion.
+  // Default off — the shipped path (host dX GEMM on the padded dlogits
+  // view) is the measured-fastest configuration.
+  static const bool fuse_dh_env = [] {
+    const char* v = std::getenv("REPLAY_AMD_CE_FUSE_DH");
+    return v != nullptr && v[0] == '1';
+  }();
+  const bool fuse_dh = fuse_dh_env && E == 64;
   auto dhidden = fuse_dh ? torch::empty_like(hidden)
                          : torch::empty({0}, hidden.options());  // host GEMM fallback
   auto labels_c = labels.contiguous();
   auto gscale_c = gscale.to(torch::kFloat32).contiguous();
   const int m_tiles = (M + 255) / 256;
   auto stream = at::cuda::getCurrentHIPStream();
-#define LAUNCH_CLB(EE)                                                                       \
+#define LAUNCH_CLB(EE, FDH)                                                                  \
   do {                                                                                       \
     constexpr bool RES = (EE <= 128);                                                        \
-    constexpr bool FDH = false; /* dh fusion measured slower: per-tile W^T-stage\
-                                   barriers serialize the WG; host dX GEMM wins */           \
-    size_t lds = (FDH ? (size_t)EE * 64 * 2 : 0) + 4 * 16 * 64 * 2 + 3 * 256 * 4 +           \
+    size_t lds = (FDH ? (size_t)4 * EE * 64 * 2 : 0) + 4 * 16 * 64 * 2 + 3 * 256 * 4 +       \
                  (RES ? 0 : (size_t)256 * EE * 2);                                           \
     hipLaunchKernelGGL((ce_linear_bwd_kernel<EE, RES, FDH>), dim3(m_tiles), dim3(256), lds,  \
                        stream, reinterpret_cast<const __hip_bfloat16*>(hidden.data_ptr()),   \
@@ -573,11 +584,15 @@ std::vector<torch::Tensor> ce_linear_bwd(torch::Tensor hidden, torch::Tensor w,
                        M, V, (int)Vp);                                                       \
   } while (0)
   if (E == 64) {
-    LAUNCH_CLB(64);
+    if (fuse_dh) {
+      LAUNCH_CLB(64, true);
+    } else {
+      LAUNCH_CLB(64, false);
+    }
   } else if (E == 128) {
-    LAUNCH_CLB(128);
+    LAUNCH_CLB(128, false);
   } else if (E == 256) {
-    LAUNCH_CLB(256);
+    LAUNCH_CLB(256, false);
   } else {
     TORCH_CHECK(false, "ce_linear supports E in {64, 128, 256}");
   }
