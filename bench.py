@@ -45,10 +45,20 @@ N_BLOCKS = 2
 N_HEADS = 2
 
 
-def build_model(device):
+def apply_model_config(args):
+    """--model sasrec (default, BASELINE config 2) or bert4rec (BASELINE
+    config 3 shape: 4 blocks, d=128, seq_len=200; run under torchrun x8 for
+    the DP=8 configuration)."""
+    global SEQ_LEN, EMB_DIM, N_BLOCKS, N_HEADS
+    if args.model == "bert4rec":
+        SEQ_LEN, EMB_DIM, N_BLOCKS, N_HEADS = 200, 128, 4, 4
+
+
+def build_model(device, model_name="sasrec"):
     from replay_amd.data.nn import TensorFeatureInfo, TensorSchema
     from replay_amd.data.schema import FeatureHint, FeatureType
     from replay_amd.nn.loss import CE
+    from replay_amd.nn.sequential.bert4rec import Bert4Rec
     from replay_amd.nn.sequential.sasrec import SasRec
 
     schema = TensorSchema(
@@ -63,7 +73,8 @@ def build_model(device):
             )
         ]
     )
-    model = SasRec.from_params(
+    cls = Bert4Rec if model_name == "bert4rec" else SasRec
+    model = cls.from_params(
         schema,
         max_sequence_length=SEQ_LEN,
         embedding_dim=EMB_DIM,
@@ -75,7 +86,7 @@ def build_model(device):
     return model
 
 
-def make_batches(n_batches, batch_size, device, seed):
+def make_batches(n_batches, batch_size, device, seed, model_name="sasrec"):
     """Synthetic ML-20M-shape sequence batches, generated on device."""
     gen = torch.Generator(device="cpu").manual_seed(seed)
     batches = []
@@ -87,6 +98,11 @@ def make_batches(n_batches, batch_size, device, seed):
             "padding_mask": torch.ones(batch_size, SEQ_LEN, dtype=torch.bool, device=device),
         }
         batch["labels_padding_mask"] = batch["padding_mask"]
+        if model_name == "bert4rec":  # masked-token objective (15% + last)
+            tm = torch.rand(batch_size, SEQ_LEN, generator=gen) < 0.15
+            tm[:, -1] = True
+            batch["labels"] = batch["item_id"].clone()
+            batch["token_mask"] = tm.to(device)
         batches.append(batch)
     return batches
 
@@ -259,6 +275,8 @@ def main() -> None:
     parser.add_argument("--tunableop", action="store_true", help="(default on)")
     parser.add_argument("--no-tunableop", action="store_true", help="disable rocBLAS TunableOp")
     parser.add_argument("--mode", choices=["train", "serve", "itemknn"], default="train")
+    parser.add_argument("--model", choices=["sasrec", "bert4rec"], default="sasrec",
+                        help="train mode: sasrec (config 2) or bert4rec (config 3 shape)")
     parser.add_argument(
         "--graphs",
         action="store_true",
@@ -291,7 +309,8 @@ def main() -> None:
             torch.distributed.destroy_process_group()
         return
 
-    model = build_model(device)
+    apply_model_config(args)
+    model = build_model(device, args.model)
     if world > 1:
         model = torch.nn.parallel.DistributedDataParallel(
             model,
@@ -302,7 +321,7 @@ def main() -> None:
     optimizer = torch.optim.Adam(
         model.parameters(), lr=args.lr, capturable=args.graphs and use_cuda and world == 1
     )
-    batches = make_batches(4, args.batch, device, seed=1000 + rank)
+    batches = make_batches(4, args.batch, device, seed=1000 + rank, model_name=args.model)
 
     amp_dtype = torch.bfloat16
     autocast = torch.autocast(device_type=device.type, dtype=amp_dtype, enabled=use_cuda)
@@ -384,7 +403,7 @@ def main() -> None:
                     "dtype": "bf16" if use_cuda else "fp32",
                     "data": "synthetic",
                     "config": {
-                        "model": f"sasrec_{N_BLOCKS}blocks_d{EMB_DIM}",
+                        "model": f"{args.model}_{N_BLOCKS}blocks_d{EMB_DIM}",
                         "global_batch": global_batch,
                         "seq_len": SEQ_LEN,
                         "n_items": N_ITEMS,
