@@ -145,11 +145,16 @@ class RandomSplitter(Splitter):
         self.seed = seed
 
     def _core_split(self, interactions: pd.DataFrame) -> SplitterReturnType:
-        rng = np.random.default_rng(self.seed)
-        mask = rng.random(len(interactions)) < self.test_size
-        is_test = pd.Series(mask, index=interactions.index)
-        is_test = self._recalculate_with_session_id_column(interactions, is_test)
-        return interactions[~is_test], interactions[is_test]
+        # the reference's exact sampling call (random_splitter.py:56):
+        # pandas .sample with the seed, so the split AND row order match
+        train = interactions.sample(frac=(1 - self.test_size), random_state=self.seed)
+        test = interactions.drop(train.index)
+        if self.session_id_column:
+            is_test = pd.Series(False, index=interactions.index)
+            is_test[test.index] = True
+            is_test = self._recalculate_with_session_id_column(interactions, is_test)
+            return interactions[~is_test], interactions[is_test]
+        return train, test
 
 
 class NewUsersSplitter(Splitter):
@@ -196,10 +201,11 @@ class ColdUserRandomSplitter(Splitter):
         self.seed = seed
 
     def _core_split(self, interactions: pd.DataFrame) -> SplitterReturnType:
-        users = interactions[self.query_column].unique()
-        rng = np.random.default_rng(self.seed)
-        test_users = set(rng.choice(users, size=int(len(users) * self.test_size), replace=False).tolist())
-        is_test = interactions[self.query_column].isin(test_users)
+        # the reference's exact sampling (cold_user_random_splitter.py:64):
+        # a pandas .sample over the unique-user frame with the seed
+        users = pd.DataFrame(interactions[self.query_column].unique(), columns=[self.query_column])
+        train_users = set(users.sample(frac=(1 - self.test_size), random_state=self.seed)[self.query_column])
+        is_test = ~interactions[self.query_column].isin(train_users)
         return interactions[~is_test], interactions[is_test]
 
 

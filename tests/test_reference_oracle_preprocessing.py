@@ -175,3 +175,28 @@ class TestPreprocessingOracle:
         r = RefDisc([RefQ("rating", n_bins=4)]).fit_transform(df)
         o = Discretizer([QuantileDiscretizingRule("rating", n_bins=4)]).fit_transform(df)
         assert r["rating"].tolist() == o["rating"].tolist(), f"seed {seed}"
+
+
+class TestRandomSplitterOracle:
+    """Seeded random splitters use the reference's exact pandas sampling
+    calls, so even the random splits match the oracle."""
+
+    @pytest.mark.parametrize("seed", range(5))
+    def test_random_and_cold_user_match(self, reference, seed):
+        from replay.splitters import (
+            ColdUserRandomSplitter as RefCU, RandomSplitter as RefRS,
+        )
+
+        from replay_amd.splitters import ColdUserRandomSplitter, RandomSplitter
+
+        df = _frames(seed + 50)
+        rt, re_ = RefRS(test_size=0.3, seed=seed).split(df)
+        ot, oe = RandomSplitter(test_size=0.3, seed=seed).split(df)
+        assert sorted(rt.index) == sorted(ot.index)
+        assert sorted(re_.index) == sorted(oe.index)
+
+        rt, re_ = RefCU(test_size=0.3, seed=seed, query_column="query_id").split(df)
+        ot, oe = ColdUserRandomSplitter(test_size=0.3, seed=seed, query_column="query_id").split(df)
+        assert set(rt["query_id"]) == set(ot["query_id"])
+        assert set(re_["query_id"]) == set(oe["query_id"])
+        assert len(rt) == len(ot) and len(re_) == len(oe)
