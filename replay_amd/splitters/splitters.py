@@ -319,12 +319,10 @@ class KFolds:
         self.query_column = query_column
 
     def split(self, interactions: pd.DataFrame) -> Iterator[Tuple[pd.DataFrame, pd.DataFrame]]:
-        rng = np.random.default_rng(self.seed)
-        fold = np.empty(len(interactions), dtype=np.int64)
-        df = interactions.reset_index(drop=True)
-        for _, group in df.groupby(self.query_column):
-            idx = group.index.to_numpy()
-            fold[idx] = rng.permutation(len(idx)) % self.n_folds
+        # the reference's exact fold assignment (k_folds.py:104): global
+        # seeded shuffle, per-user cumcount + 1 modulo n_folds
+        df = interactions.sample(frac=1, random_state=self.seed).sort_values(self.query_column)
+        fold = (df.groupby(self.query_column, sort=False).cumcount() + 1) % self.n_folds
         for k in range(self.n_folds):
             mask = fold == k
-            yield interactions[~mask], interactions[mask]
+            yield df[~mask], df[mask]

@@ -200,3 +200,16 @@ class TestRandomSplitterOracle:
         assert set(rt["query_id"]) == set(ot["query_id"])
         assert set(re_["query_id"]) == set(oe["query_id"])
         assert len(rt) == len(ot) and len(re_) == len(oe)
+
+    @pytest.mark.parametrize("seed", range(4))
+    def test_kfolds_match(self, reference, seed):
+        from replay.splitters import KFolds as RefKFolds
+
+        from replay_amd.splitters import KFolds
+
+        df = _frames(seed + 80)
+        ref_folds = RefKFolds(n_folds=3, seed=seed, query_column="query_id").split(df)
+        our_folds = KFolds(n_folds=3, seed=seed, query_column="query_id").split(df)
+        for (rt, re_), (ot, oe) in zip(ref_folds, our_folds):
+            assert sorted(rt.index) == sorted(ot.index)
+            assert sorted(re_.index) == sorted(oe.index)
