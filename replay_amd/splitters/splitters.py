@@ -11,7 +11,7 @@ KFolds (k_folds.py:16).
 
 from __future__ import annotations
 
-from typing import Iterator, Optional, Tuple
+from typing import Iterator, Optional, Tuple, Union
 
 import numpy as np
 import pandas as pd
@@ -244,53 +244,81 @@ class RandomNextNSplitter(Splitter):
 
 
 class TwoStageSplitter(Splitter):
-    """Split by users first (``second_divide_size`` share of users), then for
-    those users take ``first_divide_size`` of interactions to test
-    (reference two_stage_splitter.py:17)."""
+    """Two-stage split (reference two_stage_splitter.py:17, exact pandas
+    algorithm): ``first_divide_size`` selects the TEST USERS (count if >= 1,
+    fraction otherwise; seeded pandas .sample over the sorted unique users);
+    ``second_divide_size`` then takes each test user's newest interactions —
+    a per-user fraction when in [0, 1), or a fixed count when an int >= 1.
+    ``shuffle=True`` ranks randomly instead of by recency."""
 
-    _init_arg_names = Splitter._init_arg_names + ("second_divide_size", "first_divide_size", "seed")
+    _init_arg_names = Splitter._init_arg_names + (
+        "first_divide_size", "second_divide_size", "first_divide_column", "shuffle", "seed",
+    )
 
     def __init__(
         self,
-        second_divide_size: float = 0.5,
-        first_divide_size: float = 0.5,
+        first_divide_size: Union[float, int] = 0.5,
+        second_divide_size: Union[float, int] = 0.5,
+        first_divide_column: str = "query_id",
         shuffle: bool = False,
         seed: Optional[int] = None,
         **kwargs,
     ) -> None:
         super().__init__(**kwargs)
-        self.second_divide_size = second_divide_size
         self.first_divide_size = first_divide_size
+        self.second_divide_size = second_divide_size
+        self.first_divide_column = first_divide_column
         self.shuffle = shuffle
         self.seed = seed
 
-    def _core_split(self, interactions: pd.DataFrame) -> SplitterReturnType:
-        users = np.sort(interactions[self.query_column].unique())
-        rng = np.random.default_rng(self.seed)
-        n_test_users = (
-            int(self.second_divide_size)
-            if self.second_divide_size >= 1
-            else int(len(users) * self.second_divide_size)
+    def _get_test_values(self, interactions: pd.DataFrame) -> pd.DataFrame:
+        all_values = pd.DataFrame(
+            np.sort(interactions[self.first_divide_column].unique()),
+            columns=[self.first_divide_column],
         )
-        test_users = set(rng.choice(users, size=n_test_users, replace=False).tolist())
-        in_test_users = interactions[self.query_column].isin(test_users)
-        df_test_users = interactions[in_test_users]
+        user_count = len(all_values)
+        if isinstance(self.first_divide_size, int) and not isinstance(self.first_divide_size, bool):
+            if not 1 <= self.first_divide_size < user_count:
+                raise ValueError(f"Invalid value for user_test_size: {self.first_divide_size}")
+            test_user_count = self.first_divide_size
+        else:
+            if not 0 < self.first_divide_size < 1:
+                raise ValueError(f"Invalid value for user_test_size: {self.first_divide_size}")
+            test_user_count = user_count * self.first_divide_size
+        return all_values.sample(n=int(test_user_count), random_state=self.seed)
+
+    def _partition(self, merged: pd.DataFrame) -> pd.DataFrame:
         if self.shuffle:
-            df_test_users = df_test_users.sample(frac=1.0, random_state=self.seed)
+            res = merged.sample(frac=1, random_state=self.seed).sort_values(self.first_divide_column)
+            res["_row_num"] = res.groupby(self.first_divide_column, sort=False).cumcount() + 1
+            return res
+        res = merged.copy(deep=True)
+        res.sort_values([self.query_column, self.timestamp_column], ascending=[True, False], inplace=True)
+        res["_row_num"] = res.groupby(self.query_column, sort=False).cumcount() + 1
+        return res
+
+    def _core_split(self, interactions: pd.DataFrame) -> SplitterReturnType:
+        test_users = self._get_test_values(interactions)
+        test_users = test_users.assign(is_test=True)
+        res = self._partition(interactions.merge(test_users, how="left", on=self.first_divide_column))
+        res["is_test"] = res["is_test"].notna() & res["is_test"].eq(True)
+        drop_cols = ["_row_num", "is_test"]
+        if 0 <= self.second_divide_size < 1.0:
+            counts = res.groupby(self.first_divide_column)[self.first_divide_column].transform("size")
+            frac = res["_row_num"] / counts
+            is_test = (frac <= self.second_divide_size) & res["is_test"]
+            keep_train = (frac > self.second_divide_size) | (~res["is_test"])
+        elif self.second_divide_size >= 1 and isinstance(self.second_divide_size, int):
+            is_test = (res["_row_num"] <= self.second_divide_size) & res["is_test"]
+            keep_train = (res["_row_num"] > self.second_divide_size) | (~res["is_test"])
         else:
-            df_test_users = df_test_users.sort_values(
-                [self.query_column, self.timestamp_column], kind="stable"
+            raise ValueError(
+                f"`test_size` value must be [0, 1) or a positive integer; "
+                f"test_size={self.second_divide_size}"
             )
-        sizes = df_test_users.groupby(self.query_column)[self.query_column].transform("size")
-        pos = df_test_users.groupby(self.query_column).cumcount()
-        if self.first_divide_size >= 1:
-            n_test = np.minimum(sizes - 1, int(self.first_divide_size))
-        else:
-            n_test = np.floor(sizes * self.first_divide_size).astype(int)
-        is_test = pos >= (sizes - n_test)
-        test = df_test_users[is_test]
-        train = pd.concat([interactions[~in_test_users], df_test_users[~is_test]]).sort_index()
-        return train, test.sort_index()
+        train = res[keep_train].drop(columns=drop_cols)
+        test = res[is_test].drop(columns=drop_cols)
+        return train, test
 
 
 class KFolds:

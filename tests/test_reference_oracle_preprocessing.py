@@ -213,3 +213,18 @@ class TestRandomSplitterOracle:
         for (rt, re_), (ot, oe) in zip(ref_folds, our_folds):
             assert sorted(rt.index) == sorted(ot.index)
             assert sorted(re_.index) == sorted(oe.index)
+
+    @pytest.mark.parametrize("seed", range(4))
+    def test_two_stage_splitter_matches(self, reference, seed):
+        from replay.splitters import TwoStageSplitter as RefTwoStage
+
+        from replay_amd.splitters import TwoStageSplitter
+
+        df = _frames(seed + 30)
+        for fd, sd in [(2, 3), (0.4, 2), (0.4, 0.6)]:
+            rt, re_ = RefTwoStage(first_divide_size=fd, second_divide_size=sd,
+                                  seed=seed, query_column="query_id").split(df)
+            ot, oe = TwoStageSplitter(first_divide_size=fd, second_divide_size=sd,
+                                      seed=seed, query_column="query_id").split(df)
+            assert _same_rows(rt, ot), (fd, sd, seed)
+            assert _same_rows(re_, oe), (fd, sd, seed)
