@@ -222,25 +222,23 @@ class RandomNextNSplitter(Splitter):
         self.divide_column = divide_column or self.query_column
 
     def _core_split(self, interactions: pd.DataFrame) -> SplitterReturnType:
-        df = interactions.sort_values([self.divide_column, self.timestamp_column], kind="stable")
-        rng = np.random.default_rng(self.seed)
-        sizes = df.groupby(self.divide_column)[self.divide_column].transform("size").to_numpy()
-        pos = df.groupby(self.divide_column).cumcount().to_numpy()
-        # one random cut per user, broadcast to rows
-        users, first_idx = np.unique(df[self.divide_column].to_numpy(), return_index=True)
-        cuts_per_user = {}
-        for u, fi in zip(users, first_idx):
-            size = sizes[fi]
-            cuts_per_user[u] = rng.integers(1, max(2, size))
-        cut = df[self.divide_column].map(cuts_per_user).to_numpy()
-        is_test = (pos >= cut) & (pos < cut + self.N)
-        is_test = pd.Series(is_test, index=df.index)
+        """Exact reference algorithm (random_next_n_splitter.py:121-146):
+        cut = RandomState(seed).randint(0, count) per divide group (group
+        order = first occurrence in the time-sorted frame; a cut of 0 sends
+        the whole sequence to test), rows past cut+N are dropped, and rows
+        at rank >= cut go to test."""
+        df = interactions.sort_values([self.divide_column, self.timestamp_column])
+        rank = df.groupby(self.divide_column, sort=False).cumcount()
+        counts = df.groupby(self.divide_column, sort=False).size()
+        rng = np.random.RandomState(self.seed)
+        cuts = pd.Series(rng.randint(0, counts.values), index=counts.index)
+        cut = df[self.divide_column].map(cuts)
+        if self.N is not None:
+            keep = rank < cut + self.N
+            df, rank, cut = df[keep], rank[keep], cut[keep]
+        is_test = rank >= cut
         is_test = self._recalculate_with_session_id_column(df, is_test)
-        # interactions after cut+N are dropped (they are future relative to test)
-        keep = pd.Series(pos < cut + self.N, index=df.index)
-        df = df[keep]
-        is_test = is_test[keep]
-        return df[~is_test].sort_index(), df[is_test].sort_index()
+        return df[~is_test][interactions.columns], df[is_test][interactions.columns]
 
 
 class TwoStageSplitter(Splitter):

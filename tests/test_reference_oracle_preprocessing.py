@@ -228,3 +228,16 @@ class TestRandomSplitterOracle:
                                       seed=seed, query_column="query_id").split(df)
             assert _same_rows(rt, ot), (fd, sd, seed)
             assert _same_rows(re_, oe), (fd, sd, seed)
+
+    @pytest.mark.parametrize("seed", range(4))
+    def test_random_next_n_matches(self, reference, seed):
+        from replay.splitters import RandomNextNSplitter as RefRN
+
+        from replay_amd.splitters import RandomNextNSplitter
+
+        df = _frames(seed + 60)
+        rt, re_ = RefRN(N=2, seed=seed, divide_column="query_id", query_column="query_id").split(df)
+        ot, oe = RandomNextNSplitter(N=2, seed=seed, divide_column="query_id",
+                                     query_column="query_id").split(df)
+        assert sorted(rt.index) == sorted(ot.index)
+        assert sorted(re_.index) == sorted(oe.index)
