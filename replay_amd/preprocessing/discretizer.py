@@ -83,18 +83,59 @@ class GreedyDiscretizingRule(BaseDiscretizingRule):
         self._upper_bounds: Optional[np.ndarray] = None
 
     def fit(self, df: pd.DataFrame) -> "GreedyDiscretizingRule":
+        """LightGBM-style greedy histogram binning (the reference's exact
+        algorithm, discretizer.py:110-193): values whose count exceeds the
+        running mean bin size become single-value bins; the rest fill bins
+        to the re-estimated mean, closing early before a heavy value."""
         values = df[self.column].dropna().to_numpy(dtype=np.float64)
         uniq, counts = np.unique(values, return_counts=True)
-        total = counts.sum()
-        target = max(total / self.n_bins, self.min_data_in_bin)
+        total = int(counts.sum())
+        m = len(uniq)
+        max_bin = self.n_bins + 1
         bounds: List[float] = []
-        acc = 0
-        for v, c in zip(uniq, counts):
-            acc += c
-            if acc >= target and len(bounds) < self.n_bins - 1:
-                bounds.append(float(v))
-                acc = 0
-        self._upper_bounds = np.asarray(bounds)
+        if m <= max_bin:
+            acc = 0
+            for i in range(m - 1):
+                acc += counts[i]
+                if acc >= self.min_data_in_bin:
+                    bounds.append(float((uniq[i] + uniq[i + 1]) / 2.0))
+                    acc = 0
+        else:
+            if self.min_data_in_bin > 0:
+                max_bin = max(1, min(max_bin, total // self.min_data_in_bin))
+            mean_size = total / max_bin
+            heavy = counts >= mean_size
+            light_bins = max_bin - int(heavy.sum())
+            light_samples = total - int(counts[heavy].sum())
+            mean_size = light_samples / light_bins
+            uppers = np.full(max_bin, np.inf)
+            lowers = np.full(max_bin, np.inf)
+            n_closed = 0
+            lowers[0] = uniq[0]
+            acc = 0
+            for i in range(m - 1):
+                if not heavy[i]:
+                    light_samples -= counts[i]
+                acc += counts[i]
+                close = (
+                    heavy[i]
+                    or acc >= mean_size
+                    or (heavy[i + 1] and acc >= max(1.0, mean_size * 0.5))
+                )
+                if close:
+                    uppers[n_closed] = uniq[i]
+                    n_closed += 1
+                    lowers[n_closed] = uniq[i + 1]
+                    if n_closed >= max_bin - 1:
+                        break
+                    acc = 0
+                    if not heavy[i]:
+                        light_bins -= 1
+                        mean_size = light_samples / light_bins
+            # the last closed bin merges with the tail (reference: midpoints
+            # only for the first n_closed-1 closures, then +inf)
+            bounds = [float((uppers[i] + lowers[i + 1]) / 2.0) for i in range(n_closed - 1)]
+        self._upper_bounds = np.asarray(bounds, dtype=np.float64)
         self.is_fitted = True
         return self
 

@@ -166,6 +166,21 @@ class TestPreprocessingOracle:
         assert np.allclose(np.asarray(r.todense()), np.asarray(o.todense()))
 
     @pytest.mark.parametrize("seed", range(4))
+    def test_greedy_discretizer_matches(self, reference, seed):
+        from replay.preprocessing import Discretizer as RefDisc, GreedyDiscretizingRule as RefG
+
+        from replay_amd.preprocessing import Discretizer, GreedyDiscretizingRule
+
+        df = _frames(seed)
+        r = RefDisc([RefG("rating", n_bins=4)]).fit_transform(df)
+        o = Discretizer([GreedyDiscretizingRule("rating", n_bins=4)]).fit_transform(df)
+        assert r["rating"].tolist() == o["rating"].tolist(), f"seed {seed}"
+        heavy = pd.DataFrame({"rating": [1.0] * 30 + list(np.random.default_rng(seed).random(30) * 5)})
+        r = RefDisc([RefG("rating", n_bins=4)]).fit_transform(heavy)
+        o = Discretizer([GreedyDiscretizingRule("rating", n_bins=4)]).fit_transform(heavy)
+        assert r["rating"].tolist() == o["rating"].tolist(), f"heavy seed {seed}"
+
+    @pytest.mark.parametrize("seed", range(4))
     def test_quantile_discretizer_matches(self, reference, seed):
         from replay.preprocessing import Discretizer as RefDisc, QuantileDiscretizingRule as RefQ
 
