@@ -256,3 +256,35 @@ class TestRandomSplitterOracle:
                                      query_column="query_id").split(df)
         assert sorted(rt.index) == sorted(ot.index)
         assert sorted(re_.index) == sorted(oe.index)
+
+
+class TestDatasetLabelEncoderOracle:
+    @pytest.mark.parametrize("seed", range(3))
+    def test_dataset_encoding_matches(self, reference, seed):
+        from replay.data import (
+            Dataset as RefDS, FeatureHint as RFH, FeatureInfo as RFI,
+            FeatureSchema as RFS, FeatureType as RFT,
+        )
+        from replay.data.dataset_utils import DatasetLabelEncoder as RefDLE
+
+        from replay_amd.data import Dataset, FeatureHint, FeatureInfo, FeatureSchema, FeatureType
+        from replay_amd.data.dataset_utils import DatasetLabelEncoder
+
+        rng = np.random.default_rng(seed)
+        df = pd.DataFrame(
+            {
+                "query_id": [f"u{i}" for i in rng.integers(0, 6, 30)],
+                "item_id": [f"i{i}" for i in rng.integers(0, 9, 30)],
+                "rating": rng.random(30),
+            }
+        )
+        rs = RFS([RFI("query_id", RFT.CATEGORICAL, RFH.QUERY_ID),
+                  RFI("item_id", RFT.CATEGORICAL, RFH.ITEM_ID),
+                  RFI("rating", RFT.NUMERICAL, RFH.RATING)])
+        os_ = FeatureSchema([FeatureInfo("query_id", FeatureType.CATEGORICAL, FeatureHint.QUERY_ID),
+                             FeatureInfo("item_id", FeatureType.CATEGORICAL, FeatureHint.ITEM_ID),
+                             FeatureInfo("rating", FeatureType.NUMERICAL, FeatureHint.RATING)])
+        r = RefDLE().fit_transform(RefDS(feature_schema=rs, interactions=df))
+        o = DatasetLabelEncoder().fit_transform(Dataset(feature_schema=os_, interactions=df))
+        assert r.interactions["query_id"].tolist() == o.interactions["query_id"].tolist()
+        assert r.interactions["item_id"].tolist() == o.interactions["item_id"].tolist()
