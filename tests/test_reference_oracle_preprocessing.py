@@ -288,3 +288,32 @@ class TestDatasetLabelEncoderOracle:
         o = DatasetLabelEncoder().fit_transform(Dataset(feature_schema=os_, interactions=df))
         assert r.interactions["query_id"].tolist() == o.interactions["query_id"].tolist()
         assert r.interactions["item_id"].tolist() == o.interactions["item_id"].tolist()
+
+
+class TestExperimentalPreprocessingOracle:
+    @pytest.mark.parametrize("seed", range(4))
+    def test_padder_and_sequence_generator_match(self, reference, seed):
+        from replay.experimental.preprocessing import (
+            Padder as RefPadder, SequenceGenerator as RefSeqGen,
+        )
+
+        from replay_amd.experimental.preprocessing import Padder, SequenceGenerator
+
+        rng = np.random.default_rng(seed)
+        df = pd.DataFrame({"u": rng.integers(0, 5, 30), "i": rng.integers(0, 10, 30),
+                           "t": np.arange(30)})
+        kwargs = dict(groupby_column="u", orderby_column="t", transform_columns=["i"],
+                      len_window=4, get_list_len=True)
+        r = RefSeqGen(**kwargs).transform(df)
+        o = SequenceGenerator(**kwargs).transform(df)
+        assert r["i_list"].tolist() == o["i_list"].tolist()
+        assert r["label_i"].tolist() == o["label_i"].tolist()
+        assert r["list_len"].tolist() == o["list_len"].tolist()
+
+        lists = pd.DataFrame(
+            {"xs": [list(rng.integers(0, 9, int(rng.integers(1, 7)))) for _ in range(8)]}
+        )
+        pk = dict(padding_side="left", padding_value=-1, array_size=4, cut_side="right")
+        rp = RefPadder("xs", **pk).transform(lists)
+        op = Padder("xs", **pk).transform(lists)
+        assert [list(x) for x in rp["xs"]] == [list(x) for x in op["xs"]]
