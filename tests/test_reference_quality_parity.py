@@ -342,3 +342,31 @@ class TestTiSasRecQualityParity:
         # both learn the pattern well; TI attention variants differ slightly in
         # convergence speed at this tiny scale
         assert our_hit >= ref_hit - 0.07, f"ours {our_hit:.2f} vs reference {ref_hit:.2f}"
+
+
+class TestExperimentOracle:
+    def test_experiment_table_matches(self, reference):
+        import numpy as np
+        import pandas as pd
+
+        from replay.metrics import Experiment as RefExp, HitRate as RefHR, NDCG as RefNDCG
+
+        from replay_amd.metrics import Experiment, HitRate, NDCG
+
+        rng = np.random.default_rng(0)
+        recs = pd.DataFrame(
+            {"query_id": rng.integers(0, 6, 40), "item_id": rng.integers(0, 20, 40),
+             "rating": rng.random(40)}
+        ).drop_duplicates(["query_id", "item_id"])
+        recs2 = recs.assign(rating=1.0 - recs["rating"])
+        gt = pd.DataFrame(
+            {"query_id": rng.integers(0, 6, 25), "item_id": rng.integers(0, 20, 25)}
+        ).drop_duplicates()
+        ref_exp = RefExp([RefNDCG(3), RefHR(3)], gt)
+        our_exp = Experiment([NDCG(3), HitRate(3)], gt)
+        for name, frame in [("a", recs), ("b", recs2)]:
+            ref_exp.add_result(name, frame)
+            our_exp.add_result(name, frame)
+        pd.testing.assert_frame_equal(
+            ref_exp.results.sort_index(), our_exp.results.sort_index(), check_dtype=False
+        )
