@@ -495,3 +495,42 @@ class TestMfmaAttentionBwd:
         torch.testing.assert_close(dq.float(), q2.grad, atol=8e-2, rtol=8e-2)
         torch.testing.assert_close(dk.float(), k2.grad, atol=8e-2, rtol=8e-2)
         torch.testing.assert_close(dv.float(), v2.grad, atol=8e-2, rtol=8e-2)
+
+
+@requires_gpu
+class TestFusedLinearCEDhFusion:
+    def test_fuse_dh_env_path_matches_default(self):
+        """REPLAY_AMD_CE_FUSE_DH=1 (the round-2 in-kernel dhidden fusion)
+        must be numerically equivalent to the shipped path.  Run in a
+        subprocess: the launcher caches the env flag on first use."""
+        import os
+        import subprocess
+        import sys
+
+        script = r"""
+import torch
+from replay_amd.ops.autograd import fused_linear_cross_entropy
+
+torch.manual_seed(0)
+N, E, V = 513, 64, 1003
+hidden = (torch.randn(N, E, device="cuda") * 0.5).to(torch.bfloat16).requires_grad_(True)
+weight = (torch.randn(V, E, device="cuda") * 0.5).to(torch.bfloat16).requires_grad_(True)
+labels = torch.randint(0, V, (N,), device="cuda")
+labels[::5] = -100
+loss = fused_linear_cross_entropy(hidden, weight, labels)
+loss.backward()
+h_ref = hidden.detach().float().clone().requires_grad_(True)
+w_ref = weight.detach().float().clone().requires_grad_(True)
+ref = torch.nn.functional.cross_entropy(h_ref @ w_ref.t(), labels, ignore_index=-100)
+ref.backward()
+assert abs(float(loss) - float(ref)) < 3e-2 * max(1.0, abs(float(ref)))
+torch.testing.assert_close(hidden.grad.float(), h_ref.grad, atol=6e-3, rtol=5e-2)
+torch.testing.assert_close(weight.grad.float(), w_ref.grad, atol=6e-3, rtol=5e-2)
+print("FUSE_DH OK")
+"""
+        env = dict(os.environ, REPLAY_AMD_CE_FUSE_DH="1")
+        proc = subprocess.run(
+            [sys.executable, "-c", script], capture_output=True, text=True, timeout=300, env=env
+        )
+        assert proc.returncode == 0, proc.stderr[-1500:]
+        assert "FUSE_DH OK" in proc.stdout
