@@ -7,7 +7,17 @@ MI355X transformer implementation underneath.
 
 from replay_amd.utils import TORCH_AVAILABLE
 
-from .sequential.bert4rec import Bert4Rec
+from .sequential.bert4rec import (
+    Bert4Rec,
+    Bert4RecModel,
+    Bert4RecPredictionBatch,
+    Bert4RecPredictionDataset,
+    Bert4RecTrainingBatch,
+    Bert4RecTrainingDataset,
+    Bert4RecUniformMasker,
+    Bert4RecValidationBatch,
+    Bert4RecValidationDataset,
+)
 from .sequential.compiled import Bert4RecCompiled, SasRecCompiled
 from .sequential.sasrec import (
     SasRec,

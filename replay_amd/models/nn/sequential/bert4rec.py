@@ -86,3 +86,139 @@ class Bert4Rec(LightningModule):
         loss = self._model(batch)
         self.log("train_loss", loss, sync_dist=True)
         return loss
+
+
+# ---------------------------------------------------------------------------
+# Legacy dataset/batch surface (reference models/nn/sequential/bert4rec/
+# dataset.py): maskers, masked-token training datasets and batch tuples.
+# Note the reference's token_mask polarity on this legacy surface: 0 = MASK,
+# 1 = kept (the new-generation transform uses True = masked).
+# ---------------------------------------------------------------------------
+from typing import Dict, NamedTuple  # noqa: E402
+
+from replay_amd.data.nn import (  # noqa: E402
+    TorchSequentialDataset,
+    TorchSequentialValidationDataset,
+)
+
+Bert4RecModel = _NewGenBert4Rec  # the raw torch module (reference model.py)
+
+
+class Bert4RecMasker:
+    """Masking-strategy base (reference dataset.py:40)."""
+
+    def mask(self, paddings: torch.Tensor) -> torch.Tensor:  # pragma: no cover
+        raise NotImplementedError
+
+
+class Bert4RecUniformMasker(Bert4RecMasker):
+    """Uniform token masking over the valid positions (reference
+    dataset.py:55): returns a mask where 0 = <MASK>, 1 = kept."""
+
+    def __init__(self, mask_prob: float = 0.15, generator=None) -> None:
+        self.mask_prob = mask_prob
+        self.generator = generator
+
+    def mask(self, paddings: torch.Tensor) -> torch.Tensor:
+        rand = torch.rand(paddings.shape, generator=self.generator)
+        masked = (rand < self.mask_prob) & paddings
+        if paddings.any() and not masked.any():
+            # always mask at least the last valid position
+            last = int(paddings.nonzero()[-1])
+            masked[..., last] = True
+        return ~masked
+
+
+class Bert4RecTrainingBatch(NamedTuple):
+    """Legacy tuple view of a training batch (reference dataset.py)."""
+
+    query_id: torch.LongTensor
+    padding_mask: torch.BoolTensor
+    features: Dict[str, torch.Tensor]
+    token_mask: torch.BoolTensor
+    labels: torch.LongTensor
+
+
+class Bert4RecPredictionBatch(NamedTuple):
+    query_id: torch.LongTensor
+    padding_mask: torch.BoolTensor
+    features: Dict[str, torch.Tensor]
+    token_mask: torch.BoolTensor
+
+    def convert_to_dict(self) -> dict:
+        out = {"query_id": self.query_id, "padding_mask": self.padding_mask,
+               "token_mask": self.token_mask}
+        out.update(self.features)
+        return out
+
+
+class Bert4RecValidationBatch(NamedTuple):
+    query_id: torch.LongTensor
+    padding_mask: torch.BoolTensor
+    features: Dict[str, torch.Tensor]
+    token_mask: torch.BoolTensor
+    ground_truth: torch.LongTensor
+    train: torch.LongTensor
+
+
+class Bert4RecTrainingDataset(torch.utils.data.Dataset):
+    """Masked-token training samples (reference dataset.py:95): each item
+    carries the full labels plus a token mask drawn by the masker; ours
+    emits the dict keys the legacy Bert4Rec consumes (token_mask True =
+    masked, converted from the masker's 0-=-mask polarity)."""
+
+    def __init__(
+        self,
+        sequential,
+        max_sequence_length: int,
+        mask_prob: float = 0.15,
+        sliding_window_step=None,
+        label_feature_name=None,
+        custom_masker: Optional[Bert4RecMasker] = None,
+    ) -> None:
+        self._label_name = label_feature_name or sequential.schema.item_id_feature_name
+        self._masker = custom_masker or Bert4RecUniformMasker(mask_prob)
+        self._inner = TorchSequentialDataset(
+            sequential, max_sequence_length, sliding_window_step=sliding_window_step
+        )
+
+    def __len__(self) -> int:
+        return len(self._inner)
+
+    def __getitem__(self, index: int) -> dict:
+        item = dict(self._inner[index])
+        kept = self._masker.mask(item["padding_mask"])
+        item["token_mask"] = ~kept  # our convention: True = masked
+        item["labels"] = item[self._label_name]
+        item["labels_padding_mask"] = item["padding_mask"]
+        return item
+
+
+class Bert4RecPredictionDataset(torch.utils.data.Dataset):
+    """Inference samples (reference dataset.py:195 flow)."""
+
+    def __init__(self, sequential, max_sequence_length: int) -> None:
+        self._inner = TorchSequentialDataset(sequential, max_sequence_length)
+
+    def __len__(self) -> int:
+        return len(self._inner)
+
+    def __getitem__(self, index: int) -> dict:
+        return dict(self._inner[index])
+
+
+class Bert4RecValidationDataset(torch.utils.data.Dataset):
+    """Validation samples carrying ground_truth + train ids."""
+
+    def __init__(self, sequential, ground_truth, train, max_sequence_length: int,
+                 label_feature_name=None) -> None:
+        self._inner = TorchSequentialValidationDataset(
+            sequential, ground_truth=ground_truth, train=train,
+            max_sequence_length=max_sequence_length, label_feature_name=label_feature_name,
+        )
+
+    def __len__(self) -> int:
+        return len(self._inner)
+
+    def __getitem__(self, index: int) -> dict:
+        return dict(self._inner[index])

@@ -437,3 +437,28 @@ class TestReferenceDoctestParity:
         assert out == {
             "NDCG-PerUser@2": {1: 0.38685280723454163, 2: 0.0, 3: 0.6131471927654584}
         }
+
+    def test_legacy_bert4rec_training_dataset(self):
+        from replay_amd.data import Dataset, FeatureHint, FeatureInfo, FeatureSchema, FeatureType
+        from replay_amd.data.nn import SequenceTokenizer, TensorFeatureInfo, TensorSchema
+        from replay_amd.models.nn import Bert4RecTrainingDataset
+
+        rng = np.random.default_rng(0)
+        rows = [(q, int(rng.integers(0, 20)), t) for q in range(5) for t in range(6)]
+        inter = pd.DataFrame(rows, columns=["query_id", "item_id", "timestamp"])
+        schema = FeatureSchema([
+            FeatureInfo("query_id", FeatureType.CATEGORICAL, FeatureHint.QUERY_ID),
+            FeatureInfo("item_id", FeatureType.CATEGORICAL, FeatureHint.ITEM_ID),
+            FeatureInfo("timestamp", FeatureType.NUMERICAL, FeatureHint.TIMESTAMP),
+        ])
+        ts = TensorSchema([TensorFeatureInfo("item_id", FeatureType.CATEGORICAL, is_seq=True,
+                                             feature_hint=FeatureHint.ITEM_ID, cardinality=20,
+                                             embedding_dim=8)])
+        seqs = SequenceTokenizer(ts).fit_transform(Dataset(feature_schema=schema, interactions=inter))
+        torch.manual_seed(0)
+        ds = Bert4RecTrainingDataset(seqs, max_sequence_length=5, mask_prob=0.3)
+        item = ds[0]
+        assert set(item) >= {"item_id", "padding_mask", "token_mask", "labels", "labels_padding_mask"}
+        # at least one valid position masked, and no padding position masked
+        assert (item["token_mask"] & item["padding_mask"]).any()
+        assert not (item["token_mask"] & ~item["padding_mask"]).any()

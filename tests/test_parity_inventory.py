@@ -112,6 +112,7 @@ INVENTORY = {
     "replay_amd.models.nn.optimizer_utils": ["OptimizerFactory", "LRSchedulerFactory", "FatOptimizerFactory", "FatLRSchedulerFactory"],
     "replay_amd.models.nn.loss": ["SCEParams", "ScalableCrossEntropyLoss"],
     "replay_amd.models.nn.sequential.sasrec": ["SasRec", "SasRecModel", "SasRecTrainingDataset", "SasRecTrainingBatch", "SasRecPredictionDataset", "SasRecPredictionBatch", "SasRecValidationDataset", "SasRecValidationBatch"],
+    "replay_amd.models.nn.sequential.bert4rec": ["Bert4Rec", "Bert4RecModel", "Bert4RecTrainingDataset", "Bert4RecTrainingBatch", "Bert4RecPredictionDataset", "Bert4RecPredictionBatch", "Bert4RecValidationDataset", "Bert4RecValidationBatch", "Bert4RecUniformMasker"],
     "replay_amd.models.nn.sequential.callbacks": ["BasePredictionCallback", "PandasPredictionCallback", "TorchPredictionCallback", "QueryEmbeddingsPredictionCallback", "ValidationMetricsCallback"],
     "replay_amd.data.nn.parquet.metadata": ["get_shape", "get_padding", "get_numeric_columns", "get_1d_array_columns", "get_2d_array_columns"],
     "replay_amd.nn.sequential": ["SasRec", "Bert4Rec", "TwoTower", "DiffTransformerLayer"],
