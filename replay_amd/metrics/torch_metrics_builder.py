@@ -4,9 +4,7 @@ Parity with reference replay/metrics/torch_metrics_builder.py:196
 (``TorchMetricsBuilder``): batch accumulation of recall / precision / ndcg /
 map / mrr / novelty (reference :306-336) from a hit matrix (broadcast compare,
 reference :344-349) plus coverage via a catalog histogram (reference
-:95-168).  Runs on any torch device; on ROCm the compare+reduce runs as one
-fused HIP-friendly op sequence (a dedicated HIP kernel backs it when the
-`replay_amd.ops` extension is loaded).
+:95-168).  Runs on any torch device with vectorized eager torch ops.
 
 Conventions: ``ground_truth`` / ``train`` are padded with -1;
 ``predictions`` hold top-max_k item ids ranked best-first.

@@ -127,70 +127,95 @@ class RMSNormPerHead(torch.nn.Module):
 
 
 class MultiHeadDifferentialAttention(torch.nn.Module):
-    """Differential attention (reference replay/nn/attention.py:7)."""
+    """Differential attention (reference replay/nn/attention.py:7-157).
+
+    Parameter layout matches the reference exactly for state-dict
+    compatibility: ``W_q``/``W_k`` project E -> 2*qk_e_head*num_heads (the
+    per-head q/k split into two components of full head width), ``W_v``
+    projects E -> v_e_head*num_heads, lambda parameters are per-head of
+    shape ``(num_heads, qk_e_head)``, and the per-head RMSNorm scale is
+    ``rms_scale`` of shape ``(v_e_head,)``.
+    """
 
     def __init__(
         self,
-        embed_dim: int,
+        embedding_dim: int,
         num_heads: int,
-        lambda_init: float = 0.8,
-        dropout: float = 0.0,
-        depth: int = 1,
+        lambda_init: float,
+        bias: bool = False,
+        kdim: Optional[int] = None,
+        vdim: Optional[int] = None,
     ) -> None:
         super().__init__()
-        if embed_dim % (2 * num_heads) != 0:
-            raise ValueError("embed_dim must divide 2*num_heads")
-        self.embed_dim = embed_dim
+        kdim = kdim or embedding_dim
+        vdim = vdim or embedding_dim
+        if kdim % num_heads != 0 or vdim % num_heads != 0:
+            raise ValueError("kdim and vdim must be divisible by num_heads")
+        self.qk_e_head = kdim // num_heads
+        self.v_e_head = vdim // num_heads
         self.num_heads = num_heads
-        self.head_dim = embed_dim // (2 * num_heads)
-        self.dropout = dropout
-        self.q_proj = torch.nn.Linear(embed_dim, embed_dim, bias=False)
-        self.k_proj = torch.nn.Linear(embed_dim, embed_dim, bias=False)
-        self.v_proj = torch.nn.Linear(embed_dim, embed_dim, bias=False)
-        self.out_proj = torch.nn.Linear(embed_dim, embed_dim, bias=False)
-        self.lambda_init = lambda_init - 0.6 * math.exp(-0.3 * (depth - 1))
-        d = self.head_dim
-        self.lambda_q1 = torch.nn.Parameter(torch.randn(d) * 0.1)
-        self.lambda_k1 = torch.nn.Parameter(torch.randn(d) * 0.1)
-        self.lambda_q2 = torch.nn.Parameter(torch.randn(d) * 0.1)
-        self.lambda_k2 = torch.nn.Parameter(torch.randn(d) * 0.1)
-        self.subln = RMSNormPerHead(2 * self.head_dim)
+        self.lambda_init = lambda_init
+        self.eps = 1e-5
+
+        self.W_q = torch.nn.Linear(embedding_dim, 2 * self.qk_e_head * num_heads, bias=bias)
+        self.W_k = torch.nn.Linear(embedding_dim, 2 * self.qk_e_head * num_heads, bias=bias)
+        self.W_v = torch.nn.Linear(embedding_dim, self.v_e_head * num_heads, bias=bias)
+        self.W_o = torch.nn.Linear(self.v_e_head * num_heads, embedding_dim, bias=bias)
+
+        self.lambda_q1 = torch.nn.Parameter(torch.randn(num_heads, self.qk_e_head))
+        self.lambda_k1 = torch.nn.Parameter(torch.randn(num_heads, self.qk_e_head))
+        self.lambda_q2 = torch.nn.Parameter(torch.randn(num_heads, self.qk_e_head))
+        self.lambda_k2 = torch.nn.Parameter(torch.randn(num_heads, self.qk_e_head))
+        self.register_buffer(
+            "scaling", torch.asarray(1.0 / math.sqrt(self.qk_e_head), dtype=torch.float32)
+        )
+        self.rms_scale = torch.nn.Parameter(torch.ones(self.v_e_head))
+
+    def reset_parameters(self) -> None:
+        for _, param in self.named_parameters():
+            if param.dim() >= 2:
+                torch.nn.init.xavier_normal_(param.data)
 
     def forward(
         self,
-        x: torch.Tensor,
-        attn_mask: Optional[torch.Tensor] = None,
-        key_padding_mask: Optional[torch.Tensor] = None,
+        query: torch.Tensor,
+        key: torch.Tensor,
+        value: torch.Tensor,
+        attn_mask: torch.Tensor,
     ) -> torch.Tensor:
+        """query/key/value: [B, L, E]; attn_mask additive float, shape
+        [B*H, L, L] or [B, H, L, L] (0 = keep, -inf = masked)."""
         from replay_amd.nn.mask import MaskSpec
 
         if isinstance(attn_mask, MaskSpec):
             attn_mask = attn_mask.materialize()
-        B, L, E = x.shape
-        H, Dh = self.num_heads, self.head_dim
-        q = self.q_proj(x).view(B, L, 2 * H, Dh).transpose(1, 2)  # [B, 2H, L, Dh]
-        k = self.k_proj(x).view(B, L, 2 * H, Dh).transpose(1, 2)
-        v = self.v_proj(x).view(B, L, H, 2 * Dh).transpose(1, 2)  # [B, H, L, 2Dh]
+        B, L, _ = value.shape
+        H, Dq, Dv = self.num_heads, self.qk_e_head, self.v_e_head
+        if attn_mask is None:
+            attn_mask = torch.zeros(B, H, L, L, dtype=torch.float32, device=value.device)
+        q = self.W_q(query).view(B, L, H, 2 * Dq).transpose(1, 2)  # [B, H, L, 2Dq]
+        k = self.W_k(key).view(B, L, H, 2 * Dq).transpose(1, 2)
+        v = self.W_v(value).view(B, L, H, Dv).transpose(1, 2)  # [B, H, L, Dv]
+        q1, q2 = q.chunk(2, dim=-1)
+        k1, k2 = k.chunk(2, dim=-1)
 
-        scores = q @ k.transpose(-1, -2) / math.sqrt(Dh)  # [B, 2H, L, L]
-        if key_padding_mask is not None:
-            fill = float("-inf") if self.training else torch.finfo(torch.float32).min
-            scores = scores.masked_fill(key_padding_mask[:, None, None, :], fill)
-        if attn_mask is not None:
-            if attn_mask.dim() == 3 and attn_mask.shape[0] == B * H:
-                am = attn_mask.view(B, H, L, L).repeat_interleave(2, dim=1)
-            else:
-                am = attn_mask
-            scores = scores + am
-        probs = torch.softmax(scores, dim=-1)
-        probs = torch.nan_to_num(probs, nan=0.0)
-        probs = probs.view(B, H, 2, L, L)
-        lam1 = torch.exp((self.lambda_q1 * self.lambda_k1).sum())
-        lam2 = torch.exp((self.lambda_q2 * self.lambda_k2).sum())
-        lam = lam1 - lam2 + self.lambda_init
-        attn = probs[:, :, 0] - lam * probs[:, :, 1]  # [B, H, L, L]
-        attn = torch.nn.functional.dropout(attn, self.dropout, self.training)
-        out = attn @ v  # [B, H, L, 2Dh]
-        out = self.subln(out) * (1.0 - self.lambda_init)
-        out = out.transpose(1, 2).reshape(B, L, E)
-        return self.out_proj(out)
+        # per-head lambda reparameterization: (H,)
+        lam = (
+            torch.exp((self.lambda_q1 * self.lambda_k1).sum(-1).float())
+            - torch.exp((self.lambda_q2 * self.lambda_k2).sum(-1).float())
+            + self.lambda_init
+        ).view(1, H, 1, 1)
+
+        if attn_mask.dim() == 3:
+            attn_mask = attn_mask.view(B, H, L, L)
+        scale = self.get_buffer("scaling")
+        s1 = q1 @ k1.transpose(-2, -1) * scale + attn_mask
+        s2 = q2 @ k2.transpose(-2, -1) * scale + attn_mask
+        attn = torch.softmax(s1, dim=-1) - lam.to(s1.dtype) * torch.softmax(s2, dim=-1)
+        out = attn @ v  # [B, H, L, Dv]
+
+        # per-head RMSNorm + (1 - lambda_init) scaling
+        rms = torch.sqrt(out.pow(2).mean(dim=-1, keepdim=True) + self.eps)
+        out = (out / rms) * self.rms_scale * (1.0 - self.lambda_init)
+        out = out.transpose(1, 2).contiguous().view(B, L, H * Dv)
+        return self.W_o(out)

@@ -38,34 +38,43 @@ class PointWiseFeedForward(torch.nn.Module):
 
 
 class SwiGLU(torch.nn.Module):
-    """Gated MLP: W2(silu(W1 x) * W3 x) (reference ffn.py:60)."""
+    """Gated MLP: W2(silu(WG x) * W1 x) — parameter names/shapes match
+    reference ffn.py:60 for state-dict compatibility."""
 
     def __init__(self, embedding_dim: int, hidden_dim: int = None, dropout: float = 0.0) -> None:
         super().__init__()
-        hidden = hidden_dim or embedding_dim * 4
-        self.w1 = torch.nn.Linear(embedding_dim, hidden, bias=False)
-        self.w3 = torch.nn.Linear(embedding_dim, hidden, bias=False)
-        self.w2 = torch.nn.Linear(hidden, embedding_dim, bias=False)
+        hidden = hidden_dim or embedding_dim * 2
+        self.WG = torch.nn.Linear(embedding_dim, hidden)
+        self.W1 = torch.nn.Linear(embedding_dim, hidden)
+        self.W2 = torch.nn.Linear(hidden, embedding_dim)
         self.dropout = torch.nn.Dropout(dropout)
 
+    def reset_parameters(self) -> None:
+        for _, param in self.named_parameters():
+            if param.dim() >= 2:
+                torch.nn.init.xavier_normal_(param.data)
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return self.dropout(self.w2(torch.nn.functional.silu(self.w1(x)) * self.w3(x)))
+        return self.dropout(self.W2(torch.nn.functional.silu(self.WG(x)) * self.W1(x)))
 
 
 class SwiGLUEncoder(torch.nn.Module):
-    """SwiGLU + RMSNorm block stack for the Two-Tower item tower
-    (reference ffn.py:102)."""
+    """Two post-norm SwiGLU blocks with skip connections for the Two-Tower
+    item tower (reference ffn.py:102: x = norm(sw(x) + x), twice)."""
 
-    def __init__(self, embedding_dim: int, hidden_dim: int = None, num_blocks: int = 1, dropout: float = 0.0) -> None:
+    def __init__(self, embedding_dim: int, hidden_dim: int = None, num_blocks: int = 2, dropout: float = 0.0) -> None:
         super().__init__()
-        self.blocks = torch.nn.ModuleList(
-            [SwiGLU(embedding_dim, hidden_dim, dropout) for _ in range(num_blocks)]
-        )
-        self.norms = torch.nn.ModuleList(
-            [torch.nn.RMSNorm(embedding_dim) for _ in range(num_blocks)]
-        )
+        self.num_blocks = num_blocks
+        for i in range(1, num_blocks + 1):
+            setattr(self, f"sw{i}", SwiGLU(embedding_dim, hidden_dim, dropout))
+            setattr(self, f"norm{i}", torch.nn.RMSNorm(embedding_dim))
+
+    def reset_parameters(self) -> None:
+        for i in range(1, self.num_blocks + 1):
+            getattr(self, f"sw{i}").reset_parameters()
+            getattr(self, f"norm{i}").reset_parameters()
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        for block, norm in zip(self.blocks, self.norms):
-            x = x + block(norm(x))
+        for i in range(1, self.num_blocks + 1):
+            x = getattr(self, f"norm{i}")(getattr(self, f"sw{i}")(x) + x)
         return x

@@ -1,12 +1,16 @@
 """Differential-transformer encoder blocks for SASRec.
 
 Parity with reference replay/nn/sequential/sasrec/diff_transformer.py
-(DiffTransformerBlock:10, DiffTransformerLayer:67): RMSNorm +
-MultiHeadDifferentialAttention + SwiGLU post-norm blocks.
+(DiffTransformerBlock:10, DiffTransformerLayer:67): each block is
+differential attention (value width 2E) -> add -> RMSNorm, then
+SwiGLU(E, 2E) -> add -> RMSNorm, with a depth-dependent lambda_init of
+``0.8 - 0.6 * exp(-0.3 * block_num)``.  Module/parameter names match the
+reference for state-dict compatibility.
 """
 
 from __future__ import annotations
 
+import math
 from typing import Optional
 
 import torch
@@ -16,26 +20,24 @@ from replay_amd.nn.ffn import SwiGLU
 
 
 class DiffTransformerBlock(torch.nn.Module):
-    def __init__(
-        self,
-        embedding_dim: int,
-        num_heads: int,
-        dropout: float = 0.0,
-        lambda_init: float = 0.8,
-        depth: int = 1,
-        ffn_hidden: Optional[int] = None,
-    ) -> None:
+    def __init__(self, embedding_dim: int, num_heads: int, lambda_init: float = 0.8) -> None:
         super().__init__()
         self.attn_norm = torch.nn.RMSNorm(embedding_dim)
-        self.attention = MultiHeadDifferentialAttention(
-            embedding_dim, num_heads, lambda_init, dropout, depth
+        self.attn = MultiHeadDifferentialAttention(
+            embedding_dim, num_heads, lambda_init, vdim=2 * embedding_dim
         )
-        self.ffn_norm = torch.nn.RMSNorm(embedding_dim)
-        self.ffn = SwiGLU(embedding_dim, ffn_hidden, dropout)
+        self.ff_norm = torch.nn.RMSNorm(embedding_dim)
+        self.ff = SwiGLU(embedding_dim, 2 * embedding_dim)
 
-    def forward(self, x: torch.Tensor, attn_mask=None, key_padding_mask=None) -> torch.Tensor:
-        x = x + self.attention(self.attn_norm(x), attn_mask=attn_mask, key_padding_mask=key_padding_mask)
-        x = x + self.ffn(self.ffn_norm(x))
+    def reset_parameters(self) -> None:
+        self.attn_norm.reset_parameters()
+        self.attn.reset_parameters()
+        self.ff_norm.reset_parameters()
+        self.ff.reset_parameters()
+
+    def forward(self, x: torch.Tensor, attn_mask=None) -> torch.Tensor:
+        x = self.attn_norm(self.attn(x, x, x, attn_mask) + x)
+        x = self.ff_norm(self.ff(x) + x)
         return x
 
 
@@ -48,21 +50,24 @@ class DiffTransformerLayer(torch.nn.Module):
         embedding_dim: int,
         num_heads: int,
         num_blocks: int,
-        dropout: float = 0.0,
-        lambda_init: float = 0.8,
-        ffn_hidden: Optional[int] = None,
     ) -> None:
         super().__init__()
-        self.blocks = torch.nn.ModuleList(
+        self.layers = torch.nn.ModuleList(
             [
-                DiffTransformerBlock(embedding_dim, num_heads, dropout, lambda_init, depth + 1, ffn_hidden)
-                for depth in range(num_blocks)
+                DiffTransformerBlock(
+                    embedding_dim,
+                    num_heads,
+                    lambda_init=0.8 - 0.6 * math.exp(-0.3 * block_num),
+                )
+                for block_num in range(num_blocks)
             ]
         )
 
+    def reset_parameters(self) -> None:
+        for layer in self.layers:
+            layer.reset_parameters()
+
     def forward(self, x: torch.Tensor, attn_mask=None, padding_mask: Optional[torch.Tensor] = None) -> torch.Tensor:
-        for block in self.blocks:
-            x = block(x, attn_mask=attn_mask)
-            if padding_mask is not None:
-                x = x * padding_mask.unsqueeze(-1).to(x.dtype)
+        for layer in self.layers:
+            x = layer(x, attn_mask=attn_mask)
         return x

@@ -12,7 +12,7 @@ from typing import Optional
 
 import torch
 
-from replay_amd.ops import hip_ext, require_hip_on_gpu
+from replay_amd.ops import require_hip_on_gpu
 
 
 def eager_attention(
@@ -40,13 +40,14 @@ def fused_attention(
     attn_mask: Optional[torch.Tensor],
     dropout_p: float = 0.0,
 ) -> torch.Tensor:
-    """q,k,v: [BH, L, Dh]; attn_mask additive [BH, L, L] or None."""
-    if require_hip_on_gpu(q):
-        ext = hip_ext()
-        if hasattr(ext, "attention_fwd") and dropout_p == 0.0 and q.dtype in (torch.bfloat16, torch.float16):
-            from replay_amd.ops.autograd import FlashAttentionFunction
+    """q,k,v: [BH, L, Dh]; attn_mask additive [BH, L, L] or None.
 
-            return FlashAttentionFunction.apply(q, k, v, attn_mask)
-        # HIP extension present but this config unsupported -> eager on GPU is
-        # still torch-ROCm (hipBLASLt GEMMs), allowed for odd dtypes/dropout.
+    The flash-style HIP kernel takes a [B,L] bool padding mask + causal flag,
+    not an additive [BH,L,L] mask; callers with a MaskSpec dispatch to it
+    directly in ``MultiheadAttention.forward``.  Additive-mask callers land
+    here and run the eager math — on GPU that is still torch-ROCm
+    (hipBLASLt GEMMs + fused softmax), the correct path for materialized
+    masks, odd dtypes and dropout.
+    """
+    require_hip_on_gpu(q)  # loud failure if the extension is missing on GPU
     return eager_attention(q, k, v, attn_mask, dropout_p)

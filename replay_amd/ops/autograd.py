@@ -27,7 +27,13 @@ class LayerNormFunction(torch.autograd.Function):
 class FusedCrossEntropyFunction(torch.autograd.Function):
     """Fused softmax-CE over [N, V] logits (K10): forward reads the logits
     twice (max, sum-exp) with no fp32 copy; backward overwrites the logits
-    storage with dlogits in one pass."""
+    storage with dlogits in one pass (halves peak memory at large V).
+
+    The in-place consumption means the logits buffer is CORRUPTED after the
+    first backward: a second backward through this node (retain_graph) is
+    guarded with a RuntimeError, and when the backward itself is recorded
+    (create_graph / double backward) the kernel falls back to an out-of-place
+    pass on a clone so the saved logits stay intact."""
 
     @staticmethod
     def forward(ctx, logits2d: torch.Tensor, labels: torch.Tensor, ignore_index: int = -100):
@@ -35,12 +41,24 @@ class FusedCrossEntropyFunction(torch.autograd.Function):
         loss_sum, count, lse = ext.ce_fwd(logits2d, labels, ignore_index)
         ctx.save_for_backward(logits2d, labels, lse, count)
         ctx.ignore_index = ignore_index
+        ctx.consumed = False
         return (loss_sum / count.clamp(min=1).to(loss_sum.dtype)).squeeze(0)
 
     @staticmethod
     def backward(ctx, dloss):
         ext = hip_ext()
         logits2d, labels, lse, count = ctx.saved_tensors
+        if ctx.consumed:
+            raise RuntimeError(
+                "FusedCrossEntropyFunction.backward consumed the saved logits "
+                "in-place on a previous call; re-run the forward instead of "
+                "backpropagating twice through the same fused-CE node."
+            )
+        in_place = not torch.is_grad_enabled()
+        if in_place:
+            ctx.consumed = True
+        else:
+            logits2d = logits2d.clone()
         dlogits = ext.ce_bwd(
             logits2d, labels, lse, dloss.reshape(1), count, ctx.ignore_index, True
         )

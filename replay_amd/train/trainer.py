@@ -39,6 +39,20 @@ def move_batch(batch, device):
     return batch
 
 
+class _TrainStepShim(torch.nn.Module):
+    """DDP wrapper target whose forward delegates to ``training_step`` so
+    single- and multi-GPU runs execute identical module code (Lightning does
+    the same): DDP hooks fire on the shim's forward while token masking /
+    negative sampling / loss construction stay inside the module."""
+
+    def __init__(self, module: torch.nn.Module) -> None:
+        super().__init__()
+        self.module = module
+
+    def forward(self, batch, batch_idx):
+        return self.module.training_step(batch, batch_idx)
+
+
 class Trainer:
     def __init__(
         self,
@@ -114,6 +128,7 @@ class Trainer:
             "epoch": self.current_epoch,
             "global_step": self.global_step,
             "optimizer_states": [self._optimizer.state_dict()] if getattr(self, "_optimizer", None) else [],
+            "lr_schedulers": [self._scheduler.state_dict()] if getattr(self, "_scheduler", None) else [],
             "hyper_parameters": getattr(module, "hparams", {}),
             "pytorch-lightning_version": "replay_amd-compat",
         }
@@ -128,7 +143,7 @@ class Trainer:
         model = module
         if _dist_ok() and self.world_size > 1 and any(p.requires_grad for p in module.parameters()):
             model = torch.nn.parallel.DistributedDataParallel(
-                module,
+                _TrainStepShim(module),
                 device_ids=[self.device.index] if self.device.type == "cuda" else None,
                 # xGMI is point-to-point (7 links/GPU): few large buckets beat
                 # many small ones for ring all-reduce on small models
@@ -178,6 +193,8 @@ class Trainer:
             module.load_state_dict(ckpt["state_dict"])
             if ckpt.get("optimizer_states"):
                 self._optimizer.load_state_dict(ckpt["optimizer_states"][0])
+            if ckpt.get("lr_schedulers") and self._scheduler is not None:
+                self._scheduler.load_state_dict(ckpt["lr_schedulers"][0])
             self.current_epoch = ckpt.get("epoch", 0)
             self.global_step = ckpt.get("global_step", 0)
 
@@ -192,7 +209,7 @@ class Trainer:
                 batch = self._apply_transforms(batch, "train")
                 with self._autocast():
                     loss = (
-                        model(batch)
+                        model(batch, batch_idx)
                         if isinstance(model, torch.nn.parallel.DistributedDataParallel)
                         else module.training_step(batch, batch_idx)
                     )

@@ -104,6 +104,7 @@ def _twotower_loader(sequential_data):
 
 class TestBert4RecE2E:
     def test_training_loss_improves(self, sequential_data, tensor_schema):
+        torch.manual_seed(0)  # unseeded init made this threshold test flaky
         model = Bert4Rec.from_params(
             tensor_schema, max_sequence_length=MAX_LEN, embedding_dim=16, num_blocks=1, num_heads=2
         )
