@@ -237,7 +237,7 @@ def twotower_bench(args, device, rank, world) -> None:
         dropout=0.0,
         loss=LogInCE(),
     ).to(device)
-    if world > 1:
+    if torch.distributed.is_initialized():
         model = torch.nn.parallel.DistributedDataParallel(
             model,
             device_ids=[device.index] if use_cuda else None,
@@ -433,7 +433,10 @@ def main() -> None:
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     use_cuda = torch.cuda.is_available()
-    if world > 1:
+    # under torchrun (WORLD_SIZE set) always init the process group, even at
+    # world size 1, so RCCL init + the DDP path are exercised at every N and
+    # the scaling curve's N=1 point runs the same code as N=8
+    if "WORLD_SIZE" in os.environ:
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         torch.distributed.init_process_group(backend="nccl" if use_cuda else "gloo")
     device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
@@ -457,7 +460,7 @@ def main() -> None:
 
     apply_model_config(args)
     model = build_model(device, args.model)
-    if world > 1:
+    if torch.distributed.is_initialized():
         model = torch.nn.parallel.DistributedDataParallel(
             model,
             device_ids=[device.index] if use_cuda else None,
@@ -559,7 +562,7 @@ def main() -> None:
                 }
             )
         )
-    if world > 1:
+    if torch.distributed.is_initialized():
         torch.distributed.destroy_process_group()
 
 
