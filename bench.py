@@ -19,8 +19,8 @@ import os
 import sys
 import time
 
-_serve_mode = "serve" in sys.argv
-if "--no-tunableop" not in sys.argv and not _serve_mode:
+_no_tune_mode = "serve" in sys.argv or "twotower" in sys.argv
+if "--no-tunableop" not in sys.argv and not _no_tune_mode:
     # rocBLAS/hipBLASLt algorithm tuning (split-K for the skinny wgrad GEMM
     # shapes; +10% step time measured); must be set before torch import.
     # Results persist in ONE repo-tracked canonical CSV; TunableOp reads/writes
@@ -61,6 +61,8 @@ def apply_model_config(args):
     global SEQ_LEN, EMB_DIM, N_BLOCKS, N_HEADS
     if args.model == "bert4rec":
         SEQ_LEN, EMB_DIM, N_BLOCKS, N_HEADS = 200, 128, 4, 4
+        if args.batch == 8192:  # keep tokens/step comparable (B*L = 409600)
+            args.batch = 2048
 
 
 def build_model(device, model_name="sasrec"):

@@ -499,7 +499,6 @@ class TestMfmaAttentionBwd:
 
 @requires_gpu
 class TestFusedLinearCEDhFusion:
-    @pytest.mark.xfail(strict=False, reason="round-2 experimental path (off by default); must not block the suite")
     def test_fuse_dh_env_path_matches_default(self):
         """REPLAY_AMD_CE_FUSE_DH=1 (the round-2 in-kernel dhidden fusion)
         must be numerically equivalent to the shipped path.  Run in a
