@@ -19,16 +19,15 @@ import os
 import sys
 import time
 
-_no_tune_mode = "serve" in sys.argv or "twotower" in sys.argv
-if "--no-tunableop" not in sys.argv and not _no_tune_mode:
-    # rocBLAS/hipBLASLt algorithm tuning (split-K for the skinny wgrad GEMM
-    # shapes; +10% step time measured); must be set before torch import.
-    # Results persist in ONE repo-tracked canonical CSV; TunableOp reads/writes
+if "--no-tunableop" not in sys.argv:
+    # rocBLAS/hipBLASLt algorithm selection from the repo-tracked tuned CSV
+    # (split-K for the skinny wgrad GEMM shapes; +10% step time measured).
+    # READ-ONLY by default: tuning probes on unusual shapes (the twotower
+    # strided-batched loss GEMMs, bert4rec wgrad) memory-fault the GPU, so
+    # probing is opt-in via --tune (run it on a throwaway box, then merge the
+    # per-ordinal CSV back into tunableop_gfx950.csv).  TunableOp reads
     # per-device-ordinal files (<base><ordinal>.csv), so seed those from the
-    # canonical here — fresh boxes (same gfx950 GPU) then reuse the tuned
-    # algorithms instead of re-running the ~90 s sweep; only NEW shapes tune.
-    # Disabled in serve mode (bad-tile probes on [B, 2M]-wide GEMMs cost
-    # seconds each).
+    # canonical.  Must be set before torch import.
     _tune_file = os.path.join(os.path.dirname(os.path.abspath(__file__)), "tunableop_gfx950.csv")
     if os.path.exists(_tune_file):
         import shutil as _shutil
@@ -38,7 +37,7 @@ if "--no-tunableop" not in sys.argv and not _no_tune_mode:
             if not os.path.exists(_per_dev):
                 _shutil.copy(_tune_file, _per_dev)
     os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
-    os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "1")
+    os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "1" if "--tune" in sys.argv else "0")
     os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME", _tune_file)
     os.environ.setdefault("PYTORCH_TUNABLEOP_MAX_TUNING_DURATION_MS", "30")
     os.environ.setdefault("PYTORCH_TUNABLEOP_MAX_TUNING_ITERATIONS", "3")
@@ -397,8 +396,10 @@ def main() -> None:
     parser.add_argument("--warmup", type=int, default=5)
     parser.add_argument("--batch", type=int, default=8192, help="per-GPU batch size")
     parser.add_argument("--lr", type=float, default=1e-3)
-    parser.add_argument("--tunableop", action="store_true", help="(default on)")
+    parser.add_argument("--tunableop", action="store_true", help="(default on, read-only)")
     parser.add_argument("--no-tunableop", action="store_true", help="disable rocBLAS TunableOp")
+    parser.add_argument("--tune", action="store_true",
+                        help="enable TunableOp PROBING (crash-prone on odd shapes; throwaway runs only)")
     parser.add_argument("--mode", choices=["train", "serve", "itemknn", "twotower"], default="train")
     parser.add_argument("--model", choices=["sasrec", "bert4rec"], default="sasrec",
                         help="train mode: sasrec (config 2) or bert4rec (config 3 shape)")
@@ -541,7 +542,11 @@ def main() -> None:
         print(
             json.dumps(
                 {
-                    "metric": "training interactions/sec, SASRec ML-20M-shape",
+                    "metric": (
+                        "training interactions/sec, SASRec ML-20M-shape"
+                        if args.model == "sasrec"
+                        else "training interactions/sec, BERT4Rec ML-20M-shape"
+                    ),
                     "value": value,
                     "unit": "interactions/sec",
                     "n_gpus": n_gpus,
