@@ -213,6 +213,172 @@ __global__ __launch_bounds__(256, 2) void scored_topk_gemm_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// v2 (E = 256 shape, 8-wave): the 4-wave/128-KiB-LDS variant above runs ONE
+// wave per SIMD (LDS-bound occupancy), so every B-load stall parks the only
+// MFMA stream on that SIMD — measured SQ_WAIT 83%, MFMA busy 9%.  This
+// variant keeps the same 256-query M-tile but splits it over 8 waves
+// (32 rows each): the A fragments now fit in 64 VGPRs/lane, so there is NO
+// LDS at all, two waves share each SIMD (interleaved latency hiding), and
+// each wave runs a depth-PIPE_D B ring.  The 8x B-tile read redundancy
+// within the workgroup is served by L1/L2 (4 KB/section tile slice), and
+// the 4 M-tiles' passes over W share the die-level L3 (256 MB), so HBM
+// traffic stays ~one W stream.
+template <int E, int PIPE_D>
+__global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v2(
+    const __hip_bfloat16* __restrict__ q,  // [M, E]
+    const __hip_bfloat16* __restrict__ w,  // [V, E]
+    const float* __restrict__ thresholds,  // [M]
+    float* __restrict__ out_vals,          // [M, cap]
+    int* __restrict__ out_idx,             // [M, cap]
+    int* __restrict__ counts,              // [M]
+    int M, int64_t V, int cap) {
+  constexpr int KSTEPS = E / 32;
+  constexpr int MF = 2;  // row-fragments per wave (wave covers 32 rows)
+  static_assert(KSTEPS % PIPE_D == 0, "ring slot rotation needs KSTEPS % PIPE_D == 0");
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int m0 = blockIdx.x * 256 + wave * 32;
+
+  // ---- resident A fragments: 32 rows x E in 8*KSTEPS VGPR quads ----
+  bf16x8 a_frag[MF][KSTEPS];
+#pragma unroll
+  for (int mf = 0; mf < MF; ++mf) {
+    const int row = m0 + mf * 16 + (lane & 15);
+    const __hip_bfloat16* qr = q + (size_t)min(row, M - 1) * E + (lane >> 4) * 8;
+#pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks) {
+      a_frag[mf][ks] = *reinterpret_cast<const bf16x8*>(qr + ks * 32);
+    }
+    if (row >= M) {
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks) a_frag[mf][ks] = bf16x8{0};
+    }
+  }
+  float t_reg[MF][4];
+#pragma unroll
+  for (int mf = 0; mf < MF; ++mf) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = m0 + mf * 16 + (lane >> 4) * 4 + r;
+      t_reg[mf][r] = (row < M) ? thresholds[row] : INFINITY;
+    }
+  }
+
+  const int64_t n_tiles = (V + 63) >> 6;
+  const int64_t tile0 = blockIdx.y;
+  if (tile0 >= n_tiles) return;
+  // last tile whose 64 items are all in range on the fast (unclamped) path
+  const bool grid_tail = ((n_tiles << 6) != V);
+  const int64_t tile_stride = gridDim.y;
+
+  // per-lane W walk pointers: fragment f of section ks of tile t lives at
+  //   w + (t*64 + f*16 + (lane&15))*E + ks*32 + (lane>>4)*8
+  const int bk0 = (lane >> 4) * 8;
+  const __hip_bfloat16* wp = w + ((size_t)(tile0 << 6) + (lane & 15)) * E + bk0;
+  const size_t wp_stride = (size_t)tile_stride * 64 * E;  // elements per tile step
+
+  bf16x8 b_ring[PIPE_D][4];
+  // load group (tile, ks): 4 fragments, fast path (no clamp)
+  auto load_group_fast = [&](bf16x8 (&dst)[4], const __hip_bfloat16* base, int ks) {
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      dst[f] = *reinterpret_cast<const bf16x8*>(base + (size_t)f * 16 * E + ks * 32);
+    }
+  };
+  auto load_group_clamped = [&](bf16x8 (&dst)[4], int64_t tile, int ks) {
+    const int64_t n0 = tile << 6;
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      const int64_t item = n0 + f * 16 + (lane & 15);
+      const __hip_bfloat16* wr = w + (size_t)min(item, V - 1) * E;
+      dst[f] = *reinterpret_cast<const bf16x8*>(wr + ks * 32 + bk0);
+    }
+  };
+  auto load_group = [&](bf16x8 (&dst)[4], int64_t tile, const __hip_bfloat16* base, int ks) {
+    if (tile >= n_tiles) return;  // ring slot past the walk: leave stale, never consumed
+    if (grid_tail && tile == n_tiles - 1) {
+      load_group_clamped(dst, tile, ks);
+    } else {
+      load_group_fast(dst, base, ks);
+    }
+  };
+
+  // prologue: fill the ring PIPE_D sections ahead (may cross tile bounds)
+#pragma unroll
+  for (int d = 0; d < PIPE_D; ++d) {
+    const int64_t t = tile0 + (int64_t)(d / KSTEPS) * tile_stride;
+    load_group(b_ring[d], t, wp + (size_t)(d / KSTEPS) * wp_stride, d % KSTEPS);
+  }
+
+  for (int64_t tile = tile0; tile < n_tiles; tile += tile_stride) {
+    f32x4 acc[MF][4];
+#pragma unroll
+    for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+      for (int f = 0; f < 4; ++f) acc[mf][f] = f32x4{0.f, 0.f, 0.f, 0.f};
+    const int64_t n0 = tile << 6;
+#pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks) {
+      const int slot = ks % PIPE_D;
+      bf16x8 b_frag[4];
+#pragma unroll
+      for (int f = 0; f < 4; ++f) b_frag[f] = b_ring[slot][f];
+      // refill the slot PIPE_D sections ahead
+      {
+        const int nks = ks + PIPE_D;
+        const int64_t t_ahead = tile + (int64_t)(nks / KSTEPS) * tile_stride;
+        load_group(b_ring[slot], t_ahead, wp + (size_t)(nks / KSTEPS) * wp_stride,
+                   nks % KSTEPS);
+      }
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+#pragma unroll
+        for (int mf = 0; mf < MF; ++mf) {
+          acc[mf][f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mf][ks], b_frag[f], acc[mf][f], 0, 0, 0);
+        }
+      }
+    }
+    wp += wp_stride;
+    // ---- epilogue: threshold test on accumulators (common path: no hit) ----
+    bool any_hit = false;
+#pragma unroll
+    for (int mf = 0; mf < MF; ++mf) {
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        const int64_t item = n0 + f * 16 + (lane & 15);
+        if (item >= V) continue;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          any_hit |= (acc[mf][f][r] >= t_reg[mf][r]);
+        }
+      }
+    }
+    if (__builtin_amdgcn_ballot_w64(any_hit) == 0) continue;  // fast path
+#pragma unroll
+    for (int mf = 0; mf < MF; ++mf) {
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        const int64_t item = n0 + f * 16 + (lane & 15);
+        if (item >= V) continue;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const float v = acc[mf][f][r];
+          if (v >= t_reg[mf][r]) {
+            const int row = m0 + mf * 16 + (lane >> 4) * 4 + r;
+            const int pos = atomicAdd(&counts[row], 1);
+            if (pos < cap) {
+              out_vals[(size_t)row * cap + pos] = v;
+              out_idx[(size_t)row * cap + pos] = (int)item;
+            }
+          }
+        }
+      }
+    }
+  }
+}
+
 }  // namespace
 
 std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
@@ -248,7 +414,13 @@ std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
   } else if (E == 128) {
     LAUNCH_STG(128);
   } else if (E == 256) {
-    LAUNCH_STG(256);
+    // v2: 8-wave, resident-A, LDS-free (see kernel comment)
+    hipLaunchKernelGGL((scored_topk_gemm_kernel_v2<256, 2>), grid, dim3(512), 0, stream,
+                       reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                       reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                       thr.data_ptr<float>(), out_vals.data_ptr<float>(),
+                       out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
+                       (int)capacity);
   } else {
     TORCH_CHECK(false, "scored_topk_gemm supports E in {64, 128, 256}");
   }
