@@ -26,6 +26,7 @@
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 #include "common.h"
+#include <cstdlib>
 
 namespace {
 
@@ -379,6 +380,366 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v2(
   }
 }
 
+// ---------------------------------------------------------------------------
+// v3 (E = 256): v2 keeps A resident but has every wave re-load the shared
+// 64-item B tile (8x redundancy -> L1/L2 bound, measured 22 ms).  v3 stages
+// the B tile ONCE per workgroup into LDS with cooperative
+// global_load_lds (LDS-DMA, no staging VGPRs), double-buffered so the DMA
+// for tile t+1 overlaps the MFMA work on tile t (guide §5 canonical GEMM).
+//
+// LDS image: 64 rows x 544 B (W row 512 B + 32 B junk pad).  The pad makes
+// the row stride 136 dwords == 8 mod 64, so the 16 lanes of a
+// ds_read_b128 group land on 16 DISTINCT banks (conflict-free); the junk
+// chunks are filled from a safe dummy address (glds is lane-linear: the
+// image must be written in exactly lane order, so the pad must be part of
+// the stream, guide §5 glds caveat).
+template <int E>
+__global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v3(
+    const __hip_bfloat16* __restrict__ q,  // [M, E]
+    const __hip_bfloat16* __restrict__ w,  // [V, E]
+    const float* __restrict__ thresholds,  // [M]
+    float* __restrict__ out_vals,          // [M, cap]
+    int* __restrict__ out_idx,             // [M, cap]
+    int* __restrict__ counts,              // [M]
+    int M, int64_t V, int cap) {
+  constexpr int KSTEPS = E / 32;
+  constexpr int MF = 2;                    // row-fragments per wave (32 rows)
+  constexpr int ROW_B = E * 2 + 32;        // padded LDS row bytes (544 at E=256)
+  constexpr int CHUNKS_ROW = ROW_B / 16;   // 34
+  constexpr int TILE_CHUNKS = 64 * CHUNKS_ROW;  // 2176 chunks of 16 B
+  constexpr int TILE_B = 64 * ROW_B;       // 34816 B per buffer
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int m0 = blockIdx.x * 256 + wave * 32;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];  // 2 x TILE_B
+
+  // ---- resident A fragments: 32 rows x E ----
+  bf16x8 a_frag[MF][KSTEPS];
+#pragma unroll
+  for (int mf = 0; mf < MF; ++mf) {
+    const int row = m0 + mf * 16 + (lane & 15);
+    const __hip_bfloat16* qr = q + (size_t)min(row, M - 1) * E + (lane >> 4) * 8;
+#pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks) {
+      a_frag[mf][ks] = *reinterpret_cast<const bf16x8*>(qr + ks * 32);
+    }
+    if (row >= M) {
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks) a_frag[mf][ks] = bf16x8{0};
+    }
+  }
+  float t_reg[MF][4];
+#pragma unroll
+  for (int mf = 0; mf < MF; ++mf) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = m0 + mf * 16 + (lane >> 4) * 4 + r;
+      t_reg[mf][r] = (row < M) ? thresholds[row] : INFINITY;
+    }
+  }
+
+  const int64_t n_tiles = (V + 63) >> 6;
+  const int64_t tile0 = blockIdx.y;
+  const int64_t tile_stride = gridDim.y;
+  if (tile0 >= n_tiles) return;
+
+  // glds stage of one tile into buffer `buf`: this wave issues wave-count-
+  // strided 1 KiB pieces; lane's source chunk c = piece*64 + lane.
+  // item = c / 34, sub = c % 34; sub >= 32 is the pad (load chunk 0 again —
+  // glds always writes its lane slot, so feed it a safe in-range address).
+  auto stage_tile = [&](int buf, int64_t tile) {
+    const __hip_bfloat16* wt = w + (size_t)(tile << 6) * E;
+    const bool tail = ((tile << 6) + 64) > V;
+    char* lds_base = smem + (size_t)buf * TILE_B;
+#pragma unroll
+    for (int piece = 0; piece < TILE_CHUNKS / 64 / 8; ++piece) {
+      const int c = (piece * 8 + wave) * 64 + lane;
+      int item = c / CHUNKS_ROW;
+      int sub = c % CHUNKS_ROW;
+      if (sub >= E * 2 / 16) sub = 0;  // pad chunk: safe dummy source
+      if (tail) {
+        const int64_t gitem = (tile << 6) + item;
+        item -= (int)(gitem >= V ? (gitem - (V - 1)) : 0);  // clamp into range
+      }
+      const char* src = reinterpret_cast<const char*>(wt) + (size_t)item * (E * 2) + sub * 16;
+      __builtin_amdgcn_global_load_lds(
+          (const void*)src, (void*)(lds_base + (size_t)(piece * 8 + wave) * 1024), 16, 0, 0);
+    }
+  };
+  // TILE_CHUNKS/64 = 34 wave-pieces per tile; 8 waves round-robin -> waves
+  // 0,1 carry 5 pieces, the rest 4 (34 = 4*8 + 2)
+  auto stage_rem = [&](int buf, int64_t tile) {
+    const int base_piece = (TILE_CHUNKS / 64 / 8) * 8;  // 32
+    if (wave < (TILE_CHUNKS / 64) - base_piece) {
+      const __hip_bfloat16* wt = w + (size_t)(tile << 6) * E;
+      const bool tail = ((tile << 6) + 64) > V;
+      char* lds_base = smem + (size_t)buf * TILE_B;
+      const int c = (base_piece + wave) * 64 + lane;
+      int item = c / CHUNKS_ROW;
+      int sub = c % CHUNKS_ROW;
+      if (sub >= E * 2 / 16) sub = 0;
+      if (tail) {
+        const int64_t gitem = (tile << 6) + item;
+        item -= (int)(gitem >= V ? (gitem - (V - 1)) : 0);
+      }
+      const char* src = reinterpret_cast<const char*>(wt) + (size_t)item * (E * 2) + sub * 16;
+      __builtin_amdgcn_global_load_lds(
+          (const void*)src, (void*)(lds_base + (size_t)(base_piece + wave) * 1024), 16, 0, 0);
+    }
+  };
+
+  stage_tile(0, tile0);
+  stage_rem(0, tile0);
+  __syncthreads();  // drains the glds (vmcnt(0) folded into the barrier)
+
+  int cur = 0;
+  for (int64_t tile = tile0; tile < n_tiles; tile += tile_stride) {
+    const int64_t nxt = tile + tile_stride;
+    if (nxt < n_tiles) {
+      stage_tile(cur ^ 1, nxt);
+      stage_rem(cur ^ 1, nxt);
+    }
+    const char* bbuf = smem + (size_t)cur * TILE_B;
+    f32x4 acc[MF][4];
+#pragma unroll
+    for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+      for (int f = 0; f < 4; ++f) acc[mf][f] = f32x4{0.f, 0.f, 0.f, 0.f};
+    const int64_t n0 = tile << 6;
+#pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks) {
+      bf16x8 b_frag[4];
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        const int item = f * 16 + (lane & 15);
+        b_frag[f] = *reinterpret_cast<const bf16x8*>(
+            bbuf + (size_t)item * ROW_B + ks * 64 + (lane >> 4) * 16);
+      }
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+#pragma unroll
+        for (int mf = 0; mf < MF; ++mf) {
+          acc[mf][f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mf][ks], b_frag[f], acc[mf][f], 0, 0, 0);
+        }
+      }
+    }
+    // ---- epilogue: threshold test (common path: no hit) ----
+    bool any_hit = false;
+#pragma unroll
+    for (int mf = 0; mf < MF; ++mf) {
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        const int64_t item = n0 + f * 16 + (lane & 15);
+        if (item >= V) continue;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          any_hit |= (acc[mf][f][r] >= t_reg[mf][r]);
+        }
+      }
+    }
+    if (__builtin_amdgcn_ballot_w64(any_hit) != 0) {
+#pragma unroll
+      for (int mf = 0; mf < MF; ++mf) {
+#pragma unroll
+        for (int f = 0; f < 4; ++f) {
+          const int64_t item = n0 + f * 16 + (lane & 15);
+          if (item >= V) continue;
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const float v = acc[mf][f][r];
+            if (v >= t_reg[mf][r]) {
+              const int row = m0 + mf * 16 + (lane >> 4) * 4 + r;
+              const int pos = atomicAdd(&counts[row], 1);
+              if (pos < cap) {
+                out_vals[(size_t)row * cap + pos] = v;
+                out_idx[(size_t)row * cap + pos] = (int)item;
+              }
+            }
+          }
+        }
+      }
+    }
+    __syncthreads();  // buffer swap guard (also drains next tile's glds)
+    cur ^= 1;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// v4 (E = 256): v3 with the barrier drain removed.  v3's __syncthreads
+// folds a vmcnt(0) whenever a glds is in flight, so every tile waits for
+// the NEXT tile's full 34 KB DMA before any wave proceeds (guide §5: the
+// structural ~20% stall of the 2-buffer + __syncthreads form).  v4 keeps
+// THREE LDS buffers with 2 tiles of DMA in flight, raw s_barrier, and a
+// per-wave counted s_waitcnt vmcnt(N) right before the fragment reads —
+// the canonical glds pipeline (guide: +83% over serial vs +40%).
+//
+// Protocol per iteration i (cur = i % 3):
+//   counted vmcnt(own pieces of 1 newer tile)  -> own glds for buf[cur] done
+//   raw s_barrier                              -> everyone's glds for buf[cur]
+//                                                 done AND everyone finished
+//                                                 reading buf[(i+2)%3] in i-1
+//   stage tile i+2 into buf[(i+2)%3]
+//   compute buf[cur] + epilogue (hit path ends with a full drain so the
+//                                counted wait stays calibrated)
+template <int E>
+__global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
+    const __hip_bfloat16* __restrict__ q,  // [M, E]
+    const __hip_bfloat16* __restrict__ w,  // [V, E]
+    const float* __restrict__ thresholds,  // [M]
+    float* __restrict__ out_vals,          // [M, cap]
+    int* __restrict__ out_idx,             // [M, cap]
+    int* __restrict__ counts,              // [M]
+    int M, int64_t V, int cap) {
+  constexpr int KSTEPS = E / 32;
+  constexpr int MF = 2;
+  constexpr int ROW_B = E * 2 + 32;             // 544
+  constexpr int CHUNKS_ROW = ROW_B / 16;        // 34
+  constexpr int TILE_PIECES = 64 * CHUNKS_ROW / 64;  // 34 wave-pieces
+  constexpr int TILE_B = 64 * ROW_B;
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int m0 = blockIdx.x * 256 + wave * 32;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];  // 3 x TILE_B
+
+  bf16x8 a_frag[MF][KSTEPS];
+#pragma unroll
+  for (int mf = 0; mf < MF; ++mf) {
+    const int row = m0 + mf * 16 + (lane & 15);
+    const __hip_bfloat16* qr = q + (size_t)min(row, M - 1) * E + (lane >> 4) * 8;
+#pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks) {
+      a_frag[mf][ks] = *reinterpret_cast<const bf16x8*>(qr + ks * 32);
+    }
+    if (row >= M) {
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks) a_frag[mf][ks] = bf16x8{0};
+    }
+  }
+  float t_reg[MF][4];
+#pragma unroll
+  for (int mf = 0; mf < MF; ++mf) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = m0 + mf * 16 + (lane >> 4) * 4 + r;
+      t_reg[mf][r] = (row < M) ? thresholds[row] : INFINITY;
+    }
+  }
+  // drain the A/threshold loads so they never mix into the glds counting
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+
+  const int64_t n_tiles = (V + 63) >> 6;
+  const int64_t tile0 = blockIdx.y;
+  const int64_t tile_stride = gridDim.y;
+  if (tile0 >= n_tiles) return;
+
+  auto stage_tile = [&](int buf, int64_t tile) {
+    const __hip_bfloat16* wt = w + (size_t)(tile << 6) * E;
+    const bool tail = ((tile << 6) + 64) > V;
+    char* lds_base = smem + (size_t)buf * TILE_B;
+    for (int piece = wave; piece < TILE_PIECES; piece += 8) {
+      const int c = piece * 64 + lane;
+      int item = c / CHUNKS_ROW;
+      int sub = c % CHUNKS_ROW;
+      if (sub >= E * 2 / 16) sub = 0;
+      if (tail) {
+        const int64_t gitem = (tile << 6) + item;
+        item -= (int)(gitem >= V ? (gitem - (V - 1)) : 0);
+      }
+      const char* src = reinterpret_cast<const char*>(wt) + (size_t)item * (E * 2) + sub * 16;
+      __builtin_amdgcn_global_load_lds(
+          (const void*)src, (void*)(lds_base + (size_t)piece * 1024), 16, 0, 0);
+    }
+  };
+
+  stage_tile(0, tile0);
+  if (tile0 + tile_stride < n_tiles) stage_tile(1, tile0 + tile_stride);
+
+  int cur = 0;
+  for (int64_t tile = tile0; tile < n_tiles; tile += tile_stride) {
+    // own glds for buf[cur] complete; allow 1 newer tile in flight (when one
+    // was actually staged — at the walk's tail there is none, so full drain)
+    if (tile + tile_stride < n_tiles) {
+      if (wave < (TILE_PIECES & 7)) {
+        asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
+      } else {
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      }
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+    const int64_t t2 = tile + 2 * tile_stride;
+    if (t2 < n_tiles) stage_tile((cur + 2) % 3, t2);
+
+    const char* bbuf = smem + (size_t)cur * TILE_B;
+    f32x4 acc[MF][4];
+#pragma unroll
+    for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+      for (int f = 0; f < 4; ++f) acc[mf][f] = f32x4{0.f, 0.f, 0.f, 0.f};
+    const int64_t n0 = tile << 6;
+#pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks) {
+      bf16x8 b_frag[4];
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        const int item = f * 16 + (lane & 15);
+        b_frag[f] = *reinterpret_cast<const bf16x8*>(
+            bbuf + (size_t)item * ROW_B + ks * 64 + (lane >> 4) * 16);
+      }
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+#pragma unroll
+        for (int mf = 0; mf < MF; ++mf) {
+          acc[mf][f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mf][ks], b_frag[f], acc[mf][f], 0, 0, 0);
+        }
+      }
+    }
+    bool any_hit = false;
+#pragma unroll
+    for (int mf = 0; mf < MF; ++mf) {
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        const int64_t item = n0 + f * 16 + (lane & 15);
+        if (item >= V) continue;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          any_hit |= (acc[mf][f][r] >= t_reg[mf][r]);
+        }
+      }
+    }
+    if (__builtin_amdgcn_ballot_w64(any_hit) != 0) {
+#pragma unroll
+      for (int mf = 0; mf < MF; ++mf) {
+#pragma unroll
+        for (int f = 0; f < 4; ++f) {
+          const int64_t item = n0 + f * 16 + (lane & 15);
+          if (item >= V) continue;
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const float v = acc[mf][f][r];
+            if (v >= t_reg[mf][r]) {
+              const int row = m0 + mf * 16 + (lane >> 4) * 4 + r;
+              const int pos = atomicAdd(&counts[row], 1);
+              if (pos < cap) {
+                out_vals[(size_t)row * cap + pos] = v;
+                out_idx[(size_t)row * cap + pos] = (int)item;
+              }
+            }
+          }
+        }
+      }
+      // recalibrate the counted pipeline: drain epilogue VMEM + all glds
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    cur = (cur + 1) % 3;
+  }
+}
+
 }  // namespace
 
 std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
@@ -414,13 +775,34 @@ std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
   } else if (E == 128) {
     LAUNCH_STG(128);
   } else if (E == 256) {
-    // v2: 8-wave, resident-A, LDS-free (see kernel comment)
-    hipLaunchKernelGGL((scored_topk_gemm_kernel_v2<256, 2>), grid, dim3(512), 0, stream,
-                       reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
-                       reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
-                       thr.data_ptr<float>(), out_vals.data_ptr<float>(),
-                       out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
-                       (int)capacity);
+    static const char* variant = std::getenv("REPLAY_AMD_STG_VARIANT");
+    if (variant != nullptr && variant[0] == '2') {
+      // v2: 8-wave, resident-A, LDS-free, per-wave B ring (A/B reference)
+      hipLaunchKernelGGL((scored_topk_gemm_kernel_v2<256, 2>), grid, dim3(512), 0, stream,
+                         reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                         reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                         thr.data_ptr<float>(), out_vals.data_ptr<float>(),
+                         out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
+                         (int)capacity);
+    } else if (variant != nullptr && variant[0] == '3') {
+      // v3: glds double-buffer + __syncthreads (A/B reference)
+      const size_t lds_v3 = 2 * 64 * (256 * 2 + 32);
+      hipLaunchKernelGGL((scored_topk_gemm_kernel_v3<256>), grid, dim3(512), lds_v3, stream,
+                         reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                         reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                         thr.data_ptr<float>(), out_vals.data_ptr<float>(),
+                         out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
+                         (int)capacity);
+    } else {
+      // v4: glds triple-buffer + raw barrier + counted vmcnt (default)
+      const size_t lds_v4 = 3 * 64 * (256 * 2 + 32);
+      hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256>), grid, dim3(512), lds_v4, stream,
+                         reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                         reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                         thr.data_ptr<float>(), out_vals.data_ptr<float>(),
+                         out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
+                         (int)capacity);
+    }
   } else {
     TORCH_CHECK(false, "scored_topk_gemm supports E in {64, 128, 256}");
   }
