@@ -577,12 +577,15 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v3(
 //
 // Protocol per iteration i (cur = i % 3):
 //   counted vmcnt(own pieces of 1 newer tile)  -> own glds for buf[cur] done
+//                                                 (epilogue stores are OLDER
+//                                                 than the newest stage, so
+//                                                 the counted wait drains
+//                                                 them for free)
 //   raw s_barrier                              -> everyone's glds for buf[cur]
 //                                                 done AND everyone finished
 //                                                 reading buf[(i+2)%3] in i-1
-//   stage tile i+2 into buf[(i+2)%3]
-//   compute buf[cur] + epilogue (hit path ends with a full drain so the
-//                                counted wait stays calibrated)
+//   compute buf[cur] + epilogue
+//   stage tile i+2 into buf[(i+2)%3] (LAST, so it is the newest VMEM)
 template <int E>
 __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
     const __hip_bfloat16* __restrict__ q,  // [M, E]
@@ -671,8 +674,6 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     }
     __builtin_amdgcn_s_barrier();
-    const int64_t t2 = tile + 2 * tile_stride;
-    if (t2 < n_tiles) stage_tile((cur + 2) % 3, t2);
 
     const char* bbuf = smem + (size_t)cur * TILE_B;
     f32x4 acc[MF][4];
@@ -733,8 +734,13 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
           }
         }
       }
-      // recalibrate the counted pipeline: drain epilogue VMEM + all glds
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    // stage tile+2 LAST so the next iteration's counted vmcnt drains this
+    // iteration's (older) epilogue stores together with tile+1's glds while
+    // leaving only the newest stage in flight
+    {
+      const int64_t t2 = tile + 2 * tile_stride;
+      if (t2 < n_tiles) stage_tile((cur + 2) % 3, t2);
     }
     cur = (cur + 1) % 3;
   }
