@@ -586,7 +586,7 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v3(
 //                                                 reading buf[(i+2)%3] in i-1
 //   compute buf[cur] + epilogue
 //   stage tile i+2 into buf[(i+2)%3] (LAST, so it is the newest VMEM)
-template <int E>
+template <int E, int MF>
 __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
     const __hip_bfloat16* __restrict__ q,  // [M, E]
     const __hip_bfloat16* __restrict__ w,  // [V, E]
@@ -594,16 +594,17 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
     float* __restrict__ out_vals,          // [M, cap]
     int* __restrict__ out_idx,             // [M, cap]
     int* __restrict__ counts,              // [M]
-    int M, int64_t V, int cap) {
+    int M, int64_t V64, int cap) {
   constexpr int KSTEPS = E / 32;
-  constexpr int MF = 2;
   constexpr int ROW_B = E * 2 + 32;             // 544
   constexpr int CHUNKS_ROW = ROW_B / 16;        // 34
   constexpr int TILE_PIECES = 64 * CHUNKS_ROW / 64;  // 34 wave-pieces
   constexpr int TILE_B = 64 * ROW_B;
+  const int V = (int)V64;  // host asserts V < 2^31 (out_idx is int32 anyway);
+                           // 32-bit index math saves ~30 VGPRs at MF=4
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
-  const int m0 = blockIdx.x * 256 + wave * 32;
+  const int m0 = blockIdx.x * (128 * MF) + wave * (16 * MF);
 
   extern __shared__ __attribute__((aligned(16))) char smem[];  // 3 x TILE_B
 
@@ -633,13 +634,13 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
   // drain the A/threshold loads so they never mix into the glds counting
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
 
-  const int64_t n_tiles = (V + 63) >> 6;
-  const int64_t tile0 = blockIdx.y;
-  const int64_t tile_stride = gridDim.y;
+  const int n_tiles = (V + 63) >> 6;
+  const int tile0 = blockIdx.y;
+  const int tile_stride = gridDim.y;
   if (tile0 >= n_tiles) return;
 
-  auto stage_tile = [&](int buf, int64_t tile) {
-    const __hip_bfloat16* wt = w + (size_t)(tile << 6) * E;
+  auto stage_tile = [&](int buf, int tile) {
+    const char* wt = reinterpret_cast<const char*>(w) + (size_t)((unsigned)tile << 6) * (E * 2);
     const bool tail = ((tile << 6) + 64) > V;
     char* lds_base = smem + (size_t)buf * TILE_B;
     for (int piece = wave; piece < TILE_PIECES; piece += 8) {
@@ -648,12 +649,12 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
       int sub = c % CHUNKS_ROW;
       if (sub >= E * 2 / 16) sub = 0;
       if (tail) {
-        const int64_t gitem = (tile << 6) + item;
-        item -= (int)(gitem >= V ? (gitem - (V - 1)) : 0);
+        const int gitem = (tile << 6) + item;
+        item -= (gitem >= V ? (gitem - (V - 1)) : 0);
       }
-      const char* src = reinterpret_cast<const char*>(wt) + (size_t)item * (E * 2) + sub * 16;
+      const unsigned off = (unsigned)item * (E * 2) + sub * 16;
       __builtin_amdgcn_global_load_lds(
-          (const void*)src, (void*)(lds_base + (size_t)piece * 1024), 16, 0, 0);
+          (const void*)(wt + off), (void*)(lds_base + (size_t)piece * 1024), 16, 0, 0);
     }
   };
 
@@ -661,7 +662,7 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
   if (tile0 + tile_stride < n_tiles) stage_tile(1, tile0 + tile_stride);
 
   int cur = 0;
-  for (int64_t tile = tile0; tile < n_tiles; tile += tile_stride) {
+  for (int tile = tile0; tile < n_tiles; tile += tile_stride) {
     // own glds for buf[cur] complete; allow 1 newer tile in flight (when one
     // was actually staged — at the walk's tail there is none, so full drain)
     if (tile + tile_stride < n_tiles) {
@@ -676,59 +677,45 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
     __builtin_amdgcn_s_barrier();
 
     const char* bbuf = smem + (size_t)cur * TILE_B;
-    f32x4 acc[MF][4];
+    const int n0 = tile << 6;
+    // one 16-item column fragment at a time: acc live set = MF quads (not
+    // MF x 4), which is what lets MF = 4 (512-row M-tile) fit in 256 VGPRs
 #pragma unroll
-    for (int mf = 0; mf < MF; ++mf)
+    for (int f = 0; f < 4; ++f) {
+      f32x4 acc[MF];
 #pragma unroll
-      for (int f = 0; f < 4; ++f) acc[mf][f] = f32x4{0.f, 0.f, 0.f, 0.f};
-    const int64_t n0 = tile << 6;
+      for (int mf = 0; mf < MF; ++mf) acc[mf] = f32x4{0.f, 0.f, 0.f, 0.f};
+      const int item = n0 + f * 16 + (lane & 15);
+      const char* bcol = bbuf + (size_t)(f * 16 + (lane & 15)) * ROW_B + (lane >> 4) * 16;
 #pragma unroll
-    for (int ks = 0; ks < KSTEPS; ++ks) {
-      bf16x8 b_frag[4];
-#pragma unroll
-      for (int f = 0; f < 4; ++f) {
-        const int item = f * 16 + (lane & 15);
-        b_frag[f] = *reinterpret_cast<const bf16x8*>(
-            bbuf + (size_t)item * ROW_B + ks * 64 + (lane >> 4) * 16);
-      }
-#pragma unroll
-      for (int f = 0; f < 4; ++f) {
+      for (int ks = 0; ks < KSTEPS; ++ks) {
+        const bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(bcol + ks * 64);
 #pragma unroll
         for (int mf = 0; mf < MF; ++mf) {
-          acc[mf][f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a_frag[mf][ks], b_frag[f], acc[mf][f], 0, 0, 0);
+          acc[mf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mf][ks], b_frag, acc[mf], 0, 0, 0);
         }
       }
-    }
-    bool any_hit = false;
+      bool any_hit = false;
+      if (item < V) {
 #pragma unroll
-    for (int mf = 0; mf < MF; ++mf) {
+        for (int mf = 0; mf < MF; ++mf) {
 #pragma unroll
-      for (int f = 0; f < 4; ++f) {
-        const int64_t item = n0 + f * 16 + (lane & 15);
-        if (item >= V) continue;
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          any_hit |= (acc[mf][f][r] >= t_reg[mf][r]);
+          for (int r = 0; r < 4; ++r) any_hit |= (acc[mf][r] >= t_reg[mf][r]);
         }
       }
-    }
-    if (__builtin_amdgcn_ballot_w64(any_hit) != 0) {
+      if (__builtin_amdgcn_ballot_w64(any_hit) != 0) {
 #pragma unroll
-      for (int mf = 0; mf < MF; ++mf) {
-#pragma unroll
-        for (int f = 0; f < 4; ++f) {
-          const int64_t item = n0 + f * 16 + (lane & 15);
-          if (item >= V) continue;
+        for (int mf = 0; mf < MF; ++mf) {
 #pragma unroll
           for (int r = 0; r < 4; ++r) {
-            const float v = acc[mf][f][r];
-            if (v >= t_reg[mf][r]) {
+            const float v = acc[mf][r];
+            if (item < V && v >= t_reg[mf][r]) {
               const int row = m0 + mf * 16 + (lane >> 4) * 4 + r;
               const int pos = atomicAdd(&counts[row], 1);
               if (pos < cap) {
                 out_vals[(size_t)row * cap + pos] = v;
-                out_idx[(size_t)row * cap + pos] = (int)item;
+                out_idx[(size_t)row * cap + pos] = item;
               }
             }
           }
@@ -739,7 +726,7 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
     // iteration's (older) epilogue stores together with tile+1's glds while
     // leaving only the newest stage in flight
     {
-      const int64_t t2 = tile + 2 * tile_stride;
+      const int t2 = tile + 2 * tile_stride;
       if (t2 < n_tiles) stage_tile((cur + 2) % 3, t2);
     }
     cur = (cur + 1) % 3;
@@ -757,13 +744,18 @@ std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
   const int E = (int)q.size(1);
   const int64_t V = w.size(0);
   TORCH_CHECK(w.size(1) == E, "dim mismatch");
+  TORCH_CHECK(V < (int64_t)INT32_MAX, "catalog must fit int32 indices");
   auto opts_f = q.options().dtype(torch::kFloat32);
   auto opts_i = q.options().dtype(torch::kInt32);
   auto out_vals = torch::full({M, capacity}, -std::numeric_limits<float>::infinity(), opts_f);
   auto out_idx = torch::zeros({M, capacity}, opts_i);
   auto counts = torch::zeros({M}, opts_i);
   auto thr = thresholds.to(torch::kFloat32).contiguous();
-  const int m_tiles = (M + 255) / 256;
+  static const char* variant = std::getenv("REPLAY_AMD_STG_VARIANT");
+  // E=256 v4 default: 512-row M-tile (MF=4) halves item-table passes
+  const int v4_mf = (variant != nullptr && variant[0] == '5') ? 2 : 4;
+  const int m_tile_rows = (E == 256 && (variant == nullptr || variant[0] >= '4')) ? 128 * v4_mf : 256;
+  const int m_tiles = (M + m_tile_rows - 1) / m_tile_rows;
   // fill 256 CUs x ~4 blocks with >> WGs (guide §1); stripes over item tiles
   int stripes = (int)std::min<int64_t>((V + 63) / 64, std::max(1, 4096 / m_tiles));
   dim3 grid(m_tiles, stripes);
@@ -781,7 +773,6 @@ std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
   } else if (E == 128) {
     LAUNCH_STG(128);
   } else if (E == 256) {
-    static const char* variant = std::getenv("REPLAY_AMD_STG_VARIANT");
     if (variant != nullptr && variant[0] == '2') {
       // v2: 8-wave, resident-A, LDS-free, per-wave B ring (A/B reference)
       hipLaunchKernelGGL((scored_topk_gemm_kernel_v2<256, 2>), grid, dim3(512), 0, stream,
@@ -799,10 +790,19 @@ std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
                          thr.data_ptr<float>(), out_vals.data_ptr<float>(),
                          out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
                          (int)capacity);
-    } else {
-      // v4: glds triple-buffer + raw barrier + counted vmcnt (default)
+    } else if (v4_mf == 2) {
+      // v4 at the narrow 256-row M-tile (A/B reference)
       const size_t lds_v4 = 3 * 64 * (256 * 2 + 32);
-      hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256>), grid, dim3(512), lds_v4, stream,
+      hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 2>), grid, dim3(512), lds_v4, stream,
+                         reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                         reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                         thr.data_ptr<float>(), out_vals.data_ptr<float>(),
+                         out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
+                         (int)capacity);
+    } else {
+      // v4: glds triple-buffer + raw barrier + counted vmcnt, 512-row M-tile
+      const size_t lds_v4 = 3 * 64 * (256 * 2 + 32);
+      hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4>), grid, dim3(512), lds_v4, stream,
                          reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
                          reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
                          thr.data_ptr<float>(), out_vals.data_ptr<float>(),
