@@ -586,7 +586,7 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v3(
 //                                                 reading buf[(i+2)%3] in i-1
 //   compute buf[cur] + epilogue
 //   stage tile i+2 into buf[(i+2)%3] (LAST, so it is the newest VMEM)
-template <int E, int MF>
+template <int E, int MF, int NBUF = 3>
 __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
     const __hip_bfloat16* __restrict__ q,  // [M, E]
     const __hip_bfloat16* __restrict__ w,  // [V, E]
@@ -658,18 +658,22 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
     }
   };
 
-  stage_tile(0, tile0);
-  if (tile0 + tile_stride < n_tiles) stage_tile(1, tile0 + tile_stride);
+#pragma unroll
+  for (int d = 0; d < NBUF - 1; ++d) {
+    if (tile0 + d * tile_stride < n_tiles) stage_tile(d, tile0 + d * tile_stride);
+  }
 
   int cur = 0;
   for (int tile = tile0; tile < n_tiles; tile += tile_stride) {
-    // own glds for buf[cur] complete; allow 1 newer tile in flight (when one
-    // was actually staged — at the walk's tail there is none, so full drain)
-    if (tile + tile_stride < n_tiles) {
+    // own glds for buf[cur] complete; allow NBUF-2 newer tiles in flight
+    // (when they were actually staged — at the walk's tail, full drain)
+    if (tile + (int)(NBUF - 2) * tile_stride < n_tiles) {
       if (wave < (TILE_PIECES & 7)) {
-        asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
+        if constexpr (NBUF == 3) asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
+        else asm volatile("s_waitcnt vmcnt(10)" ::: "memory");
       } else {
-        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        if constexpr (NBUF == 3) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        else asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
       }
     } else {
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -726,10 +730,10 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
     // iteration's (older) epilogue stores together with tile+1's glds while
     // leaving only the newest stage in flight
     {
-      const int t2 = tile + 2 * tile_stride;
-      if (t2 < n_tiles) stage_tile((cur + 2) % 3, t2);
+      const int t2 = tile + (NBUF - 1) * tile_stride;
+      if (t2 < n_tiles) stage_tile((cur + NBUF - 1) % NBUF, t2);
     }
-    cur = (cur + 1) % 3;
+    cur = (cur + 1) % NBUF;
   }
 }
 
@@ -794,6 +798,16 @@ std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
       // v4 at the narrow 256-row M-tile (A/B reference)
       const size_t lds_v4 = 3 * 64 * (256 * 2 + 32);
       hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 2>), grid, dim3(512), lds_v4, stream,
+                         reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                         reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                         thr.data_ptr<float>(), out_vals.data_ptr<float>(),
+                         out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
+                         (int)capacity);
+    } else if (variant != nullptr && variant[0] == '6') {
+      // v4 with a 4-deep buffer ring (3 tiles of DMA in flight)
+      const size_t lds_v46 = 4 * 64 * (256 * 2 + 32);
+      hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 4>), grid, dim3(512), lds_v46,
+                         stream,
                          reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
                          reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
                          thr.data_ptr<float>(), out_vals.data_ptr<float>(),
