@@ -586,7 +586,7 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v3(
 //                                                 reading buf[(i+2)%3] in i-1
 //   compute buf[cur] + epilogue
 //   stage tile i+2 into buf[(i+2)%3] (LAST, so it is the newest VMEM)
-template <int E, int MF, int NBUF = 3>
+template <int E, int MF, int NBUF = 3, int ABLATE = 0>  // 1 = no-stage, 2 = no-mfma
 __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
     const __hip_bfloat16* __restrict__ q,  // [M, E]
     const __hip_bfloat16* __restrict__ w,  // [V, E]
@@ -685,7 +685,7 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
     // one 16-item column fragment at a time: acc live set = MF quads (not
     // MF x 4), which is what lets MF = 4 (512-row M-tile) fit in 256 VGPRs
 #pragma unroll
-    for (int f = 0; f < 4; ++f) {
+    for (int f = 0; f < 4 && ABLATE != 2; ++f) {
       f32x4 acc[MF];
 #pragma unroll
       for (int mf = 0; mf < MF; ++mf) acc[mf] = f32x4{0.f, 0.f, 0.f, 0.f};
@@ -729,7 +729,7 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
     // stage tile+2 LAST so the next iteration's counted vmcnt drains this
     // iteration's (older) epilogue stores together with tile+1's glds while
     // leaving only the newest stage in flight
-    {
+    if constexpr (ABLATE != 1) {
       const int t2 = tile + (NBUF - 1) * tile_stride;
       if (t2 < n_tiles) stage_tile((cur + NBUF - 1) % NBUF, t2);
     }
@@ -803,6 +803,25 @@ std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
                          thr.data_ptr<float>(), out_vals.data_ptr<float>(),
                          out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
                          (int)capacity);
+    } else if (variant != nullptr && (variant[0] == '7' || variant[0] == '8')) {
+      // ablation probes: 7 = no staging after the prologue (compute+barrier
+      // only, WRONG RESULTS), 8 = no MFMA/epilogue (DMA+barrier only)
+      const size_t lds_ab = 3 * 64 * (256 * 2 + 32);
+      if (variant[0] == '7') {
+        hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 3, 1>), grid, dim3(512), lds_ab,
+                           stream, reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                           reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                           thr.data_ptr<float>(), out_vals.data_ptr<float>(),
+                           out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
+                           (int)capacity);
+      } else {
+        hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 3, 2>), grid, dim3(512), lds_ab,
+                           stream, reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                           reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                           thr.data_ptr<float>(), out_vals.data_ptr<float>(),
+                           out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
+                           (int)capacity);
+      }
     } else if (variant != nullptr && variant[0] == '6') {
       // v4 with a 4-deep buffer ring (3 tiles of DMA in flight)
       const size_t lds_v46 = 4 * 64 * (256 * 2 + 32);
