@@ -683,21 +683,39 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
     const char* bbuf = smem + (size_t)cur * TILE_B;
     const int n0 = tile << 6;
     // one 16-item column fragment at a time: acc live set = MF quads (not
-    // MF x 4), which is what lets MF = 4 (512-row M-tile) fit in 256 VGPRs
+    // MF x 4), which is what lets MF = 4 (512-row M-tile) fit in 256 VGPRs.
+    // B fragments are prefetched a full column ahead (ping-pong bfr[2][8])
+    // so the 32-MFMA run per column never parks on lgkmcnt — without this
+    // the ds_read -> 4xMFMA chains serialize (measured 5x off the MFMA
+    // floor with DMA fully hidden).
+    bf16x8 bfr[2][KSTEPS];
+    {
+      const char* bcol0 = bbuf + (size_t)(lane & 15) * ROW_B + (lane >> 4) * 16;
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks) {
+        bfr[0][ks] = *reinterpret_cast<const bf16x8*>(bcol0 + ks * 64);
+      }
+    }
 #pragma unroll
     for (int f = 0; f < 4 && ABLATE != 2; ++f) {
+      if (f < 3) {
+        const char* bcoln =
+            bbuf + (size_t)((f + 1) * 16 + (lane & 15)) * ROW_B + (lane >> 4) * 16;
+#pragma unroll
+        for (int ks = 0; ks < KSTEPS; ++ks) {
+          bfr[(f + 1) & 1][ks] = *reinterpret_cast<const bf16x8*>(bcoln + ks * 64);
+        }
+      }
       f32x4 acc[MF];
 #pragma unroll
       for (int mf = 0; mf < MF; ++mf) acc[mf] = f32x4{0.f, 0.f, 0.f, 0.f};
       const int item = n0 + f * 16 + (lane & 15);
-      const char* bcol = bbuf + (size_t)(f * 16 + (lane & 15)) * ROW_B + (lane >> 4) * 16;
 #pragma unroll
       for (int ks = 0; ks < KSTEPS; ++ks) {
-        const bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(bcol + ks * 64);
 #pragma unroll
         for (int mf = 0; mf < MF; ++mf) {
           acc[mf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              a_frag[mf][ks], b_frag, acc[mf], 0, 0, 0);
+              a_frag[mf][ks], bfr[f & 1][ks], acc[mf], 0, 0, 0);
         }
       }
       bool any_hit = false;
