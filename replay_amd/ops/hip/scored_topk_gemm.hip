@@ -586,7 +586,9 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v3(
 //                                                 reading buf[(i+2)%3] in i-1
 //   compute buf[cur] + epilogue
 //   stage tile i+2 into buf[(i+2)%3] (LAST, so it is the newest VMEM)
-template <int E, int MF, int NBUF = 3, int ABLATE = 0>  // 1 = no-stage, 2 = no-mfma
+// ABLATE: 0 full; 1 no-stage; 2 no-mfma; 3 no-epilogue; 4 no-epilogue +
+// no-barrier/vmcnt (pure ds_read+MFMA loop); 5 no-epilogue + no-stage
+template <int E, int MF, int NBUF = 3, int ABLATE = 0>
 __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
     const __hip_bfloat16* __restrict__ q,  // [M, E]
     const __hip_bfloat16* __restrict__ w,  // [V, E]
@@ -667,18 +669,20 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
   for (int tile = tile0; tile < n_tiles; tile += tile_stride) {
     // own glds for buf[cur] complete; allow NBUF-2 newer tiles in flight
     // (when they were actually staged — at the walk's tail, full drain)
-    if (tile + (int)(NBUF - 2) * tile_stride < n_tiles) {
-      if (wave < (TILE_PIECES & 7)) {
-        if constexpr (NBUF == 3) asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
-        else asm volatile("s_waitcnt vmcnt(10)" ::: "memory");
+    if constexpr (ABLATE != 4) {
+      if (tile + (int)(NBUF - 2) * tile_stride < n_tiles) {
+        if (wave < (TILE_PIECES & 7)) {
+          if constexpr (NBUF == 3) asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
+          else asm volatile("s_waitcnt vmcnt(10)" ::: "memory");
+        } else {
+          if constexpr (NBUF == 3) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+          else asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+        }
       } else {
-        if constexpr (NBUF == 3) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-        else asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       }
-    } else {
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
     }
-    __builtin_amdgcn_s_barrier();
 
     const char* bbuf = smem + (size_t)cur * TILE_B;
     const int n0 = tile << 6;
@@ -726,6 +730,11 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
           for (int r = 0; r < 4; ++r) any_hit |= (acc[mf][r] >= t_reg[mf][r]);
         }
       }
+      if constexpr (ABLATE == 3 || ABLATE == 4 || ABLATE == 5) {
+        // keep acc live without the atomic epilogue (never true at runtime)
+        if (any_hit && cap == -1) counts[0] = 1;
+        continue;
+      }
       if (__builtin_amdgcn_ballot_w64(any_hit) != 0) {
 #pragma unroll
         for (int mf = 0; mf < MF; ++mf) {
@@ -747,7 +756,7 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
     // stage tile+2 LAST so the next iteration's counted vmcnt drains this
     // iteration's (older) epilogue stores together with tile+1's glds while
     // leaving only the newest stage in flight
-    if constexpr (ABLATE != 1) {
+    if constexpr (ABLATE != 1 && ABLATE != 5) {
       const int t2 = tile + (NBUF - 1) * tile_stride;
       if (t2 < n_tiles) stage_tile((cur + NBUF - 1) % NBUF, t2);
     }
@@ -821,6 +830,31 @@ std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
                          thr.data_ptr<float>(), out_vals.data_ptr<float>(),
                          out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
                          (int)capacity);
+    } else if (variant != nullptr &&
+               (variant[0] == '9' || variant[0] == 'a' || variant[0] == 'b')) {
+      const size_t lds_ab = 3 * 64 * (256 * 2 + 32);
+      if (variant[0] == '9') {
+        hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 3, 3>), grid, dim3(512), lds_ab,
+                           stream, reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                           reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                           thr.data_ptr<float>(), out_vals.data_ptr<float>(),
+                           out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
+                           (int)capacity);
+      } else if (variant[0] == 'a') {
+        hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 3, 4>), grid, dim3(512), lds_ab,
+                           stream, reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                           reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                           thr.data_ptr<float>(), out_vals.data_ptr<float>(),
+                           out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
+                           (int)capacity);
+      } else {
+        hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 3, 5>), grid, dim3(512), lds_ab,
+                           stream, reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                           reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                           thr.data_ptr<float>(), out_vals.data_ptr<float>(),
+                           out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
+                           (int)capacity);
+      }
     } else if (variant != nullptr && (variant[0] == '7' || variant[0] == '8')) {
       // ablation probes: 7 = no staging after the prologue (compute+barrier
       // only, WRONG RESULTS), 8 = no MFMA/epilogue (DMA+barrier only)
