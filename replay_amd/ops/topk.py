@@ -67,7 +67,7 @@ def fast_row_topk(
         j += math.ceil(seen.shape[1] * q) + 1  # seen scores may pollute the sample
     j = min(j, sample.shape[1])
     thresholds = sample.topk(j, dim=1).values[:, -1]
-    capacity = max(4 * k, int(2.5 * j / q))
+    capacity = max(4 * k, int(5.0 * j / q))
     # seen filtering happens POST-compaction on the ~k-sized candidate list:
     # an in-kernel scan diverges the wave on serial seen-list loads (measured
     # 2.5 -> 7.8 ms per pass)
@@ -125,7 +125,12 @@ def fused_catalog_topk(
         j += math.ceil(seen.shape[1] * qr) + 1
     j = min(j, sample.shape[1])
     thresholds = sample.topk(j, dim=1).values[:, -1]
-    capacity = max(4 * k, int(2.5 * j / qr))
+    # candidate count per row ~ j/qr with std ~ (j/qr)/sqrt(j): at j ~ 6 a
+    # 2.5x capacity sits ~3 sigma out and the overflow fallback (chunked
+    # re-scoring of the full catalog) fired nearly every step (measured
+    # ~1.6 ms/step of fallback GEMMs); 5x is ~10 sigma and costs only
+    # candidate-buffer memory
+    capacity = max(4 * k, int(5.0 * j / qr))
     vals, idx, counts = ext.scored_topk_gemm(query_emb.contiguous(), item_emb.contiguous(), thresholds, capacity)
     if seen is not None:
         gid = idx.long() + item_offset
