@@ -586,6 +586,20 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v3(
 //                                                 reading buf[(i+2)%3] in i-1
 //   compute buf[cur] + epilogue
 //   stage tile i+2 into buf[(i+2)%3] (LAST, so it is the newest VMEM)
+template <int N>
+__device__ __forceinline__ void waitcnt_vm() {
+  if constexpr (N <= 0) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  else if constexpr (N == 1) asm volatile("s_waitcnt vmcnt(1)" ::: "memory");
+  else if constexpr (N == 2) asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+  else if constexpr (N == 3) asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
+  else if constexpr (N == 4) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  else if constexpr (N == 5) asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
+  else if constexpr (N == 6) asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+  else if constexpr (N == 8) asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+  else if constexpr (N == 10) asm volatile("s_waitcnt vmcnt(10)" ::: "memory");
+  else static_assert(N == 0, "add a vmcnt immediate case");
+}
+
 // ABLATE: 0 full; 1 no-stage; 2 no-mfma; 3 no-epilogue; 4 no-epilogue +
 // no-barrier/vmcnt (pure ds_read+MFMA loop); 5 no-epilogue + no-stage
 template <int E, int MF, int NBUF = 3, int ABLATE = 0>
@@ -670,13 +684,13 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
     // own glds for buf[cur] complete; allow NBUF-2 newer tiles in flight
     // (when they were actually staged — at the walk's tail, full drain)
     if constexpr (ABLATE != 4) {
+      constexpr int P_HI = (TILE_PIECES + 7) / 8;
+      constexpr int P_LO = TILE_PIECES / 8;
       if (tile + (int)(NBUF - 2) * tile_stride < n_tiles) {
         if (wave < (TILE_PIECES & 7)) {
-          if constexpr (NBUF == 3) asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
-          else asm volatile("s_waitcnt vmcnt(10)" ::: "memory");
+          waitcnt_vm<P_HI * (NBUF - 2)>();
         } else {
-          if constexpr (NBUF == 3) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-          else asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+          waitcnt_vm<P_LO * (NBUF - 2)>();
         }
       } else {
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -783,9 +797,11 @@ std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
   auto counts = torch::zeros({M}, opts_i);
   auto thr = thresholds.to(torch::kFloat32).contiguous();
   static const char* variant = std::getenv("REPLAY_AMD_STG_VARIANT");
-  // E=256 v4 default: 512-row M-tile (MF=4) halves item-table passes
-  const int v4_mf = (variant != nullptr && variant[0] == '5') ? 2 : 4;
-  const int m_tile_rows = (E == 256 && (variant == nullptr || variant[0] >= '4')) ? 128 * v4_mf : 256;
+  const bool legacy = (variant != nullptr && variant[0] < '4');
+  // v4 defaults: 512-row M-tile at E=256 (MF=4), 1024-row at E<=128 (MF=8);
+  // wider M-tiles divide the number of passes over the streamed item table
+  const int v4_mf = (variant != nullptr && variant[0] == '5') ? 2 : (E == 64 ? 8 : 4);
+  const int m_tile_rows = legacy ? 256 : 128 * v4_mf;
   const int m_tiles = (M + m_tile_rows - 1) / m_tile_rows;
   // fill 256 CUs x ~4 blocks with >> WGs (guide §1); stripes over item tiles
   int stripes = (int)std::min<int64_t>((V + 63) / 64, std::max(1, 4096 / m_tiles));
@@ -799,10 +815,26 @@ std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
                      thr.data_ptr<float>(), out_vals.data_ptr<float>(),                  \
                      out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,              \
                      (int)capacity)
-  if (E == 64) {
+  if (E == 64 && legacy) {
     LAUNCH_STG(64);
-  } else if (E == 128) {
+  } else if (E == 128 && legacy) {
     LAUNCH_STG(128);
+  } else if (E == 64) {
+    const size_t lds64 = 3 * 64 * (64 * 2 + 32);
+    hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<64, 8>), grid, dim3(512), lds64, stream,
+                       reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                       reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                       thr.data_ptr<float>(), out_vals.data_ptr<float>(),
+                       out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
+                       (int)capacity);
+  } else if (E == 128) {
+    const size_t lds128 = 3 * 64 * (128 * 2 + 32);
+    hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<128, 4>), grid, dim3(512), lds128, stream,
+                       reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                       reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                       thr.data_ptr<float>(), out_vals.data_ptr<float>(),
+                       out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
+                       (int)capacity);
   } else if (E == 256) {
     if (variant != nullptr && variant[0] == '2') {
       // v2: 8-wave, resident-A, LDS-free, per-wave B ring (A/B reference)
