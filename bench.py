@@ -64,10 +64,10 @@ def apply_model_config(args):
             args.batch = 2048
 
 
-def build_model(device, model_name="sasrec"):
+def build_model(device, model_name="sasrec", loss_name="ce"):
     from replay_amd.data.nn import TensorFeatureInfo, TensorSchema
     from replay_amd.data.schema import FeatureHint, FeatureType
-    from replay_amd.nn.loss import CE
+    from replay_amd.nn.loss import CE, CESampled
     from replay_amd.nn.sequential.bert4rec import Bert4Rec
     from replay_amd.nn.sequential.sasrec import SasRec
 
@@ -84,6 +84,11 @@ def build_model(device, model_name="sasrec"):
         ]
     )
     cls = Bert4Rec if model_name == "bert4rec" else SasRec
+    loss = (
+        CESampled(log_correction=True, vocab_size=N_ITEMS)
+        if loss_name == "sampledce"
+        else CE()
+    )
     model = cls.from_params(
         schema,
         max_sequence_length=SEQ_LEN,
@@ -91,12 +96,12 @@ def build_model(device, model_name="sasrec"):
         num_blocks=N_BLOCKS,
         num_heads=N_HEADS,
         dropout=0.0,
-        loss=CE(),
+        loss=loss,
     ).to(device)
     return model
 
 
-def make_batches(n_batches, batch_size, device, seed, model_name="sasrec"):
+def make_batches(n_batches, batch_size, device, seed, model_name="sasrec", n_negatives=0):
     """Synthetic ML-20M-shape sequence batches, generated on device."""
     gen = torch.Generator(device="cpu").manual_seed(seed)
     batches = []
@@ -108,6 +113,10 @@ def make_batches(n_batches, batch_size, device, seed, model_name="sasrec"):
             "padding_mask": torch.ones(batch_size, SEQ_LEN, dtype=torch.bool, device=device),
         }
         batch["labels_padding_mask"] = batch["padding_mask"]
+        if n_negatives:
+            batch["negatives"] = torch.randint(
+                0, N_ITEMS, (n_negatives,), generator=gen
+            ).to(device)
         if model_name == "bert4rec":  # masked-token objective (15% + last)
             tm = torch.rand(batch_size, SEQ_LEN, generator=gen) < 0.15
             tm[:, -1] = True
@@ -403,6 +412,10 @@ def main() -> None:
     parser.add_argument("--mode", choices=["train", "serve", "itemknn", "twotower"], default="train")
     parser.add_argument("--model", choices=["sasrec", "bert4rec"], default="sasrec",
                         help="train mode: sasrec (config 2) or bert4rec (config 3 shape)")
+    parser.add_argument("--loss", choices=["ce", "sampledce"], default="ce",
+                        help="train loss: full-softmax CE or shared-pool sampled CE (K9 fused)")
+    parser.add_argument("--negatives", type=int, default=8192,
+                        help="sampledce: shared negative-pool size per step")
     parser.add_argument(
         "--graphs",
         action="store_true",
@@ -462,7 +475,10 @@ def main() -> None:
         return
 
     apply_model_config(args)
-    model = build_model(device, args.model)
+    if args.loss == "sampledce":
+        global N_ITEMS
+        N_ITEMS = args.items  # sampled CE exists for huge catalogs (config 4/5 scale)
+    model = build_model(device, args.model, args.loss)
     if torch.distributed.is_initialized():
         model = torch.nn.parallel.DistributedDataParallel(
             model,
@@ -473,7 +489,10 @@ def main() -> None:
     optimizer = torch.optim.Adam(
         model.parameters(), lr=args.lr, capturable=args.graphs and use_cuda and world == 1
     )
-    batches = make_batches(4, args.batch, device, seed=1000 + rank, model_name=args.model)
+    batches = make_batches(
+        4, args.batch, device, seed=1000 + rank, model_name=args.model,
+        n_negatives=args.negatives if args.loss == "sampledce" else 0,
+    )
 
     amp_dtype = torch.bfloat16
     autocast = torch.autocast(device_type=device.type, dtype=amp_dtype, enabled=use_cuda)
@@ -563,7 +582,7 @@ def main() -> None:
                         "global_batch": global_batch,
                         "seq_len": SEQ_LEN,
                         "n_items": N_ITEMS,
-                        "loss": "full-softmax CE",
+                        "loss": "full-softmax CE" if args.loss == "ce" else f"sampled CE (pool {args.negatives}, log-corrected)",
                         "parallelism": f"dp{n_gpus}",
                     },
                 }

@@ -129,6 +129,28 @@ class CESampled(SampledLossBase):
     ) -> torch.Tensor:
         if negative_labels is None:
             raise ValueError("CESampled requires negative_labels")
+        mask = target_padding_mask if target_padding_mask is not None else padding_mask
+        from replay_amd.ops.sampled_ce import can_fuse_sampled_ce, fused_sampled_ce_parts
+
+        if can_fuse_sampled_ce(embeddings, negative_labels, self.logits_callback):
+            # K9 fused path: pool LSE from the MFMA linear+LSE kernel — the
+            # [B, L, n] logits never materialize
+            pos_logit, lse_neg, n_coll = fused_sampled_ce_parts(
+                embeddings, positive_labels, negative_labels, self.logits_callback
+            )
+            if self.log_correction:
+                n_neg = negative_labels.shape[0]
+                vocab = self.vocab_size or (int(negative_labels.max()) + 1)
+                correction = torch.log(
+                    torch.tensor(float(max(vocab - 1, 1)), device=lse_neg.device)
+                ) - torch.log((n_neg - n_coll).clamp(min=1))
+                lse_neg = lse_neg + correction
+            per_pos = torch.nn.functional.softplus(lse_neg - pos_logit)
+            valid = mask.to(per_pos.dtype)
+            if weights is not None:
+                valid = valid * weights
+            return (per_pos * valid).sum() / valid.sum().clamp(min=1e-12)
+
         pos, neg = self.get_sampled_logits(embeddings, positive_labels, negative_labels)
         if self.log_correction:
             n_neg = neg.shape[-1]
@@ -139,7 +161,6 @@ class CESampled(SampledLossBase):
             )
             neg = neg + correction
         logits = torch.cat([pos, neg], dim=-1).float()  # [B, L, 1+n]
-        mask = target_padding_mask if target_padding_mask is not None else padding_mask
         target = torch.zeros(logits.shape[:-1], dtype=torch.long, device=logits.device)
         per_pos = torch.nn.functional.cross_entropy(
             logits.reshape(-1, logits.shape[-1]),
