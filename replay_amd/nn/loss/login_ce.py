@@ -35,6 +35,22 @@ class LogInCE(SampledLossBase):
             # negative pool (reference login_ce.py LogInCE vs LogInCESampled:
             # the only difference is the negative set)
             negative_labels = positive_labels[padding_mask].reshape(-1)
+        if self.temperature == 1.0:
+            from replay_amd.ops.sampled_ce import can_fuse_sampled_ce, fused_sampled_ce_parts
+
+            if can_fuse_sampled_ce(embeddings, negative_labels, self.logits_callback):
+                # K9 fused path (see ops/sampled_ce.py): positives-in-logits
+                # InfoNCE is softplus(lse_neg - pos) once collisions are
+                # excluded, so the shared-pool LSE kernel covers it
+                pos_logit, lse_neg, _ = fused_sampled_ce_parts(
+                    embeddings, positive_labels, negative_labels, self.logits_callback
+                )
+                per_pos = torch.nn.functional.softplus(lse_neg - pos_logit)
+                mask = target_padding_mask if target_padding_mask is not None else padding_mask
+                valid = mask.to(per_pos.dtype)
+                if weights is not None:
+                    valid = valid * weights
+                return (per_pos * valid).sum() / valid.sum().clamp(min=1e-12)
         pos, neg = self.get_sampled_logits(embeddings, positive_labels, negative_labels)
         logits = torch.cat([pos, neg], dim=-1).float() / self.temperature
         lse = torch.logsumexp(logits, dim=-1)

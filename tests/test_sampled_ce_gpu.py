@@ -95,3 +95,29 @@ class TestFusedSampledCE:
         finally:
             sce.can_fuse_sampled_ce = orig
         assert abs(float(out_f) - float(out_e)) < 0.05 * max(1.0, abs(float(out_e)))
+
+
+@requires_gpu
+def test_logince_fused_parity():
+    """TwoTower's InfoNCE (positives in-logits) through the fused pool-LSE."""
+    from replay_amd.nn.loss import LogInCE
+
+    B, L, E, V, n_neg = 16, 1, 128, 5000, 2048
+    head, emb, h, labels, negs, mask = _setup(B, L, E, V, n_neg, seed=5)
+    loss = LogInCE()
+    loss.set_logits_callback(head)
+    out_f = loss(h, labels, mask, negative_labels=negs)
+    gh_f, gw_f = torch.autograd.grad(out_f, [h, emb.item_emb.weight])
+
+    import replay_amd.ops.sampled_ce as sce
+
+    orig = sce.can_fuse_sampled_ce
+    sce.can_fuse_sampled_ce = lambda *a, **k: False
+    try:
+        out_e = loss(h, labels, mask, negative_labels=negs)
+        gh_e, gw_e = torch.autograd.grad(out_e, [h, emb.item_emb.weight])
+    finally:
+        sce.can_fuse_sampled_ce = orig
+    assert abs(float(out_f) - float(out_e)) < 0.02 * max(1.0, abs(float(out_e)))
+    assert float(torch.nn.functional.cosine_similarity(gh_f.flatten(), gh_e.flatten(), dim=0)) > 0.999
+    assert float(torch.nn.functional.cosine_similarity(gw_f.flatten(), gw_e.flatten(), dim=0)) > 0.999
