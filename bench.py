@@ -64,7 +64,7 @@ def apply_model_config(args):
             args.batch = 2048
 
 
-def build_model(device, model_name="sasrec", loss_name="ce"):
+def build_model(device, model_name="sasrec", loss_name="ce", sparse_embedding=False):
     from replay_amd.data.nn import TensorFeatureInfo, TensorSchema
     from replay_amd.data.schema import FeatureHint, FeatureType
     from replay_amd.nn.loss import CE, CESampled
@@ -89,6 +89,9 @@ def build_model(device, model_name="sasrec", loss_name="ce"):
         if loss_name == "sampledce"
         else CE()
     )
+    kwargs = {}
+    if sparse_embedding and cls is SasRec:
+        kwargs["sparse_embedding"] = True
     model = cls.from_params(
         schema,
         max_sequence_length=SEQ_LEN,
@@ -97,6 +100,7 @@ def build_model(device, model_name="sasrec", loss_name="ce"):
         num_heads=N_HEADS,
         dropout=0.0,
         loss=loss,
+        **kwargs,
     ).to(device)
     return model
 
@@ -124,6 +128,21 @@ def make_batches(n_batches, batch_size, device, seed, model_name="sasrec", n_neg
             batch["token_mask"] = tm.to(device)
         batches.append(batch)
     return batches
+
+
+def sparse_embedding_setup(model, lr):
+    """Split params for hybrid optimization: SparseAdam over sparse embedding
+    tables (K6: O(touched-rows) updates at 10M+ catalogs), Adam for the rest.
+    Returns (optimizers, sparse_params, ddp_ignore_names)."""
+    sparse_mods = [m for m in model.modules() if isinstance(m, torch.nn.Embedding) and m.sparse]
+    sparse_ids = {id(m.weight) for m in sparse_mods}
+    sparse_params = [m.weight for m in sparse_mods]
+    dense_params = [p for p in model.parameters() if id(p) not in sparse_ids]
+    ignore = [n for n, p in model.named_parameters() if id(p) in sparse_ids]
+    opts = [torch.optim.Adam(dense_params, lr=lr)]
+    if sparse_params:
+        opts.append(torch.optim.SparseAdam(sparse_params, lr=lr))
+    return opts, sparse_params, ignore
 
 
 def serve_bench(args, device, rank, world) -> None:
@@ -246,15 +265,20 @@ def twotower_bench(args, device, rank, world) -> None:
         num_heads=2,
         dropout=0.0,
         loss=LogInCE(),
+        sparse_embedding=not args.dense_emb,
     ).to(device)
+    optimizers, sparse_params, ddp_ignore = sparse_embedding_setup(model, args.lr)
     if torch.distributed.is_initialized():
+        if ddp_ignore:
+            torch.nn.parallel.DistributedDataParallel._set_params_and_buffers_to_ignore_for_model(
+                model, ddp_ignore
+            )
         model = torch.nn.parallel.DistributedDataParallel(
             model,
             device_ids=[device.index] if use_cuda else None,
             bucket_cap_mb=64,
             gradient_as_bucket_view=True,
         )
-    optimizer = torch.optim.Adam(model.parameters(), lr=args.lr)
     gen = torch.Generator(device="cpu").manual_seed(1000 + rank)
     batches = []
     for _ in range(4):
@@ -270,11 +294,17 @@ def twotower_bench(args, device, rank, world) -> None:
     autocast = torch.autocast(device_type=device.type, dtype=torch.bfloat16, enabled=use_cuda)
 
     def step(i: int) -> None:
+        from replay_amd.parallel import sync_sparse_grads
+
         with autocast:
             loss = model(batches[i % len(batches)])
-        optimizer.zero_grad(set_to_none=True)
+        for opt in optimizers:
+            opt.zero_grad(set_to_none=True)
         loss.backward()
-        optimizer.step()
+        if world > 1 and sparse_params:
+            sync_sparse_grads(sparse_params)
+        for opt in optimizers:
+            opt.step()
 
     for i in range(args.warmup):
         step(i)
@@ -416,6 +446,8 @@ def main() -> None:
                         help="train loss: full-softmax CE or shared-pool sampled CE (K9 fused)")
     parser.add_argument("--negatives", type=int, default=8192,
                         help="sampledce: shared negative-pool size per step")
+    parser.add_argument("--dense-emb", action="store_true",
+                        help="disable sparse embedding gradients (twotower / sampledce modes)")
     parser.add_argument(
         "--graphs",
         action="store_true",
@@ -478,17 +510,29 @@ def main() -> None:
     if args.loss == "sampledce":
         global N_ITEMS
         N_ITEMS = args.items  # sampled CE exists for huge catalogs (config 4/5 scale)
-    model = build_model(device, args.model, args.loss)
+    use_sparse = args.loss == "sampledce" and not args.dense_emb
+    model = build_model(device, args.model, args.loss, sparse_embedding=use_sparse)
+    optimizers, sparse_params, ddp_ignore = sparse_embedding_setup(model, args.lr)
     if torch.distributed.is_initialized():
+        if ddp_ignore:
+            torch.nn.parallel.DistributedDataParallel._set_params_and_buffers_to_ignore_for_model(
+                model, ddp_ignore
+            )
         model = torch.nn.parallel.DistributedDataParallel(
             model,
             device_ids=[device.index] if use_cuda else None,
             bucket_cap_mb=64,
             gradient_as_bucket_view=True,
         )
-    optimizer = torch.optim.Adam(
-        model.parameters(), lr=args.lr, capturable=args.graphs and use_cuda and world == 1
-    )
+    if not sparse_params:
+        optimizers = [
+            torch.optim.Adam(
+                model.parameters(), lr=args.lr, capturable=args.graphs and use_cuda and world == 1
+            )
+        ]
+    elif args.graphs:
+        raise SystemExit("--graphs is incompatible with sparse embeddings (SparseAdam)")
+    optimizer = optimizers[0]  # graph-capture path uses the single dense Adam
     batches = make_batches(
         4, args.batch, device, seed=1000 + rank, model_name=args.model,
         n_negatives=args.negatives if args.loss == "sampledce" else 0,
@@ -498,12 +542,18 @@ def main() -> None:
     autocast = torch.autocast(device_type=device.type, dtype=amp_dtype, enabled=use_cuda)
 
     def step(i: int) -> None:
+        from replay_amd.parallel import sync_sparse_grads
+
         batch = batches[i % len(batches)]
         with autocast:
             loss = model(batch)
-        optimizer.zero_grad(set_to_none=True)
+        for opt in optimizers:
+            opt.zero_grad(set_to_none=True)
         loss.backward()
-        optimizer.step()
+        if world > 1 and sparse_params:
+            sync_sparse_grads(sparse_params)
+        for opt in optimizers:
+            opt.step()
 
     if args.graphs and use_cuda and world == 1:
         # hipGraph capture of the whole step (HIP graphs instead of a tracing

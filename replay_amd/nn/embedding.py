@@ -25,11 +25,17 @@ class CategoricalEmbedding(torch.nn.Module):
     ``n_extra`` trainable special-token rows (e.g. BERT4Rec's mask token at
     index cardinality+1)."""
 
-    def __init__(self, cardinality: int, embedding_dim: int, n_extra: int = 0) -> None:
+    def __init__(self, cardinality: int, embedding_dim: int, n_extra: int = 0, sparse: bool = False) -> None:
         super().__init__()
         self.cardinality = cardinality
         self.embedding_dim = embedding_dim
-        self.item_emb = torch.nn.Embedding(cardinality + 1 + n_extra, embedding_dim, padding_idx=cardinality)
+        # sparse=True: backward produces a COO gradient over the touched rows
+        # only — at catalog scale (10M+ rows) the dense scatter-add grad +
+        # dense Adam state walk dominate the step (K6 / SURVEY §7 risk);
+        # sparse rows + SparseAdam keep the update O(touched)
+        self.item_emb = torch.nn.Embedding(
+            cardinality + 1 + n_extra, embedding_dim, padding_idx=cardinality, sparse=sparse
+        )
         # reference embedding.py:199: xavier-normal table init (the torch
         # N(0,1) default puts the initial full-softmax CE at ~5x ln(V)
         # through the sqrt(d)-scaled tied head)
@@ -48,12 +54,19 @@ class CategoricalEmbedding(torch.nn.Module):
         """Full table minus the padding row (reference embedding.py:105-118)."""
         return self.item_emb.weight[: self.cardinality]
 
+    def gather(self, item_ids: torch.Tensor) -> torch.Tensor:
+        """Row gather that preserves the sparse-gradient path (a plain
+        ``weight[ids]`` would build a dense [V, E] grad in backward)."""
+        return torch.nn.functional.embedding(
+            item_ids, self.item_emb.weight, sparse=self.item_emb.sparse
+        )
+
 
 class CategoricalListEmbedding(CategoricalEmbedding):
     """Embeds a list feature [B, L, N] and aggregates over N."""
 
-    def __init__(self, cardinality: int, embedding_dim: int, aggregation: str = "mean", n_extra: int = 0) -> None:
-        super().__init__(cardinality, embedding_dim, n_extra)
+    def __init__(self, cardinality: int, embedding_dim: int, aggregation: str = "mean", n_extra: int = 0, sparse: bool = False) -> None:
+        super().__init__(cardinality, embedding_dim, n_extra, sparse)
         if aggregation not in ("sum", "mean", "max"):
             raise ValueError("aggregation must be sum/mean/max")
         self.aggregation = aggregation
@@ -105,6 +118,7 @@ class SequenceEmbedding(torch.nn.Module):
         categorical_list_aggregation: str = "mean",
         excluded_features: Optional[list] = None,
         n_extra_tokens: int = 0,
+        sparse: bool = False,
     ) -> None:
         super().__init__()
         self.schema = schema
@@ -120,10 +134,10 @@ class SequenceEmbedding(torch.nn.Module):
                     raise ValueError(f"No embedding_dim for categorical feature {name}")
                 if feature.is_list:
                     self.embedders[name] = CategoricalListEmbedding(
-                        feature.cardinality, dim, categorical_list_aggregation, n_extra_tokens
+                        feature.cardinality, dim, categorical_list_aggregation, n_extra_tokens, sparse
                     )
                 else:
-                    self.embedders[name] = CategoricalEmbedding(feature.cardinality, dim, n_extra_tokens)
+                    self.embedders[name] = CategoricalEmbedding(feature.cardinality, dim, n_extra_tokens, sparse)
             else:
                 tensor_dim = feature.tensor_dim or 1
                 if dim is not None and dim != tensor_dim:

@@ -26,6 +26,11 @@ class EmbeddingTyingHead(torch.nn.Module):
         self._item_feature_name = item_feature_name
 
     def get_item_weights(self, item_ids: Optional[torch.Tensor] = None) -> torch.Tensor:
+        if item_ids is not None:
+            gatherer = self._item_embedder()
+            if gatherer is not None and hasattr(gatherer, "gather"):
+                # sparse-gradient-preserving row gather (K6)
+                return gatherer.gather(item_ids)
         if self._item_feature_name is not None:
             weights = self._embedder.get_item_weights(self._item_feature_name)
         else:
@@ -33,6 +38,15 @@ class EmbeddingTyingHead(torch.nn.Module):
         if item_ids is not None:
             weights = weights[item_ids]
         return weights
+
+    def _item_embedder(self):
+        if self._item_feature_name is not None and hasattr(self._embedder, "embedders"):
+            if self._item_feature_name in self._embedder.embedders:
+                return self._embedder.embedders[self._item_feature_name]
+            return None
+        if hasattr(self._embedder, "gather"):
+            return self._embedder
+        return None
 
     def forward(
         self,
@@ -44,8 +58,12 @@ class EmbeddingTyingHead(torch.nn.Module):
         [n] -> shared candidate set; pairwise=True with [..., n] ->
         per-position candidate ids scored against their own hidden state."""
         if pairwise:
-            weights = self.get_item_weights(None)  # [V, E]
-            cand_emb = weights[candidates_to_score]  # [..., n, E]
+            gatherer = self._item_embedder()
+            if gatherer is not None and hasattr(gatherer, "gather"):
+                cand_emb = gatherer.gather(candidates_to_score)  # sparse-grad path
+            else:
+                weights = self.get_item_weights(None)  # [V, E]
+                cand_emb = weights[candidates_to_score]  # [..., n, E]
             return torch.einsum("...e,...ne->...n", hidden, cand_emb.to(hidden.dtype))
         weights = self.get_item_weights(candidates_to_score)
         if weights.dim() == 3:  # per-query candidate sets [B, n, E]

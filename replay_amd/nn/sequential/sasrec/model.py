@@ -132,10 +132,13 @@ class SasRec(torch.nn.Module):
         activation: str = "relu",
         loss: Optional[LossBase] = None,
         excluded_features: Optional[list] = None,
+        sparse_embedding: bool = False,
     ) -> "SasRec":
         from replay_amd.nn.loss import CE
 
-        embedder = SequenceEmbedding(schema, embedding_dim, excluded_features=excluded_features)
+        embedder = SequenceEmbedding(
+            schema, embedding_dim, excluded_features=excluded_features, sparse=sparse_embedding
+        )
         aggregator = PositionAwareAggregator(embedding_dim, max_sequence_length, dropout)
         mask = DefaultAttentionMask(num_heads=num_heads, causal=True)
         encoder = SasRecTransformerLayer(embedding_dim, num_heads, num_blocks, dropout, activation)

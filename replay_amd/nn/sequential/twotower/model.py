@@ -59,6 +59,7 @@ class ItemTower(torch.nn.Module):
         hidden_dim: Optional[int] = None,
         num_blocks: int = 1,
         item_feature_name: Optional[str] = None,
+        sparse: bool = False,
     ) -> None:
         super().__init__()
         self.item_schema = item_schema
@@ -71,11 +72,11 @@ class ItemTower(torch.nn.Module):
         for name, feature in item_schema.items():
             if feature.feature_hint is not None and name == self.item_feature_name:
                 dim = feature.embedding_dim or embedding_dim
-                self.embedders[name] = torch.nn.Embedding(feature.cardinality + 1, dim)
+                self.embedders[name] = torch.nn.Embedding(feature.cardinality + 1, dim, sparse=sparse)
                 in_dim += dim
             elif feature.is_cat:
                 dim = feature.embedding_dim or embedding_dim
-                self.embedders[name] = torch.nn.Embedding(feature.cardinality + 1, dim)
+                self.embedders[name] = torch.nn.Embedding(feature.cardinality + 1, dim, sparse=sparse)
                 in_dim += dim
             else:
                 in_dim += feature.tensor_dim or 1
@@ -281,15 +282,19 @@ class TwoTower(torch.nn.Module):
         item_tower_blocks: int = 1,
         loss: Optional[LossBase] = None,
         item_features: Optional[Dict[str, torch.Tensor]] = None,
+        sparse_embedding: bool = False,
     ) -> "TwoTower":
         from replay_amd.nn.loss import LogInCE
 
-        embedder = SequenceEmbedding(query_schema, embedding_dim)
+        embedder = SequenceEmbedding(query_schema, embedding_dim, sparse=sparse_embedding)
         aggregator = PositionAwareAggregator(embedding_dim, max_sequence_length, dropout)
         mask = DefaultAttentionMask(num_heads=num_heads, causal=True)
         encoder = SasRecTransformerLayer(embedding_dim, num_heads, num_blocks, dropout)
         query_tower = QueryTower(SasRecBody(embedder, aggregator, mask, encoder))
-        item_tower = ItemTower(item_schema or query_schema, embedding_dim, num_blocks=item_tower_blocks)
+        item_tower = ItemTower(
+            item_schema or query_schema, embedding_dim, num_blocks=item_tower_blocks,
+            sparse=sparse_embedding,
+        )
         if item_features:
             item_tower.set_item_features(item_features)
         body = TwoTowerBody(query_tower, item_tower)

@@ -1,3 +1,3 @@
-from .collectives import gather_embeddings, gather_ids, world_info
+from .collectives import gather_embeddings, gather_ids, sync_sparse_grads, world_info
 
-__all__ = ["gather_embeddings", "gather_ids", "world_info"]
+__all__ = ["gather_embeddings", "gather_ids", "sync_sparse_grads", "world_info"]

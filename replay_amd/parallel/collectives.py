@@ -61,3 +61,38 @@ class _GatherEmbeddings(torch.autograd.Function):
 
 def gather_embeddings(x: torch.Tensor) -> torch.Tensor:
     return _GatherEmbeddings.apply(x)
+
+
+def sync_sparse_grads(params) -> None:
+    """Average sparse COO gradients across ranks.
+
+    RCCL/NCCL all-reduce requires dense tensors, so sparse embedding grads
+    (K6: touched-rows-only at 10M+ catalogs) are synchronized by an explicit
+    all-gather of (indices, values) padded to the max nnz, then coalesced.
+    Payload is O(touched rows x E) — tiny next to the dense [V, E] all-reduce
+    it replaces.
+    """
+    rank, world = world_info()
+    if world == 1:
+        return
+    for p in params:
+        g = p.grad
+        if g is None or not g.is_sparse:
+            continue
+        g = g.coalesce()
+        idx, val = g.indices(), g.values()
+        n = torch.tensor([idx.shape[1]], device=val.device, dtype=torch.long)
+        counts = [torch.zeros_like(n) for _ in range(world)]
+        dist.all_gather(counts, n)
+        maxn = int(torch.stack(counts).max())
+        pad_idx = torch.zeros(idx.shape[0], maxn, dtype=idx.dtype, device=idx.device)
+        pad_val = torch.zeros((maxn,) + tuple(val.shape[1:]), dtype=val.dtype, device=val.device)
+        pad_idx[:, : idx.shape[1]] = idx
+        pad_val[: val.shape[0]] = val
+        gi = [torch.empty_like(pad_idx) for _ in range(world)]
+        gv = [torch.empty_like(pad_val) for _ in range(world)]
+        dist.all_gather(gi, pad_idx.contiguous())
+        dist.all_gather(gv, pad_val.contiguous())
+        cat_i = torch.cat([t[:, : int(c)] for t, c in zip(gi, counts)], dim=1)
+        cat_v = torch.cat([t[: int(c)] for t, c in zip(gv, counts)], dim=0)
+        p.grad = torch.sparse_coo_tensor(cat_i, cat_v / world, g.shape).coalesce()
