@@ -65,6 +65,15 @@ class MultiheadAttention(torch.nn.Module):
         L, Dh = x.shape[1], self.head_dim
         if Dh > 64 or 256 % Dh != 0 or L > 512:
             return False
+        # the MFMA backward (bf16, Dh 32/64, L<=256) has a much smaller LDS
+        # footprint than the VALU fallback — gate on the kernel that will
+        # actually run (the VALU formula wrongly rejected bert4rec's
+        # L=200/Dh=32 shape: 168 KB vs the MFMA path's 52 KB)
+        runs_bf16 = x.dtype == torch.bfloat16 or torch.is_autocast_enabled()
+        if runs_bf16 and Dh in (32, 64) and L <= 256:
+            lpad = (L + 31) & ~31
+            lds_bwd = 3 * Dh * lpad * 2 + lpad * 4 + 8 * 512 * 2 + L + 64
+            return lds_bwd <= 160 * 1024
         # backward LDS budget check (fp32 staging): D*Lpad + 5*L*D + extras
         lpad = (L + 63) & ~63
         lds_bwd = 4 * (Dh * lpad + 5 * L * Dh + L + 8 * L) + L

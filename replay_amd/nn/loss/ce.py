@@ -46,6 +46,15 @@ class CE(LossBase):
                     # custom autograd functions bypass autocast: cast here so
                     # the GEMMs and CE kernels run in bf16 (MFMA path)
                     hidden2d = hidden2d.to(torch.get_autocast_dtype("cuda"))
+                labels1d = labels.reshape(-1)
+                if float(mask.float().mean()) < 0.5:
+                    # sparse-target objectives (BERT4Rec masks ~15% of
+                    # tokens): compact to the scored rows before the CE —
+                    # a 6x cut of the linear+CE work; index_select backward
+                    # scatters zeros to the dropped rows
+                    keep = (labels1d != -100).nonzero(as_tuple=True)[0]
+                    hidden2d = hidden2d.index_select(0, keep)
+                    labels1d = labels1d.index_select(0, keep)
                 weight = head.get_item_weights()
                 n_elems = hidden2d.shape[0] * weight.shape[0]
                 if n_elems * hidden2d.element_size() > 24 * 2**30:
@@ -53,7 +62,7 @@ class CE(LossBase):
                     # recomputed in backward and never fully materialized
                     from replay_amd.ops.fused_ce import chunked_fused_ce
 
-                    return chunked_fused_ce(hidden2d, weight, labels.reshape(-1), -100)
+                    return chunked_fused_ce(hidden2d, weight, labels1d, -100)
                 weight = weight.to(hidden2d.dtype)
                 if (
                     hidden2d.dtype == torch.bfloat16
@@ -64,14 +73,14 @@ class CE(LossBase):
                     # pass; backward recomputes them and fuses dhidden
                     from replay_amd.ops.autograd import fused_linear_cross_entropy
 
-                    return fused_linear_cross_entropy(hidden2d, weight, labels.reshape(-1), -100)
+                    return fused_linear_cross_entropy(hidden2d, weight, labels1d, -100)
                 # materialized path: ONE bf16 logits buffer, fused one-pass
                 # LSE forward + in-place dlogits backward (measured faster
                 # than chunking at V<=1e5: no recompute, full-width GEMMs)
                 from replay_amd.ops.autograd import fused_cross_entropy
 
                 logits2d = hidden2d @ weight.t()
-                return fused_cross_entropy(logits2d, labels.reshape(-1), -100)
+                return fused_cross_entropy(logits2d, labels1d, -100)
         logits = self.logits_callback(embeddings)  # [B, L, V]
         return torch.nn.functional.cross_entropy(
             logits.reshape(-1, logits.shape[-1]).float(),
