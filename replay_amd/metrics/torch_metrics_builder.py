@@ -4,7 +4,11 @@ Parity with reference replay/metrics/torch_metrics_builder.py:196
 (``TorchMetricsBuilder``): batch accumulation of recall / precision / ndcg /
 map / mrr / novelty (reference :306-336) from a hit matrix (broadcast compare,
 reference :344-349) plus coverage via a catalog histogram (reference
-:95-168).  Runs on any torch device with vectorized eager torch ops.
+:95-168).  On GPU with the replay_amd extension loaded, the whole per-batch
+accumulation runs as ONE HIP kernel (``metrics_reduce``, K13 in SURVEY
+§2.12: per-row hit scan + wave-reduced fp64 sums for every cutoff);
+elsewhere it falls back to the vectorized eager ops below — both paths are
+parity-tested against each other and against the reference's doctests.
 
 Conventions: ``ground_truth`` / ``train`` are padded with -1;
 ``predictions`` hold top-max_k item ids ranked best-first.
@@ -106,6 +110,31 @@ class TorchMetricsBuilder:
     def _add(self, name: str, value: torch.Tensor) -> None:
         self._sums[name] = self._sums.get(name, 0.0) + float(value.sum().item())
 
+    _KERNEL_METRICS = ("hitrate", "recall", "precision", "ndcg", "map", "mrr", "novelty")
+
+    def _try_kernel(self, predictions, ground_truth, train) -> bool:
+        """K13: one-launch HIP reduction of every metric sum on GPU."""
+        if not predictions.is_cuda or self.max_k > 64 or len(self._ks) > 8:
+            return False
+        from replay_amd.ops import hip_ext
+
+        ext = hip_ext()
+        if ext is None or not hasattr(ext, "metrics_reduce"):
+            return False
+        ks = torch.tensor(self._ks, dtype=torch.int32, device=predictions.device)
+        tr = train.long() if train is not None else None
+        sums = ext.metrics_reduce(predictions.long(), ground_truth.long(), tr, ks)
+        sums = sums.cpu()
+        for mi, mname in enumerate(self._KERNEL_METRICS):
+            if mname not in self._metrics:
+                continue
+            if mname == "novelty" and train is None:
+                continue
+            for ci, k in enumerate(self._ks):
+                key = f"{mname}@{k}"
+                self._sums[key] = self._sums.get(key, 0.0) + float(sums[mi, ci])
+        return True
+
     def add_prediction(
         self,
         predictions: torch.Tensor,
@@ -117,6 +146,12 @@ class TorchMetricsBuilder:
         predictions = predictions[:, : self.max_k]
         batch = predictions.shape[0]
         self._n_users += batch
+        if self._try_kernel(predictions, ground_truth, train):
+            if self._coverage is not None:
+                self._coverage.add_prediction(predictions)
+                if train is not None:
+                    self._coverage.add_train(train)
+            return
 
         # hit matrix: hits[b, k] = pred[b, k] in gt[b]  (reference :344-349)
         gt_valid = ground_truth >= 0

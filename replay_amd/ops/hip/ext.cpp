@@ -38,6 +38,8 @@ std::vector<torch::Tensor> ce_linear_fwd(torch::Tensor hidden, torch::Tensor w,
 std::vector<torch::Tensor> ce_linear_bwd(torch::Tensor hidden, torch::Tensor w,
                                          torch::Tensor labels, torch::Tensor lse,
                                          torch::Tensor gscale, double gsign);
+torch::Tensor metrics_reduce(torch::Tensor preds, torch::Tensor gt,
+                             c10::optional<torch::Tensor> train, torch::Tensor ks);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layer_norm_fwd", &layer_norm_fwd, "fused LayerNorm forward (gfx950)");
@@ -55,4 +57,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused linear+CE forward: online LSE in the GEMM epilogue (gfx950)");
   m.def("ce_linear_bwd", &ce_linear_bwd,
         "fused linear+CE backward: recomputed dlogits + fused dhidden (gfx950)");
+  m.def("metrics_reduce", &metrics_reduce,
+        "fused ranking-metric sums: hits + all cutoffs in one launch (gfx950)");
 }
