@@ -33,6 +33,17 @@ def _to_pandas(df) -> pd.DataFrame:
     if isinstance(df, pd.DataFrame):
         return df
     if _is_polars(df):  # pragma: no cover
+        import warnings
+
+        warnings.warn(
+            "polars input is converted to pandas at the Dataset boundary: the "
+            "MI355X build's single tabular execution backend is pandas/Arrow "
+            "(a native polars pipeline is a documented descope — see "
+            "docs/pages/parity.md). All downstream splitters/filters/metrics "
+            "run the pandas path, which is oracle-equality-tested against the "
+            "reference.",
+            stacklevel=3,
+        )
         return df.to_pandas()
     raise TypeError(f"Unsupported dataframe type: {type(df)} (Spark is not supported by the MI355X build)")
 
