@@ -62,7 +62,41 @@ def test_u_lin_ucb(ds):
 def test_neural_ts(ds):
     from replay_amd.experimental.models import NeuralTS
 
-    _check(NeuralTS(epochs=2, embedding_dim=8, hidden_dim=8, device="cpu"), ds)
+    model = NeuralTS(
+        embedding_sizes=[8, 8, 8], hidden_layers=[16, 8], n_epochs=2,
+        cnt_neg_samples=8, cnt_samples_for_predict=4, seed=0,
+    )
+    _check(model, ds)
+
+
+def test_neural_ts_save_load(ds, tmp_path):
+    import numpy as np
+
+    from replay_amd.experimental.models import NeuralTS
+
+    model = NeuralTS(
+        embedding_sizes=[4, 4, 4], hidden_layers=[8], n_epochs=1,
+        cnt_neg_samples=4, cnt_samples_for_predict=64, exploration_coef=0.0, seed=0,
+    )
+    model.fit(ds)
+    recs1 = model.predict(ds, k=3)
+    model.model_save(str(tmp_path / "nts"))
+    model2 = NeuralTS(
+        embedding_sizes=[4, 4, 4], hidden_layers=[8], n_epochs=1,
+        cnt_neg_samples=4, cnt_samples_for_predict=64, exploration_coef=0.0, seed=0,
+    )
+    model2.model_load(str(tmp_path / "nts"))
+    for attr in ("fit_queries", "fit_items", "_query_dim_size", "_item_dim_size",
+                 "query_column", "item_column", "rating_column", "_user_features",
+                 "_item_features", "_device"):
+        setattr(model2, attr, getattr(model, attr, None))
+    recs2 = model2.predict(ds, k=3)
+    # MC-dropout predictions are stochastic; with exploration_coef=0 and many
+    # samples the means agree loosely
+    m1 = recs1.groupby("query_id")["item_id"].apply(set)
+    m2 = recs2.groupby("query_id")["item_id"].apply(set)
+    overlap = np.mean([len(a & b) / max(len(a), 1) for a, b in zip(m1, m2)])
+    assert overlap > 0.3
 
 
 def test_hierarchical(ds):
@@ -143,3 +177,51 @@ def test_obp_estimators():
     probs = learner.predict(ctx)
     assert probs.shape == (n, A, 1)
     np.testing.assert_allclose(probs[:, :, 0].sum(1), 1.0, atol=1e-6)
+
+
+def test_neural_ts_with_features(ds):
+    """Wide&Deep feature path: continuous + categorical user/item columns."""
+    import numpy as np
+    import pandas as pd
+
+    from replay_amd.data import Dataset, FeatureHint, FeatureInfo, FeatureSchema, FeatureType
+    from replay_amd.experimental.models import NeuralTS
+
+    inter = ds.interactions
+    n_u = int(inter["query_id"].max()) + 1
+    n_i = int(inter["item_id"].max()) + 1
+    rng = np.random.default_rng(0)
+    uf = pd.DataFrame({
+        "query_id": np.arange(n_u),
+        "age": rng.normal(size=n_u),
+        "segment": rng.choice(["a", "b"], size=n_u),
+    })
+    itf = pd.DataFrame({
+        "item_id": np.arange(n_i),
+        "price": rng.normal(size=n_i),
+        "genre": rng.choice(["x", "y", "z"], size=n_i),
+    })
+    schema = FeatureSchema([
+        FeatureInfo("query_id", FeatureType.CATEGORICAL, FeatureHint.QUERY_ID),
+        FeatureInfo("item_id", FeatureType.CATEGORICAL, FeatureHint.ITEM_ID),
+        FeatureInfo("rating", FeatureType.NUMERICAL, FeatureHint.RATING),
+        FeatureInfo("timestamp", FeatureType.NUMERICAL, FeatureHint.TIMESTAMP),
+        FeatureInfo("age", FeatureType.NUMERICAL),
+        FeatureInfo("segment", FeatureType.CATEGORICAL),
+        FeatureInfo("price", FeatureType.NUMERICAL),
+        FeatureInfo("genre", FeatureType.CATEGORICAL),
+    ])
+    full = Dataset(
+        feature_schema=schema, interactions=inter, query_features=uf, item_features=itf,
+        check_consistency=False, categorical_encoded=True,
+    )
+    model = NeuralTS(
+        user_cols={"continuous_cols": ["age"], "cat_embed_cols": ["segment"], "wide_cols": []},
+        item_cols={"continuous_cols": ["price"], "cat_embed_cols": [], "wide_cols": ["genre"]},
+        embedding_sizes=[4, 4, 4], hidden_layers=[8], n_epochs=1,
+        cnt_neg_samples=4, cnt_samples_for_predict=2, seed=0,
+    )
+    model.fit(full)
+    recs = model.predict(full, k=3)
+    assert len(recs) > 0
+    assert recs.groupby("query_id").size().max() <= 3
