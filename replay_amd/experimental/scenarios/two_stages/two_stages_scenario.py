@@ -4,9 +4,14 @@ Parity with reference TwoStagesScenario (replay/experimental/scenarios/
 two_stages/two_stages_scenario.py:112,476,573): first-level models produce
 candidates on a train split; a second-level ranker is trained on a held-out
 split with positives = real interactions among candidates, negatives =
-non-interacted candidates; history-based features feed the ranker.  The
-reference's LightAutoML reranker is replaced by sklearn
-GradientBoostingClassifier (LightAutoML is not in the ROCm stack).
+non-interacted candidates (``negatives_type`` "first_level") or random
+pairs ("random", reference :476-533); a fallback model fills queries the
+first level missed (reference fallback_model); per-model embedding features
+(user/item factors + their elementwise product, reference
+get_first_level_model_features:30-110) and history-based features feed the
+ranker.  The reference's LightAutoML reranker is replaced by sklearn
+GradientBoostingClassifier (LightAutoML is not in the ROCm stack); the
+``optimize`` passthrough tunes first-level models like reference :650-705.
 """
 
 from __future__ import annotations
@@ -29,20 +34,60 @@ class TwoStagesScenario:
         num_candidates: int = 100,
         train_splitter: Optional[RatioSplitter] = None,
         use_generated_features: bool = True,
+        fallback_model: Optional[BaseRecommender] = None,
+        use_first_level_models_feat=False,
+        num_negatives: int = 100,
+        negatives_type: str = "first_level",
         seed: int = 0,
     ) -> None:
-        from replay_amd.models import ALSWrap
+        from replay_amd.models import ALSWrap, PopRec
 
         self.first_level_models = first_level_models or [ALSWrap(rank=16, num_iterations=5, seed=seed)]
         self.num_candidates = num_candidates
         self.train_splitter = train_splitter
         self.use_generated_features = use_generated_features
+        self.fallback_model = PopRec() if fallback_model is None else fallback_model
+        if isinstance(use_first_level_models_feat, bool):
+            use_first_level_models_feat = [use_first_level_models_feat] * len(self.first_level_models)
+        if len(use_first_level_models_feat) != len(self.first_level_models):
+            raise ValueError("use_first_level_models_feat must match first_level_models length")
+        self.use_first_level_models_feat = use_first_level_models_feat
+        if negatives_type not in ("random", "first_level"):
+            raise ValueError(f"Invalid negatives_type: {negatives_type}")
+        self.num_negatives = num_negatives
+        self.negatives_type = negatives_type
         self.seed = seed
         self._ranker = None
         self._fp: Optional[HistoryBasedFeaturesProcessor] = None
         self.query_column = "query_id"
         self.item_column = "item_id"
         self.rating_column = "rating"
+
+    def _model_features(self, model, idx: int, pairs: pd.DataFrame) -> pd.DataFrame:
+        """First-level embedding features (reference
+        get_first_level_model_features:30-110): user factors, item factors
+        and their elementwise product per candidate pair; zero vectors when
+        a model has no embedding for an id."""
+        uf = getattr(model, "user_factors", None)
+        itf = getattr(model, "item_factors", None)
+        if uf is None or itf is None:
+            return pairs
+        rank = uf.shape[1]
+        u = pairs[self.query_column].to_numpy()
+        i = pairs[self.item_column].to_numpy()
+        u_ok = (u >= 0) & (u < uf.shape[0])
+        i_ok = (i >= 0) & (i < itf.shape[0])
+        uvec = np.zeros((len(pairs), rank), dtype=np.float32)
+        ivec = np.zeros((len(pairs), rank), dtype=np.float32)
+        uvec[u_ok] = uf[u[u_ok]]
+        ivec[i_ok] = itf[i[i_ok]]
+        mult = uvec * ivec
+        cols = {}
+        for f in range(rank):
+            cols[f"m{idx}_uf{f}"] = uvec[:, f]
+            cols[f"m{idx}_if{f}"] = ivec[:, f]
+            cols[f"m{idx}_fm{f}"] = mult[:, f]
+        return pd.concat([pairs.reset_index(drop=True), pd.DataFrame(cols)], axis=1)
 
     def _candidate_frame(self, dataset, queries=None) -> pd.DataFrame:
         frames = []
@@ -55,7 +100,33 @@ class TwoStagesScenario:
         out = frames[0]
         for f in frames[1:]:
             out = out.merge(f, on=[self.query_column, self.item_column], how="outer")
-        return out.fillna(0.0)
+        # fallback model fills queries the first level could not cover
+        if self.fallback_model is not None and getattr(self.fallback_model, "is_fitted", True):
+            want = (
+                queries[self.query_column].unique()
+                if queries is not None
+                else dataset.interactions[self.query_column].unique()
+            )
+            missing = np.setdiff1d(want, out[self.query_column].unique())
+            if len(missing):
+                try:
+                    fb = self.fallback_model.predict(
+                        dataset,
+                        self.num_candidates,
+                        queries=pd.DataFrame({self.query_column: missing}),
+                        filter_seen_items=True,
+                    )
+                    fb = fb.rename(columns={self.rating_column: "rel_0"})
+                    out = pd.concat([out, fb], ignore_index=True)
+                except Exception:  # noqa: BLE001 — fallback is best-effort
+                    pass
+        out = out.fillna(0.0)
+        for idx, (model, use_feat) in enumerate(
+            zip(self.first_level_models, self.use_first_level_models_feat)
+        ):
+            if use_feat:
+                out = self._model_features(model, idx, out)
+        return out
 
     def fit(self, dataset: Dataset) -> "TwoStagesScenario":
         from sklearn.ensemble import GradientBoostingClassifier
@@ -74,10 +145,32 @@ class TwoStagesScenario:
         )
         for model in self.first_level_models:
             model.fit(first_ds)
+        if self.fallback_model is not None:
+            self.fallback_model.fit(first_ds)
         candidates = self._candidate_frame(first_ds)
         positives = second_train[[self.query_column, self.item_column]].assign(__target=1)
         labeled = candidates.merge(positives, on=[self.query_column, self.item_column], how="left")
         labeled["__target"] = labeled["__target"].fillna(0).astype(int)
+        # negative-example strategy (reference :476-533): "first_level" keeps
+        # the non-interacted candidates (most relevant negatives); "random"
+        # replaces them with uniformly sampled pairs
+        rng = np.random.default_rng(self.seed)
+        neg_mask = labeled["__target"] == 0
+        if self.negatives_type == "random":
+            n_items = int(first_train[self.item_column].max()) + 1
+            negs = labeled[neg_mask]
+            rand_items = rng.integers(0, n_items, size=len(negs))
+            labeled.loc[neg_mask, self.item_column] = rand_items
+        if self.num_negatives and int(neg_mask.sum()) > 0:
+            keep_per_query = (
+                labeled[neg_mask]
+                .groupby(self.query_column, sort=False)
+                .head(self.num_negatives)
+                .index
+            )
+            labeled = pd.concat(
+                [labeled[~neg_mask], labeled.loc[keep_per_query]], ignore_index=False
+            ).sort_index()
         if self.use_generated_features:
             self._fp = HistoryBasedFeaturesProcessor(
                 query_column=self.query_column, item_column=self.item_column
@@ -115,3 +208,28 @@ class TwoStagesScenario:
 
     def fit_predict(self, dataset: Dataset, k: int) -> pd.DataFrame:
         return self.fit(dataset).predict(dataset, k)
+
+    def optimize(
+        self,
+        train_dataset: Dataset,
+        test_dataset: Dataset,
+        param_borders: Optional[List[Optional[dict]]] = None,
+        criterion=None,
+        k: int = 10,
+        budget: int = 10,
+    ) -> List[Optional[dict]]:
+        """Tune every first-level model (reference :650-705): each model's own
+        ``optimize`` runs with its slice of ``param_borders``."""
+        borders = param_borders or [None] * len(self.first_level_models)
+        if len(borders) != len(self.first_level_models):
+            raise ValueError("param_borders must have one entry per first-level model")
+        results = []
+        for model, border in zip(self.first_level_models, borders):
+            if hasattr(model, "optimize"):
+                kwargs = {"param_borders": border, "k": k, "budget": budget}
+                if criterion is not None:
+                    kwargs["criterion"] = criterion
+                results.append(model.optimize(train_dataset, test_dataset, **kwargs))
+            else:
+                results.append(None)
+        return results

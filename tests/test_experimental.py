@@ -225,3 +225,25 @@ def test_neural_ts_with_features(ds):
     recs = model.predict(full, k=3)
     assert len(recs) > 0
     assert recs.groupby("query_id").size().max() <= 3
+
+
+def test_two_stages_deepened(ds):
+    """Fallback fill, first-level embedding features, negatives strategies,
+    optimize passthrough."""
+    from replay_amd.experimental.scenarios.two_stages.two_stages_scenario import TwoStagesScenario
+    from replay_amd.models import ALSWrap
+
+    for negatives_type in ("first_level", "random"):
+        sc = TwoStagesScenario(
+            first_level_models=[ALSWrap(rank=4, num_iterations=2, seed=0)],
+            num_candidates=5,
+            use_first_level_models_feat=True,
+            num_negatives=3,
+            negatives_type=negatives_type,
+            seed=0,
+        )
+        recs = sc.fit_predict(ds, k=3)
+        assert len(recs) > 0
+        assert recs.groupby("query_id").size().max() <= 3
+    # embedding features reached the ranker
+    assert any(c.startswith("m0_fm") for c in sc._feature_cols)
