@@ -105,16 +105,27 @@ def build_model(device, model_name="sasrec", loss_name="ce", sparse_embedding=Fa
     return model
 
 
-def make_batches(n_batches, batch_size, device, seed, model_name="sasrec", n_negatives=0):
-    """Synthetic ML-20M-shape sequence batches, generated on device."""
+def make_batches(n_batches, batch_size, device, seed, model_name="sasrec", n_negatives=0,
+                 ragged=False):
+    """Synthetic ML-20M-shape sequence batches, generated on device.
+
+    ragged=True draws per-row history lengths (left-padded, like the real
+    tokenized ML-20M sequences) so the masked attention/CE paths are in the
+    timed region instead of the idealized all-valid mask."""
     gen = torch.Generator(device="cpu").manual_seed(seed)
     batches = []
     for _ in range(n_batches):
         items = torch.randint(0, N_ITEMS, (batch_size, SEQ_LEN + 1), generator=gen)
+        if ragged:
+            lengths = torch.randint(2, SEQ_LEN + 1, (batch_size,), generator=gen)
+            pos = torch.arange(SEQ_LEN).unsqueeze(0)
+            pmask = pos >= (SEQ_LEN - lengths.unsqueeze(1))  # left padding
+        else:
+            pmask = torch.ones(batch_size, SEQ_LEN, dtype=torch.bool)
         batch = {
             "item_id": items[:, :-1].to(device),
             "labels": items[:, 1:].to(device),
-            "padding_mask": torch.ones(batch_size, SEQ_LEN, dtype=torch.bool, device=device),
+            "padding_mask": pmask.to(device),
         }
         batch["labels_padding_mask"] = batch["padding_mask"]
         if n_negatives:
@@ -446,6 +457,8 @@ def main() -> None:
                         help="train loss: full-softmax CE or shared-pool sampled CE (K9 fused)")
     parser.add_argument("--negatives", type=int, default=8192,
                         help="sampledce: shared negative-pool size per step")
+    parser.add_argument("--ragged", action="store_true",
+                        help="train: realistic variable-length (left-padded) sequences")
     parser.add_argument("--dense-emb", action="store_true",
                         help="disable sparse embedding gradients (twotower / sampledce modes)")
     parser.add_argument(
@@ -536,6 +549,7 @@ def main() -> None:
     batches = make_batches(
         4, args.batch, device, seed=1000 + rank, model_name=args.model,
         n_negatives=args.negatives if args.loss == "sampledce" else 0,
+        ragged=args.ragged,
     )
 
     amp_dtype = torch.bfloat16

@@ -534,3 +534,47 @@ print("FUSE_DH OK")
         )
         assert proc.returncode == 0, proc.stderr[-1500:]
         assert "FUSE_DH OK" in proc.stdout
+
+
+@requires_gpu
+class TestALSOnGPU:
+    def test_als_fit_predict_cuda(self):
+        """The batched-Cholesky ALS solve on the GPU (torch.linalg.cholesky
+        over hipSOLVER) matches the CPU solve."""
+        import numpy as np
+        import pandas as pd
+
+        from replay_amd.data import Dataset, FeatureHint, FeatureInfo, FeatureSchema, FeatureType
+        from replay_amd.models import ALSWrap
+
+        rng = np.random.default_rng(0)
+        df = pd.DataFrame(
+            {
+                "query_id": rng.integers(0, 200, 5000),
+                "item_id": rng.integers(0, 300, 5000),
+                "rating": rng.random(5000) + 0.5,
+                "timestamp": np.arange(5000),
+            }
+        ).drop_duplicates(["query_id", "item_id"])
+        schema = FeatureSchema(
+            [
+                FeatureInfo("query_id", FeatureType.CATEGORICAL, FeatureHint.QUERY_ID),
+                FeatureInfo("item_id", FeatureType.CATEGORICAL, FeatureHint.ITEM_ID),
+                FeatureInfo("rating", FeatureType.NUMERICAL, FeatureHint.RATING),
+                FeatureInfo("timestamp", FeatureType.NUMERICAL, FeatureHint.TIMESTAMP),
+            ]
+        )
+        ds = Dataset(feature_schema=schema, interactions=df, categorical_encoded=True)
+        gpu = ALSWrap(rank=16, num_iterations=3, seed=0, device="cuda")
+        gpu.fit(ds)
+        recs_gpu = gpu.predict(ds, k=5)
+        cpu = ALSWrap(rank=16, num_iterations=3, seed=0, device="cpu")
+        cpu.fit(ds)
+        recs_cpu = cpu.predict(ds, k=5)
+        # same seed + deterministic alternating solves: factor spaces agree
+        # up to numerics; compare top-5 overlap per user
+        g = recs_gpu.groupby("query_id")["item_id"].apply(set)
+        c = recs_cpu.groupby("query_id")["item_id"].apply(set)
+        common = g.index.intersection(c.index)
+        overlap = np.mean([len(g[q] & c[q]) / 5 for q in common])
+        assert overlap > 0.8, overlap
