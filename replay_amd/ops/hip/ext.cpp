@@ -40,6 +40,8 @@ std::vector<torch::Tensor> ce_linear_bwd(torch::Tensor hidden, torch::Tensor w,
                                          torch::Tensor gscale, double gsign);
 torch::Tensor metrics_reduce(torch::Tensor preds, torch::Tensor gt,
                              c10::optional<torch::Tensor> train, torch::Tensor ks);
+std::vector<torch::Tensor> scored_topk_gemm_fp8(torch::Tensor q, torch::Tensor w,
+                                                torch::Tensor thresholds, int64_t capacity);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layer_norm_fwd", &layer_norm_fwd, "fused LayerNorm forward (gfx950)");
@@ -57,6 +59,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused linear+CE forward: online LSE in the GEMM epilogue (gfx950)");
   m.def("ce_linear_bwd", &ce_linear_bwd,
         "fused linear+CE backward: recomputed dlogits + fused dhidden (gfx950)");
+  m.def("scored_topk_gemm_fp8", &scored_topk_gemm_fp8,
+        "fused e4m3 MFMA score-GEMM + top-k candidate selection (gfx950)");
   m.def("metrics_reduce", &metrics_reduce,
         "fused ranking-metric sums: hits + all cutoffs in one launch (gfx950)");
 }

@@ -778,7 +778,203 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
   }
 }
 
+// ---------------------------------------------------------------------------
+// fp8 (OCP e4m3) variant of the v4 pipeline for the config-5 serve path:
+// both operands quantized e4m3 (global amax scales folded into the
+// thresholds by the host; the raw fp32 accumulators come back and the host
+// multiplies by scale AFTER selection — monotonic, so selection is
+// unchanged).  Same glds triple-buffer + raw-barrier + counted-vmcnt
+// structure; W rows are E bytes (half the bf16 stream), LDS rows padded to
+// E+48 so stride/4 == 12 (mod 64) keeps the two 32-lane ds_read_b64 groups
+// conflict-free.  MFMA: v_mfma_f32_16x16x32_fp8_fp8 (i64 operands = 8
+// packed e4m3; same fragment index map as the bf16 form, bf16-rate).
+template <int E, int MF>
+__global__ __launch_bounds__(512, 2) void scored_topk_gemm_fp8_kernel(
+    const unsigned char* __restrict__ q,  // [M, E] e4m3
+    const unsigned char* __restrict__ w,  // [V, E] e4m3
+    const float* __restrict__ thresholds,  // [M] (pre-divided by scale)
+    float* __restrict__ out_vals,          // [M, cap] RAW accumulators
+    int* __restrict__ out_idx,             // [M, cap]
+    int* __restrict__ counts,              // [M]
+    int M, int64_t V64, int cap) {
+  constexpr int KSTEPS = E / 32;
+  constexpr int ROW_B = E + 48;                 // 304 at E=256
+  constexpr int CHUNKS_ROW = ROW_B / 16;        // 19
+  constexpr int TILE_PIECES = 64 * CHUNKS_ROW / 64;
+  constexpr int TILE_B = 64 * ROW_B;
+  const int V = (int)V64;
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int m0 = blockIdx.x * (128 * MF) + wave * (16 * MF);
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];  // 3 x TILE_B
+
+  long a_frag[MF][KSTEPS];
+#pragma unroll
+  for (int mf = 0; mf < MF; ++mf) {
+    const int row = m0 + mf * 16 + (lane & 15);
+    const unsigned char* qr = q + (size_t)min(row, M - 1) * E + (lane >> 4) * 8;
+#pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks) {
+      a_frag[mf][ks] = *reinterpret_cast<const long*>(qr + ks * 32);
+    }
+    if (row >= M) {
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks) a_frag[mf][ks] = 0;
+    }
+  }
+  float t_reg[MF][4];
+#pragma unroll
+  for (int mf = 0; mf < MF; ++mf) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = m0 + mf * 16 + (lane >> 4) * 4 + r;
+      t_reg[mf][r] = (row < M) ? thresholds[row] : INFINITY;
+    }
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+
+  const int n_tiles = (V + 63) >> 6;
+  const int tile0 = blockIdx.y;
+  const int tile_stride = gridDim.y;
+  if (tile0 >= n_tiles) return;
+
+  auto stage_tile = [&](int buf, int tile) {
+    const char* wt = reinterpret_cast<const char*>(w) + (size_t)((unsigned)tile << 6) * E;
+    const bool tail = ((tile << 6) + 64) > V;
+    char* lds_base = smem + (size_t)buf * TILE_B;
+    for (int piece = wave; piece < TILE_PIECES; piece += 8) {
+      const int c = piece * 64 + lane;
+      int item = c / CHUNKS_ROW;
+      int sub = c % CHUNKS_ROW;
+      if (sub >= E / 16) sub = 0;
+      if (tail) {
+        const int gitem = (tile << 6) + item;
+        item -= (gitem >= V ? (gitem - (V - 1)) : 0);
+      }
+      const unsigned off = (unsigned)item * E + sub * 16;
+      __builtin_amdgcn_global_load_lds(
+          (const void*)(wt + off), (void*)(lds_base + (size_t)piece * 1024), 16, 0, 0);
+    }
+  };
+
+  stage_tile(0, tile0);
+  if (tile0 + tile_stride < n_tiles) stage_tile(1, tile0 + tile_stride);
+
+  int cur = 0;
+  for (int tile = tile0; tile < n_tiles; tile += tile_stride) {
+    constexpr int P_HI = (TILE_PIECES + 7) / 8;
+    constexpr int P_LO = TILE_PIECES / 8;
+    if (tile + tile_stride < n_tiles) {
+      if (wave < (TILE_PIECES & 7)) {
+        waitcnt_vm<P_HI>();
+      } else {
+        waitcnt_vm<P_LO>();
+      }
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+
+    const char* bbuf = smem + (size_t)cur * TILE_B;
+    const int n0 = tile << 6;
+    long bfr[2][KSTEPS];
+    {
+      const char* bcol0 = bbuf + (size_t)(lane & 15) * ROW_B + (lane >> 4) * 8;
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks) {
+        bfr[0][ks] = *reinterpret_cast<const long*>(bcol0 + ks * 32);
+      }
+    }
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      if (f < 3) {
+        const char* bcoln = bbuf + (size_t)((f + 1) * 16 + (lane & 15)) * ROW_B + (lane >> 4) * 8;
+#pragma unroll
+        for (int ks = 0; ks < KSTEPS; ++ks) {
+          bfr[(f + 1) & 1][ks] = *reinterpret_cast<const long*>(bcoln + ks * 32);
+        }
+      }
+      f32x4 acc[MF];
+#pragma unroll
+      for (int mf = 0; mf < MF; ++mf) acc[mf] = f32x4{0.f, 0.f, 0.f, 0.f};
+      const int item = n0 + f * 16 + (lane & 15);
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks) {
+#pragma unroll
+        for (int mf = 0; mf < MF; ++mf) {
+          acc[mf] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+              a_frag[mf][ks], bfr[f & 1][ks], acc[mf], 0, 0, 0);
+        }
+      }
+      bool any_hit = false;
+      if (item < V) {
+#pragma unroll
+        for (int mf = 0; mf < MF; ++mf) {
+#pragma unroll
+          for (int r = 0; r < 4; ++r) any_hit |= (acc[mf][r] >= t_reg[mf][r]);
+        }
+      }
+      if (__builtin_amdgcn_ballot_w64(any_hit) != 0) {
+#pragma unroll
+        for (int mf = 0; mf < MF; ++mf) {
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const float v = acc[mf][r];
+            if (item < V && v >= t_reg[mf][r]) {
+              const int row = m0 + mf * 16 + (lane >> 4) * 4 + r;
+              const int pos = atomicAdd(&counts[row], 1);
+              if (pos < cap) {
+                out_vals[(size_t)row * cap + pos] = v;
+                out_idx[(size_t)row * cap + pos] = item;
+              }
+            }
+          }
+        }
+      }
+    }
+    {
+      const int t2 = tile + 2 * tile_stride;
+      if (t2 < n_tiles) stage_tile((cur + 2) % 3, t2);
+    }
+    cur = (cur + 1) % 3;
+  }
+}
+
 }  // namespace
+
+std::vector<torch::Tensor> scored_topk_gemm_fp8(torch::Tensor q, torch::Tensor w,
+                                                torch::Tensor thresholds, int64_t capacity) {
+  TORCH_CHECK(q.is_cuda() && q.dim() == 2 && q.is_contiguous());
+  TORCH_CHECK(w.is_cuda() && w.dim() == 2 && w.is_contiguous());
+  TORCH_CHECK(q.scalar_type() == torch::kFloat8_e4m3fn && w.scalar_type() == torch::kFloat8_e4m3fn,
+              "scored_topk_gemm_fp8 wants e4m3 operands");
+  const int M = (int)q.size(0);
+  const int E = (int)q.size(1);
+  const int64_t V = w.size(0);
+  TORCH_CHECK(w.size(1) == E, "dim mismatch");
+  TORCH_CHECK(V < (int64_t)INT32_MAX, "catalog must fit int32 indices");
+  TORCH_CHECK(E == 256 || E == 128, "scored_topk_gemm_fp8 supports E in {128, 256}");
+  auto opts_f = q.options().dtype(torch::kFloat32);
+  auto opts_i = q.options().dtype(torch::kInt32);
+  auto out_vals = torch::full({M, capacity}, -std::numeric_limits<float>::infinity(), opts_f);
+  auto out_idx = torch::zeros({M, capacity}, opts_i);
+  auto counts = torch::zeros({M}, opts_i);
+  auto thr = thresholds.to(torch::kFloat32).contiguous();
+  const int m_tile_rows = 512;
+  const int m_tiles = (M + m_tile_rows - 1) / m_tile_rows;
+  int stripes = (int)std::min<int64_t>((V + 63) / 64, std::max(1, 4096 / m_tiles));
+  dim3 grid(m_tiles, stripes);
+  auto stream = at::cuda::getCurrentHIPStream();
+#define LAUNCH_FP8(EE)                                                                       hipLaunchKernelGGL((scored_topk_gemm_fp8_kernel<EE, 4>), grid, dim3(512),                                     3 * 64 * (EE + 48), stream,                                                                reinterpret_cast<const unsigned char*>(q.data_ptr()),                                      reinterpret_cast<const unsigned char*>(w.data_ptr()),                                      thr.data_ptr<float>(), out_vals.data_ptr<float>(),                                         out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,                                     (int)capacity)
+  if (E == 256) {
+    LAUNCH_FP8(256);
+  } else {
+    LAUNCH_FP8(128);
+  }
+#undef LAUNCH_FP8
+  return {out_vals, out_idx, counts};
+}
 
 std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
                                             torch::Tensor thresholds, int64_t capacity) {

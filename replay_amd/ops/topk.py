@@ -255,19 +255,8 @@ def quantize_fp8(t: torch.Tensor, rowwise: bool = False):
     return q, scale.to(torch.float32)
 
 
-def catalog_topk_fp8(
-    q8: torch.Tensor,  # [B, E] float8_e4m3fn
-    scale_q: torch.Tensor,
-    w8: torch.Tensor,  # [V, E] float8_e4m3fn
-    scale_w: torch.Tensor,
-    k: int,
-    seen: Optional[torch.Tensor] = None,
-    chunk_items: int = 2**21,
-) -> Tuple[torch.Tensor, torch.Tensor]:
-    """Full-catalog top-K with the score GEMM on the fp8 MFMA pipes
-    (hipBLASLt _scaled_mm; e4m3 inputs, bf16 out).  Selection via
-    fast_row_topk; ranking is by fp8-precision scores (the config-5
-    contract — half the item-table bytes of bf16)."""
+def _catalog_topk_fp8_chunked(q8, scale_q, w8, scale_w, k, seen, chunk_items):
+    """Chunked _scaled_mm fallback (pre-round-2 path)."""
     V = w8.shape[0]
     k = min(k, V)
     run_scores = run_ids = None
@@ -287,3 +276,79 @@ def catalog_topk_fp8(
             sel_s, sel_pos = torch.topk(merged_s, min(k, merged_s.shape[1]), dim=1)
             run_scores, run_ids = sel_s, merged_i.gather(1, sel_pos)
     return run_scores, run_ids
+
+
+def catalog_topk_fp8(
+    q8: torch.Tensor,  # [B, E] float8_e4m3fn
+    scale_q: torch.Tensor,
+    w8: torch.Tensor,  # [V, E] float8_e4m3fn
+    scale_w: torch.Tensor,
+    k: int,
+    seen: Optional[torch.Tensor] = None,
+    chunk_items: int = 2**21,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Full-catalog top-K with the score GEMM on the fp8 MFMA pipes.
+
+    Default: the FUSED e4m3 kernel (scored_topk_gemm_fp8 — half the
+    item-table stream bytes of bf16, v_mfma_f32_16x16x32_fp8_fp8, in-epilogue
+    selection against scale-folded thresholds; the raw accumulators are
+    rescaled after selection, which is monotonic so the ranking contract is
+    unchanged).  Falls back to chunked _scaled_mm for unsupported shapes or
+    rows whose threshold guess fails."""
+    import math
+
+    from replay_amd.ops import hip_ext
+
+    ext = hip_ext()
+    V, E = w8.shape
+    M = q8.shape[0]
+    k = min(k, V)
+    fused_ok = (
+        ext is not None
+        and hasattr(ext, "scored_topk_gemm_fp8")
+        and E in (128, 256)
+        and V >= 65536
+        and k <= 512
+    )
+    if not fused_ok:
+        return _catalog_topk_fp8_chunked(q8, scale_q, w8, scale_w, k, seen, chunk_items)
+    scale = (scale_q.float() * scale_w.float()).reshape(())
+    # threshold estimate from a strided subsample (fp8 GEMM, bf16 out)
+    stride = max(1, V // 32768)
+    sample_items = w8[::stride].contiguous()
+    sample = torch._scaled_mm(
+        q8, sample_items.t(), scale_a=scale_q, scale_b=scale_w, out_dtype=torch.bfloat16
+    ).float()
+    qr = sample.shape[1] / V
+    j = max(1, math.ceil(k * qr + 3.0 * math.sqrt(max(k * qr, 1e-9)) + 2))
+    if seen is not None:
+        j += math.ceil(seen.shape[1] * qr) + 1
+    j = min(j, sample.shape[1])
+    thresholds = sample.topk(j, dim=1).values[:, -1] / scale  # raw-accumulator units
+    capacity = max(4 * k, int(5.0 * j / qr))
+    vals, idx, counts = ext.scored_topk_gemm_fp8(
+        q8.contiguous(), w8.contiguous(), thresholds, capacity
+    )
+    if seen is not None:
+        gid = idx.long()
+        written = torch.isfinite(vals)
+        sorted_seen, _ = seen.sort(dim=1)
+        pos = torch.searchsorted(sorted_seen, gid).clamp(max=sorted_seen.shape[1] - 1)
+        hit = (sorted_seen.gather(1, pos) == gid) & written
+        vals = vals.masked_fill(hit, float("-inf"))
+        survivors = counts.clamp(max=capacity) - hit.sum(-1, dtype=counts.dtype)
+        bad = (survivors < k) | (counts > capacity)
+    else:
+        bad = (counts < k) | (counts > capacity)
+    kk = min(k, capacity)
+    top_s, top_pos = torch.topk(vals, kk, dim=1)
+    top_i = idx.gather(1, top_pos).long()
+    top_s = (top_s * scale).to(torch.bfloat16)
+    if bool(bad.any()):
+        rows = torch.nonzero(bad).squeeze(-1)
+        sub_s, sub_i = _catalog_topk_fp8_chunked(
+            q8[rows], scale_q, w8, scale_w, kk,
+            seen[rows] if seen is not None else None, chunk_items,
+        )
+        top_s[rows], top_i[rows] = sub_s.to(top_s.dtype), sub_i
+    return top_s, top_i
