@@ -315,7 +315,8 @@ def catalog_topk_fp8(
     scale = (scale_q.float() * scale_w.float()).reshape(())
     # threshold estimate from a strided subsample (fp8 GEMM, bf16 out)
     stride = max(1, V // 32768)
-    sample_items = w8[::stride].contiguous()
+    sample_items = w8[::stride]
+    sample_items = sample_items[: (sample_items.shape[0] // 16) * 16].contiguous()
     sample = torch._scaled_mm(
         q8, sample_items.t(), scale_a=scale_q, scale_b=scale_w, out_dtype=torch.bfloat16
     ).float()
