@@ -262,9 +262,13 @@ def _catalog_topk_fp8_chunked(q8, scale_q, w8, scale_w, k, seen, chunk_items):
     run_scores = run_ids = None
     for lo in range(0, V, chunk_items):
         hi = min(lo + chunk_items, V)
+        chunk = w8[lo:hi]
+        if chunk.shape[0] % 16:  # _scaled_mm wants multiples of 16 rows
+            pad = 16 - chunk.shape[0] % 16
+            chunk = torch.cat([chunk, torch.zeros(pad, chunk.shape[1], dtype=chunk.dtype, device=chunk.device)])
         scores = torch._scaled_mm(
-            q8, w8[lo:hi].t(), scale_a=scale_q, scale_b=scale_w, out_dtype=torch.bfloat16
-        )
+            q8, chunk.t(), scale_a=scale_q, scale_b=scale_w, out_dtype=torch.bfloat16
+        )[:, : hi - lo]
         kk = min(k, hi - lo)
         top_s, top_i = fast_row_topk(scores, kk, seen=seen, col_offset=lo)
         top_i = top_i + lo

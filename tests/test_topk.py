@@ -77,3 +77,25 @@ def test_catalog_topk_property_random(seed):
     same = ids == ref_i
     ties = torch.isclose(scores, ref_s, atol=1e-4)
     assert (same | ties).all()
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(not torch.cuda.is_available(), reason="needs GPU")
+def test_fused_fp8_matches_chunked():
+    """The fused e4m3 kernel ranks like the chunked _scaled_mm path (same
+    fp8 precision; near-tie boundary reshuffling allowed)."""
+    from replay_amd.ops.topk import _catalog_topk_fp8_chunked, catalog_topk_fp8, quantize_fp8
+
+    torch.manual_seed(3)
+    for M, V, E in [(1024, 1_000_000, 256), (512, 100_003, 128)]:
+        q = torch.randn(M, E, device="cuda", dtype=torch.bfloat16)
+        w = torch.randn(V, E, device="cuda", dtype=torch.bfloat16)
+        seen = torch.randint(0, V, (M, 64), device="cuda")
+        q8, sq = quantize_fp8(q)
+        w8, sw = quantize_fp8(w)
+        v_f, i_f = catalog_topk_fp8(q8, sq, w8, sw, 100, seen)
+        v_c, i_c = _catalog_topk_fp8_chunked(q8, sq, w8, sw, 100, seen, 2**21)
+        inter = sum(
+            len(set(i_f[r].tolist()) & set(i_c[r].tolist())) for r in range(M)
+        )
+        assert inter / (M * 100) > 0.97
