@@ -36,6 +36,8 @@ class CategoricalEmbedding(torch.nn.Module):
         self.item_emb = torch.nn.Embedding(
             cardinality + 1 + n_extra, embedding_dim, padding_idx=cardinality, sparse=sparse
         )
+        if sparse:
+            self.item_emb.weight._replay_sparse_grad = True
         # reference embedding.py:199: xavier-normal table init (the torch
         # N(0,1) default puts the initial full-softmax CE at ~5x ln(V)
         # through the sqrt(d)-scaled tied head)

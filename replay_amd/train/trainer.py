@@ -141,9 +141,21 @@ class Trainer:
         module.to(self.device)
         self._module = module
         model = module
+        # K6: sparse-gradient embedding tables are excluded from DDP (RCCL
+        # cannot all-reduce sparse) and synced explicitly after backward
+        self._sparse_params = [
+            p for p in module.parameters() if getattr(p, "_replay_sparse_grad", False)
+        ]
         if _dist_ok() and self.world_size > 1 and any(p.requires_grad for p in module.parameters()):
+            shim = _TrainStepShim(module)
+            if self._sparse_params:
+                sparse_ids = {id(p) for p in self._sparse_params}
+                ignore = [n for n, p in shim.named_parameters() if id(p) in sparse_ids]
+                torch.nn.parallel.DistributedDataParallel._set_params_and_buffers_to_ignore_for_model(
+                    shim, ignore
+                )
             model = torch.nn.parallel.DistributedDataParallel(
-                _TrainStepShim(module),
+                shim,
                 device_ids=[self.device.index] if self.device.type == "cuda" else None,
                 # xGMI is point-to-point (7 links/GPU): few large buckets beat
                 # many small ones for ring all-reduce on small models
@@ -217,8 +229,16 @@ class Trainer:
                         loss = loss["loss"]
                 self._optimizer.zero_grad(set_to_none=True)
                 loss.backward()
+                if self._sparse_params and _dist_ok() and self.world_size > 1:
+                    from replay_amd.parallel import sync_sparse_grads
+
+                    sync_sparse_grads(self._sparse_params)
                 if self.gradient_clip_val:
-                    torch.nn.utils.clip_grad_norm_(module.parameters(), self.gradient_clip_val)
+                    dense = [
+                        p for p in module.parameters()
+                        if not getattr(p, "_replay_sparse_grad", False)
+                    ]
+                    torch.nn.utils.clip_grad_norm_(dense, self.gradient_clip_val)
                 self._optimizer.step()
                 if self._scheduler is not None:
                     self._scheduler.step()

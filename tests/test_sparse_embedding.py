@@ -97,3 +97,58 @@ def test_sync_sparse_grads_two_ranks():
     for p in procs:
         p.join(120)
     assert all(p.exitcode == 0 for p in procs)
+
+
+def test_trainer_with_sparse_embedding():
+    """The framework Trainer + OptimizerFactory run a sparse-embedding model
+    end-to-end (HybridSparseOptimizer: Adam + SparseAdam)."""
+    from replay_amd.data.nn import TensorFeatureInfo, TensorSchema
+    from replay_amd.data.schema import FeatureHint, FeatureType
+    from replay_amd.nn.lightning import LightningModule, OptimizerFactory
+    from replay_amd.nn.lightning.optimizer import HybridSparseOptimizer
+    from replay_amd.nn.loss import CESampled
+    from replay_amd.nn.sequential.sasrec import SasRec
+    from replay_amd.train.trainer import Trainer
+
+    V, L, E = 200, 6, 8
+    schema = TensorSchema(
+        [
+            TensorFeatureInfo(
+                "item_id", FeatureType.CATEGORICAL, is_seq=True,
+                feature_hint=FeatureHint.ITEM_ID, cardinality=V, embedding_dim=E,
+            )
+        ]
+    )
+    torch.manual_seed(0)
+    model = SasRec.from_params(
+        schema, max_sequence_length=L, embedding_dim=E, num_blocks=1, num_heads=2,
+        dropout=0.0, loss=CESampled(log_correction=True, vocab_size=V),
+        sparse_embedding=True,
+    )
+    module = LightningModule(model, OptimizerFactory(lr=1e-2))
+    opt, _ = module.configure_optimizers()
+    assert isinstance(opt, HybridSparseOptimizer)
+
+    def batches():
+        torch.manual_seed(1)
+        for _ in range(4):
+            items = torch.randint(0, V, (4, L + 1))
+            b = {
+                "item_id": items[:, :-1],
+                "labels": items[:, 1:],
+                "padding_mask": torch.ones(4, L, dtype=torch.bool),
+                "negatives": torch.randint(0, V, (32,)),
+            }
+            b["labels_padding_mask"] = b["padding_mask"]
+            yield b
+
+    class Loader:
+        def __iter__(self):
+            return batches()
+
+        def __len__(self):
+            return 4
+
+    trainer = Trainer(max_epochs=2, accelerator="cpu", precision="32")
+    trainer.fit(module, Loader())
+    assert "train_loss" in trainer.logged_metrics
