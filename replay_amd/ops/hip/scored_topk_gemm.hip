@@ -941,6 +941,129 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_fp8_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// v6 (E = 256): occupancy experiment — MF=2 (116 VGPRs) + 2 LDS buffers
+// (70 KB) lets TWO workgroups co-reside per CU (4 waves/SIMD), so one WG's
+// barrier/atomic stalls overlap the other's MFMA stream.  Costs 2x the W
+// passes of MF=4 (4 M-tiles), bets on L3 dedup of the co-walking tiles.
+template <int E>
+__global__ __launch_bounds__(512, 4) void scored_topk_gemm_kernel_v6(
+    const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ w,
+    const float* __restrict__ thresholds, float* __restrict__ out_vals,
+    int* __restrict__ out_idx, int* __restrict__ counts, int M, int64_t V64, int cap) {
+  constexpr int MF = 2;
+  constexpr int KSTEPS = E / 32;
+  constexpr int ROW_B = E * 2 + 32;
+  constexpr int CHUNKS_ROW = ROW_B / 16;
+  constexpr int TILE_PIECES = 64 * CHUNKS_ROW / 64;
+  constexpr int TILE_B = 64 * ROW_B;
+  const int V = (int)V64;
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int m0 = blockIdx.x * (128 * MF) + wave * (16 * MF);
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];  // 2 x TILE_B
+
+  bf16x8 a_frag[MF][KSTEPS];
+#pragma unroll
+  for (int mf = 0; mf < MF; ++mf) {
+    const int row = m0 + mf * 16 + (lane & 15);
+    const __hip_bfloat16* qr = q + (size_t)min(row, M - 1) * E + (lane >> 4) * 8;
+#pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks) a_frag[mf][ks] = *reinterpret_cast<const bf16x8*>(qr + ks * 32);
+    if (row >= M) {
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks) a_frag[mf][ks] = bf16x8{0};
+    }
+  }
+  float t_reg[MF][4];
+#pragma unroll
+  for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = m0 + mf * 16 + (lane >> 4) * 4 + r;
+      t_reg[mf][r] = (row < M) ? thresholds[row] : INFINITY;
+    }
+
+  const int n_tiles = (V + 63) >> 6;
+  const int tile0 = blockIdx.y;
+  const int tile_stride = gridDim.y;
+  if (tile0 >= n_tiles) return;
+
+  auto stage_tile = [&](int buf, int tile) {
+    const char* wt = reinterpret_cast<const char*>(w) + (size_t)((unsigned)tile << 6) * (E * 2);
+    const bool tail = ((tile << 6) + 64) > V;
+    char* lds_base = smem + (size_t)buf * TILE_B;
+    for (int piece = wave; piece < TILE_PIECES; piece += 8) {
+      const int c = piece * 64 + lane;
+      int item = c / CHUNKS_ROW;
+      int sub = c % CHUNKS_ROW;
+      if (sub >= E * 2 / 16) sub = 0;
+      if (tail) {
+        const int gitem = (tile << 6) + item;
+        item -= (gitem >= V ? (gitem - (V - 1)) : 0);
+      }
+      const unsigned off = (unsigned)item * (E * 2) + sub * 16;
+      __builtin_amdgcn_global_load_lds(
+          (const void*)(wt + off), (void*)(lds_base + (size_t)piece * 1024), 16, 0, 0);
+    }
+  };
+
+  stage_tile(0, tile0);
+  int cur = 0;
+  for (int tile = tile0; tile < n_tiles; tile += tile_stride) {
+    const int nxt = tile + tile_stride;
+    if (nxt < n_tiles) stage_tile(cur ^ 1, nxt);
+    __syncthreads();  // drains buf[cur]'s glds (vmcnt folded by hipcc)
+    const char* bbuf = smem + (size_t)cur * TILE_B;
+    const int n0 = tile << 6;
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      // single-buffer fragment reads: at 4 waves/SIMD the ds_read latency
+      // hides behind the other waves (TLP), keeping VGPRs at the
+      // occupancy-4 cap
+      const char* bcol = bbuf + (size_t)(f * 16 + (lane & 15)) * ROW_B + (lane >> 4) * 16;
+      bf16x8 bfr[KSTEPS];
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks) bfr[ks] = *reinterpret_cast<const bf16x8*>(bcol + ks * 64);
+      f32x4 acc[MF];
+#pragma unroll
+      for (int mf = 0; mf < MF; ++mf) acc[mf] = f32x4{0.f, 0.f, 0.f, 0.f};
+      const int item = n0 + f * 16 + (lane & 15);
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks)
+#pragma unroll
+        for (int mf = 0; mf < MF; ++mf)
+          acc[mf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag[mf][ks], bfr[ks], acc[mf], 0, 0, 0);
+      bool any_hit = false;
+      if (item < V) {
+#pragma unroll
+        for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) any_hit |= (acc[mf][r] >= t_reg[mf][r]);
+      }
+      if (__builtin_amdgcn_ballot_w64(any_hit) != 0) {
+#pragma unroll
+        for (int mf = 0; mf < MF; ++mf)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const float v = acc[mf][r];
+            if (item < V && v >= t_reg[mf][r]) {
+              const int row = m0 + mf * 16 + (lane >> 4) * 4 + r;
+              const int pos = atomicAdd(&counts[row], 1);
+              if (pos < cap) {
+                out_vals[(size_t)row * cap + pos] = v;
+                out_idx[(size_t)row * cap + pos] = item;
+              }
+            }
+          }
+      }
+    }
+    __syncthreads();  // buffer-swap guard
+    cur ^= 1;
+  }
+}
+
 }  // namespace
 
 std::vector<torch::Tensor> scored_topk_gemm_fp8(torch::Tensor q, torch::Tensor w,
@@ -996,7 +1119,8 @@ std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
   const bool legacy = (variant != nullptr && variant[0] < '4');
   // v4 defaults: 512-row M-tile at E=256 (MF=4), 1024-row at E<=128 (MF=8);
   // wider M-tiles divide the number of passes over the streamed item table
-  const int v4_mf = (variant != nullptr && variant[0] == '5') ? 2 : (E == 64 ? 8 : 4);
+  const bool v6 = (variant != nullptr && variant[0] == 'o');
+  const int v4_mf = ((variant != nullptr && variant[0] == '5') || v6) ? 2 : (E == 64 ? 8 : 4);
   const int m_tile_rows = legacy ? 256 : 128 * v4_mf;
   const int m_tiles = (M + m_tile_rows - 1) / m_tile_rows;
   // fill 256 CUs x ~4 blocks with >> WGs (guide §1); stripes over item tiles
@@ -1026,6 +1150,14 @@ std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
   } else if (E == 128) {
     const size_t lds128 = 3 * 64 * (128 * 2 + 32);
     hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<128, 4>), grid, dim3(512), lds128, stream,
+                       reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                       reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                       thr.data_ptr<float>(), out_vals.data_ptr<float>(),
+                       out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
+                       (int)capacity);
+  } else if (E == 256 && v6) {
+    const size_t lds_v6 = 2 * 64 * (256 * 2 + 32);
+    hipLaunchKernelGGL((scored_topk_gemm_kernel_v6<256>), grid, dim3(512), lds_v6, stream,
                        reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
                        reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
                        thr.data_ptr<float>(), out_vals.data_ptr<float>(),
