@@ -1116,7 +1116,8 @@ std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
   auto counts = torch::zeros({M}, opts_i);
   auto thr = thresholds.to(torch::kFloat32).contiguous();
   static const char* variant = std::getenv("REPLAY_AMD_STG_VARIANT");
-  const bool legacy = (variant != nullptr && variant[0] < '4');
+  if (variant != nullptr && variant[0] == '\0') variant = nullptr;  // empty = default
+  const bool legacy = (variant != nullptr && variant[0] >= '1' && variant[0] < '4');
   // v4 defaults: 512-row M-tile at E=256 (MF=4), 1024-row at E<=128 (MF=8);
   // wider M-tiles divide the number of passes over the streamed item table
   const bool v6 = (variant != nullptr && variant[0] == 'o');
