@@ -597,12 +597,13 @@ __device__ __forceinline__ void waitcnt_vm() {
   else if constexpr (N == 6) asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
   else if constexpr (N == 8) asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
   else if constexpr (N == 10) asm volatile("s_waitcnt vmcnt(10)" ::: "memory");
+  else if constexpr (N == 17) asm volatile("s_waitcnt vmcnt(17)" ::: "memory");
   else static_assert(N == 0, "add a vmcnt immediate case");
 }
 
 // ABLATE: 0 full; 1 no-stage; 2 no-mfma; 3 no-epilogue; 4 no-epilogue +
 // no-barrier/vmcnt (pure ds_read+MFMA loop); 5 no-epilogue + no-stage
-template <int E, int MF, int NBUF = 3, int ABLATE = 0>
+template <int E, int MF, int NBUF = 3, int ABLATE = 0, int LOADERS = 0>
 __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
     const __hip_bfloat16* __restrict__ q,  // [M, E]
     const __hip_bfloat16* __restrict__ w,  // [V, E]
@@ -656,10 +657,12 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
   if (tile0 >= n_tiles) return;
 
   auto stage_tile = [&](int buf, int tile) {
+    constexpr int NLOAD = (LOADERS == 0) ? 8 : LOADERS;
+    if (LOADERS != 0 && wave >= LOADERS) return;
     const char* wt = reinterpret_cast<const char*>(w) + (size_t)((unsigned)tile << 6) * (E * 2);
     const bool tail = ((tile << 6) + 64) > V;
     char* lds_base = smem + (size_t)buf * TILE_B;
-    for (int piece = wave; piece < TILE_PIECES; piece += 8) {
+    for (int piece = wave; piece < TILE_PIECES; piece += NLOAD) {
       const int c = piece * 64 + lane;
       int item = c / CHUNKS_ROW;
       int sub = c % CHUNKS_ROW;
@@ -684,16 +687,20 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
     // own glds for buf[cur] complete; allow NBUF-2 newer tiles in flight
     // (when they were actually staged — at the walk's tail, full drain)
     if constexpr (ABLATE != 4) {
-      constexpr int P_HI = (TILE_PIECES + 7) / 8;
-      constexpr int P_LO = TILE_PIECES / 8;
-      if (tile + (int)(NBUF - 2) * tile_stride < n_tiles) {
-        if (wave < (TILE_PIECES & 7)) {
-          waitcnt_vm<P_HI * (NBUF - 2)>();
+      constexpr int NLOAD = (LOADERS == 0) ? 8 : LOADERS;
+      constexpr int P_HI = (TILE_PIECES + NLOAD - 1) / NLOAD;
+      constexpr int P_LO = TILE_PIECES / NLOAD;
+      const bool is_loader = (LOADERS == 0) || (wave < LOADERS);
+      if (is_loader) {
+        if (tile + (int)(NBUF - 2) * tile_stride < n_tiles) {
+          if (wave < (TILE_PIECES % NLOAD)) {
+            waitcnt_vm<P_HI * (NBUF - 2)>();
+          } else {
+            waitcnt_vm<P_LO * (NBUF - 2)>();
+          }
         } else {
-          waitcnt_vm<P_LO * (NBUF - 2)>();
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         }
-      } else {
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       }
       __builtin_amdgcn_s_barrier();
     }
@@ -951,6 +958,7 @@ __global__ __launch_bounds__(512, 4) void scored_topk_gemm_kernel_v6(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ w,
     const float* __restrict__ thresholds, float* __restrict__ out_vals,
     int* __restrict__ out_idx, int* __restrict__ counts, int M, int64_t V64, int cap) {
+  constexpr int LOADERS = 0;  // all waves stage (the shared stage_tile text)
   constexpr int MF = 2;
   constexpr int KSTEPS = E / 32;
   constexpr int ROW_B = E * 2 + 32;
@@ -991,10 +999,12 @@ __global__ __launch_bounds__(512, 4) void scored_topk_gemm_kernel_v6(
   if (tile0 >= n_tiles) return;
 
   auto stage_tile = [&](int buf, int tile) {
+    constexpr int NLOAD = (LOADERS == 0) ? 8 : LOADERS;
+    if (LOADERS != 0 && wave >= LOADERS) return;
     const char* wt = reinterpret_cast<const char*>(w) + (size_t)((unsigned)tile << 6) * (E * 2);
     const bool tail = ((tile << 6) + 64) > V;
     char* lds_base = smem + (size_t)buf * TILE_B;
-    for (int piece = wave; piece < TILE_PIECES; piece += 8) {
+    for (int piece = wave; piece < TILE_PIECES; piece += NLOAD) {
       const int c = piece * 64 + lane;
       int item = c / CHUNKS_ROW;
       int sub = c % CHUNKS_ROW;
@@ -1235,6 +1245,15 @@ std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
                            out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
                            (int)capacity);
       }
+    } else if (variant != nullptr && variant[0] == 'p') {
+      // loader-specialized staging: 2 waves carry the whole DMA stream
+      const size_t lds_p = 3 * 64 * (256 * 2 + 32);
+      hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 3, 0, 2>), grid, dim3(512), lds_p,
+                         stream, reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                         reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                         thr.data_ptr<float>(), out_vals.data_ptr<float>(),
+                         out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
+                         (int)capacity);
     } else if (variant != nullptr && variant[0] == '6') {
       // v4 with a 4-deep buffer ring (3 tiles of DMA in flight)
       const size_t lds_v46 = 4 * 64 * (256 * 2 + 32);
