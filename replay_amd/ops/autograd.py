@@ -92,6 +92,8 @@ class FusedLinearCrossEntropyFunction(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dloss):
+        import os
+
         ext = hip_ext()
         hidden2d, weight, labels, lse, valid, count = ctx.saved_tensors
         g = dloss.to(torch.float32) / count.clamp(min=1).to(torch.float32)
@@ -99,6 +101,17 @@ class FusedLinearCrossEntropyFunction(torch.autograd.Function):
         # the upstream gradient travels as a scalar
         gsign = float(torch.sign(g))
         gscale = torch.where(valid, g.abs(), torch.zeros((), device=valid.device))
+        if (
+            hidden2d.shape[1] <= 128
+            and hasattr(ext, "ce_linear_wgrad")
+            and os.environ.get("REPLAY_AMD_CE_WGRAD") != "0"
+        ):
+            # phase-split backward: dW from the item-owner wgrad kernel and
+            # dhidden from the store-free fused-dh pass — the [M, Vp] bf16
+            # dlogits tensor (22 GB at the flagship shape) never exists
+            dweight = ext.ce_linear_wgrad(hidden2d, weight, labels, lse, gscale, gsign)
+            dhidden = ext.ce_linear_bwd_fused_dh(hidden2d, weight, labels, lse, gscale, gsign)
+            return dhidden, dweight.to(weight.dtype), None, None
         dlogits, dhidden = ext.ce_linear_bwd(hidden2d, weight, labels, lse, gscale, gsign)
         if dhidden.numel() == 0:  # E > 128: host GEMM for the input gradient
             dhidden = dlogits @ weight

@@ -251,7 +251,7 @@ __global__ __launch_bounds__(256, 2) void ce_linear_fwd_kernel(
 //   - labels as int32 (catalogs < 2^31)
 //   - the score MFMA runs in two MF halves (B fragments re-read from L1),
 //     halving the live score-accumulator set: 64 -> 32
-template <int E, bool RESIDENT, bool FUSE_DH>
+template <int E, bool RESIDENT, bool FUSE_DH, bool WRITE_DL = true>
 __global__ __launch_bounds__(256, 2) void ce_linear_bwd_kernel(
     const __hip_bfloat16* __restrict__ hidden,  // [M, E]
     const __hip_bfloat16* __restrict__ w,       // [V, E]
@@ -475,13 +475,15 @@ __global__ __launch_bounds__(256, 2) void ce_linear_bwd_kernel(
           }
         }
       }
-      // coalesced store: each lane writes 16-B chunks of the bounce tile
-      for (int i = lane; i < 16 * 8; i += WAVE) {
-        const int row = row0 + i / 8;
-        const int c0 = (i % 8) * 8;
-        if (row < M) {
-          *reinterpret_cast<bf16x8*>(dlogits + (size_t)row * ldd + n0 + c0) =
-              *reinterpret_cast<const bf16x8*>(my_dl + dl_off(i / 8, c0));
+      if constexpr (WRITE_DL) {
+        // coalesced store: each lane writes 16-B chunks of the bounce tile
+        for (int i = lane; i < 16 * 8; i += WAVE) {
+          const int row = row0 + i / 8;
+          const int c0 = (i % 8) * 8;
+          if (row < M) {
+            *reinterpret_cast<bf16x8*>(dlogits + (size_t)row * ldd + n0 + c0) =
+                *reinterpret_cast<const bf16x8*>(my_dl + dl_off(i / 8, c0));
+          }
         }
       }
     }
@@ -504,7 +506,187 @@ __global__ __launch_bounds__(256, 2) void ce_linear_bwd_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// wgrad phase B (item-owner): dW = dlogits^T . hidden computed WITHOUT ever
+// materializing dlogits.  Each workgroup owns a 64-item column tile (wave =
+// 16 items, A fragments of W resident) and walks 64-row tiles of hidden:
+//   S^T[item, row] recomputed by MFMA (B fragments read straight from the
+//     row-major hidden — the operand swap keeps them contiguous),
+//   dS^T = (exp(S - lse) - onehot) * g formed on the accumulators and
+//     bounced through a wave-private LDS tile into A layout,
+//   dW_tile += dS^T . hidden via a second MFMA chain whose B fragments come
+//     from a per-row-tile hidden^T LDS stage (attention-bwd phase-B recipe).
+// The dW tile lives in registers for the whole walk and is stored once per
+// (item-tile, row-stripe) into a per-stripe fp32 slab; the host sums the
+// S slabs.  Replaces the 2x[M, Vp] bf16 dlogits round trip (22 GB at the
+// flagship shape) + the hipBLASLt wgrad GEMM.
+template <int E>
+__global__ __launch_bounds__(256, 4) void ce_linear_wgrad_kernel(
+    const __hip_bfloat16* __restrict__ hidden,  // [M, E]
+    const __hip_bfloat16* __restrict__ w,       // [V, E]
+    const int64_t* __restrict__ labels,         // [M]
+    const float* __restrict__ lse,              // [M]
+    const float* __restrict__ gscale,           // [M]
+    float gsign,
+    float* __restrict__ dw_slabs,  // [S, V, E] fp32 (plain stores)
+    int M, int V) {
+  constexpr int KSTEPS = E / 32;  // k over E (S^T GEMM)
+  constexpr int OF = E / 16;      // dW column fragments
+  const int wave = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int item0 = blockIdx.x * 64;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // [hidden^T stage: E x 64 bf16 swizzled][adj/g: 64 f32 each][lab: 64 i32]
+  // [4 waves x dS bounce 16 x 64 bf16]
+  __hip_bfloat16* ht = reinterpret_cast<__hip_bfloat16*>(smem);
+  float* adj_l = reinterpret_cast<float*>(ht + (size_t)E * 64);
+  float* g_l = adj_l + 64;
+  int* lab_l = reinterpret_cast<int*>(g_l + 64);
+  __hip_bfloat16* ds_tiles = reinterpret_cast<__hip_bfloat16*>(lab_l + 64);
+  __hip_bfloat16* my_ds = ds_tiles + (size_t)wave * 16 * 64;
+  auto ht_off = [&](int e, int row) { return e * 64 + (row ^ ((e & 7) << 3)); };
+  auto ds_off = [&](int itm, int row) { return itm * 64 + (row ^ ((itm & 7) << 3)); };
+
+  // resident A: this wave's 16 item rows of W
+  bf16x8 a_w[KSTEPS];
+  {
+    const int item = item0 + wave * 16 + (lane & 15);
+    const __hip_bfloat16* wr = w + (size_t)min(item, V - 1) * E + (lane >> 4) * 8;
+#pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks) a_w[ks] = *reinterpret_cast<const bf16x8*>(wr + ks * 32);
+    if (item >= V) {
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks) a_w[ks] = bf16x8{0};
+    }
+  }
+  f32x4 dwacc[OF];
+#pragma unroll
+  for (int f = 0; f < OF; ++f) dwacc[f] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  for (int r0 = blockIdx.y * 64; r0 < M; r0 += gridDim.y * 64) {
+    // ---- stage hidden^T [E][64] + per-row state ----
+    for (int i = threadIdx.x; i < 64 * (E / 8); i += blockDim.x) {
+      const int row = i / (E / 8);
+      const int e0 = (i % (E / 8)) * 8;
+      const int src = min(r0 + row, M - 1);
+      bf16x8 vv = *reinterpret_cast<const bf16x8*>(hidden + (size_t)src * E + e0);
+      if (r0 + row >= M) vv = bf16x8{0};
+#pragma unroll
+      for (int j = 0; j < 8; ++j) ht[ht_off(e0 + j, row)] = ((const __hip_bfloat16*)&vv)[j];
+    }
+    for (int i = threadIdx.x; i < 64; i += blockDim.x) {
+      const int row = r0 + i;
+      const float g = (row < M) ? gscale[row] : 0.f;
+      g_l[i] = g;
+      adj_l[i] = (g > 0.f && row < M) ? lse[row] - __logf(g) : INFINITY;
+      lab_l[i] = (row < M) ? (int)labels[row] : -1;
+    }
+    __syncthreads();
+
+    // ---- S^T = W . hidden^T (B fragments straight from global hidden) ----
+    f32x4 sacc[4];
+#pragma unroll
+    for (int f = 0; f < 4; ++f) sacc[f] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks) {
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        const int row = min(r0 + f * 16 + (lane & 15), M - 1);
+        bf16x8 b_h = *reinterpret_cast<const bf16x8*>(
+            hidden + (size_t)row * E + ks * 32 + (lane >> 4) * 8);
+        sacc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_w[ks], b_h, sacc[f], 0, 0, 0);
+      }
+    }
+    // ---- dS^T on the accumulators -> wave-private bounce tile ----
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int lrow = f * 16 + (lane & 15);
+        const int item = item0 + wave * 16 + (lane >> 4) * 4 + r;
+        float dl = 0.f;
+        if (item < V) {
+          dl = __expf(sacc[f][r] - adj_l[lrow]);
+          if (item == lab_l[lrow]) dl -= g_l[lrow];
+        }
+        my_ds[ds_off((lane >> 4) * 4 + r, lrow)] = __float2bfloat16(dl * gsign);
+      }
+    }
+    // ---- dW += dS^T . hidden (A from bounce, B from the hidden^T stage) ----
+#pragma unroll
+    for (int ks2 = 0; ks2 < 2; ++ks2) {
+      bf16x8 a_ds = *reinterpret_cast<const bf16x8*>(
+          my_ds + ds_off(lane & 15, ks2 * 32 + (lane >> 4) * 8));
+#pragma unroll
+      for (int f = 0; f < OF; ++f) {
+        bf16x8 b_ht = *reinterpret_cast<const bf16x8*>(
+            ht + ht_off(f * 16 + (lane & 15), ks2 * 32 + (lane >> 4) * 8));
+        dwacc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_ds, b_ht, dwacc[f], 0, 0, 0);
+      }
+    }
+    __syncthreads();  // everyone done with ht before the next stage
+  }
+  // ---- store this (item-tile, stripe)'s dW partial ----
+  float* slab = dw_slabs + (size_t)blockIdx.y * V * E;
+#pragma unroll
+  for (int f = 0; f < OF; ++f) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int item = item0 + wave * 16 + (lane >> 4) * 4 + r;
+      if (item < V) slab[(size_t)item * E + f * 16 + (lane & 15)] = dwacc[f][r];
+    }
+  }
+}
+
 }  // namespace
+
+torch::Tensor ce_linear_wgrad(torch::Tensor hidden, torch::Tensor w, torch::Tensor labels,
+                              torch::Tensor lse, torch::Tensor gscale, double gsign) {
+  const int M = (int)hidden.size(0);
+  const int E = (int)hidden.size(1);
+  const int64_t V = w.size(0);
+  TORCH_CHECK(V < (int64_t)INT32_MAX - 64, "catalog must fit int32");
+  TORCH_CHECK(E == 64 || E == 128, "ce_linear_wgrad supports E in {64, 128}");
+  const int item_tiles = (int)((V + 63) / 64);
+  const int stripes = std::max(1, std::min(16, 4096 / std::max(item_tiles, 1)));
+  auto slabs = torch::empty({stripes, V, (int64_t)E}, hidden.options().dtype(torch::kFloat32));
+  auto labels_c = labels.contiguous();
+  auto stream = at::cuda::getCurrentHIPStream();
+  const size_t lds = (size_t)E * 64 * 2 + 64 * 12 + 4 * 16 * 64 * 2 + 64;
+#define LAUNCH_WG(EE)                                                                        hipLaunchKernelGGL((ce_linear_wgrad_kernel<EE>), dim3(item_tiles, stripes), dim3(256),                        lds, stream,                                                                               reinterpret_cast<const __hip_bfloat16*>(hidden.data_ptr()),                                reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),                                     labels_c.data_ptr<int64_t>(), lse.data_ptr<float>(),                                       gscale.data_ptr<float>(), (float)gsign, slabs.data_ptr<float>(),                           M, (int)V)
+  if (E == 64) {
+    LAUNCH_WG(64);
+  } else {
+    LAUNCH_WG(128);
+  }
+#undef LAUNCH_WG
+  return slabs.sum(0);
+}
+
+torch::Tensor ce_linear_bwd_fused_dh(torch::Tensor hidden, torch::Tensor w,
+                                     torch::Tensor labels, torch::Tensor lse,
+                                     torch::Tensor gscale, double gsign) {
+  // dhidden ONLY: the bwd kernel with FUSE_DH and the dlogits stores compiled
+  // out (pairs with ce_linear_wgrad, which supplies dW)
+  const int M = (int)hidden.size(0);
+  const int E = (int)hidden.size(1);
+  const int64_t V = w.size(0);
+  TORCH_CHECK(V < (int64_t)INT32_MAX - 64, "catalog must fit int32");
+  TORCH_CHECK(E == 64 || E == 128, "fused-dh bwd supports E in {64, 128}");
+  auto dhidden = torch::empty_like(hidden);
+  auto labels_c = labels.contiguous();
+  const int m_tiles = (M + 255) / 256;
+  auto stream = at::cuda::getCurrentHIPStream();
+#define LAUNCH_BF(EE)                                                                         do {                                                                                          const size_t lds = (size_t)4 * EE * 64 * 2 + 4 * 16 * 64 * 2 + 3 * 256 * 4 + 64;            hipLaunchKernelGGL((ce_linear_bwd_kernel<EE, true, true, false>), dim3(m_tiles),                               dim3(256), lds, stream,                                                                     reinterpret_cast<const __hip_bfloat16*>(hidden.data_ptr()),                                 reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),                                      labels_c.data_ptr<int64_t>(), lse.data_ptr<float>(),                                        gscale.data_ptr<float>(), (float)gsign, nullptr,                                            reinterpret_cast<__hip_bfloat16*>(dhidden.data_ptr()), M, V, 0);       } while (0)
+  if (E == 64) {
+    LAUNCH_BF(64);
+  } else {
+    LAUNCH_BF(128);
+  }
+#undef LAUNCH_BF
+  return dhidden;
+}
 
 std::vector<torch::Tensor> ce_linear_fwd(torch::Tensor hidden, torch::Tensor w,
                                          torch::Tensor labels) {

@@ -38,6 +38,11 @@ std::vector<torch::Tensor> ce_linear_fwd(torch::Tensor hidden, torch::Tensor w,
 std::vector<torch::Tensor> ce_linear_bwd(torch::Tensor hidden, torch::Tensor w,
                                          torch::Tensor labels, torch::Tensor lse,
                                          torch::Tensor gscale, double gsign);
+torch::Tensor ce_linear_wgrad(torch::Tensor hidden, torch::Tensor w, torch::Tensor labels,
+                              torch::Tensor lse, torch::Tensor gscale, double gsign);
+torch::Tensor ce_linear_bwd_fused_dh(torch::Tensor hidden, torch::Tensor w,
+                                     torch::Tensor labels, torch::Tensor lse,
+                                     torch::Tensor gscale, double gsign);
 torch::Tensor metrics_reduce(torch::Tensor preds, torch::Tensor gt,
                              c10::optional<torch::Tensor> train, torch::Tensor ks);
 std::vector<torch::Tensor> scored_topk_gemm_fp8(torch::Tensor q, torch::Tensor w,
@@ -59,6 +64,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused linear+CE forward: online LSE in the GEMM epilogue (gfx950)");
   m.def("ce_linear_bwd", &ce_linear_bwd,
         "fused linear+CE backward: recomputed dlogits + fused dhidden (gfx950)");
+  m.def("ce_linear_wgrad", &ce_linear_wgrad,
+        "item-owner dW accumulation without materializing dlogits (gfx950)");
+  m.def("ce_linear_bwd_fused_dh", &ce_linear_bwd_fused_dh,
+        "dhidden-only backward (no dlogits stores; pairs with ce_linear_wgrad)");
   m.def("scored_topk_gemm_fp8", &scored_topk_gemm_fp8,
         "fused e4m3 MFMA score-GEMM + top-k candidate selection (gfx950)");
   m.def("metrics_reduce", &metrics_reduce,
