@@ -520,8 +520,8 @@ __global__ __launch_bounds__(256, 2) void ce_linear_bwd_kernel(
 // (item-tile, row-stripe) into a per-stripe fp32 slab; the host sums the
 // S slabs.  Replaces the 2x[M, Vp] bf16 dlogits round trip (22 GB at the
 // flagship shape) + the hipBLASLt wgrad GEMM.
-template <int E>
-__global__ __launch_bounds__(256, 4) void ce_linear_wgrad_kernel(
+template <int E, int MI>  // MI = item fragments per wave (WG owns MI*64 items)
+__global__ __launch_bounds__(256, 2) void ce_linear_wgrad_kernel(
     const __hip_bfloat16* __restrict__ hidden,  // [M, E]
     const __hip_bfloat16* __restrict__ w,       // [V, E]
     const int64_t* __restrict__ labels,         // [M]
@@ -532,14 +532,16 @@ __global__ __launch_bounds__(256, 4) void ce_linear_wgrad_kernel(
     int M, int V) {
   constexpr int KSTEPS = E / 32;  // k over E (S^T GEMM)
   constexpr int OF = E / 16;      // dW column fragments
+  constexpr int HN_B = E * 2 + 16;  // padded row-major hidden stage stride
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
-  const int item0 = blockIdx.x * 64;
+  const int item0 = blockIdx.x * (64 * MI) + wave * (16 * MI);
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // [hidden^T stage: E x 64 bf16 swizzled][adj/g: 64 f32 each][lab: 64 i32]
-  // [4 waves x dS bounce 16 x 64 bf16]
-  __hip_bfloat16* ht = reinterpret_cast<__hip_bfloat16*>(smem);
+  // [hn: 64 x HN_B row-major hidden][ht: E x 64 transposed swizzled]
+  // [adj/g: 64 f32][lab: 64 i32][4 waves x dS bounce 16 x 64 bf16]
+  char* hn = smem;
+  __hip_bfloat16* ht = reinterpret_cast<__hip_bfloat16*>(hn + (size_t)64 * HN_B);
   float* adj_l = reinterpret_cast<float*>(ht + (size_t)E * 64);
   float* g_l = adj_l + 64;
   int* lab_l = reinterpret_cast<int*>(g_l + 64);
@@ -548,30 +550,36 @@ __global__ __launch_bounds__(256, 4) void ce_linear_wgrad_kernel(
   auto ht_off = [&](int e, int row) { return e * 64 + (row ^ ((e & 7) << 3)); };
   auto ds_off = [&](int itm, int row) { return itm * 64 + (row ^ ((itm & 7) << 3)); };
 
-  // resident A: this wave's 16 item rows of W
-  bf16x8 a_w[KSTEPS];
-  {
-    const int item = item0 + wave * 16 + (lane & 15);
+  // resident A: this wave's MI x 16 item rows of W
+  bf16x8 a_w[MI][KSTEPS];
+#pragma unroll
+  for (int mi = 0; mi < MI; ++mi) {
+    const int item = item0 + mi * 16 + (lane & 15);
     const __hip_bfloat16* wr = w + (size_t)min(item, V - 1) * E + (lane >> 4) * 8;
 #pragma unroll
-    for (int ks = 0; ks < KSTEPS; ++ks) a_w[ks] = *reinterpret_cast<const bf16x8*>(wr + ks * 32);
+    for (int ks = 0; ks < KSTEPS; ++ks) a_w[mi][ks] = *reinterpret_cast<const bf16x8*>(wr + ks * 32);
     if (item >= V) {
 #pragma unroll
-      for (int ks = 0; ks < KSTEPS; ++ks) a_w[ks] = bf16x8{0};
+      for (int ks = 0; ks < KSTEPS; ++ks) a_w[mi][ks] = bf16x8{0};
     }
   }
-  f32x4 dwacc[OF];
+  f32x4 dwacc[MI][OF];
 #pragma unroll
-  for (int f = 0; f < OF; ++f) dwacc[f] = f32x4{0.f, 0.f, 0.f, 0.f};
+  for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+    for (int f = 0; f < OF; ++f) dwacc[mi][f] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   for (int r0 = blockIdx.y * 64; r0 < M; r0 += gridDim.y * 64) {
-    // ---- stage hidden^T [E][64] + per-row state ----
+    // ---- stage hidden ONCE per (WG, row-tile): row-major hn (direct b128
+    // writes, feeds the S^T B fragments) + transposed ht (feeds the dW B
+    // fragments) ----
     for (int i = threadIdx.x; i < 64 * (E / 8); i += blockDim.x) {
       const int row = i / (E / 8);
       const int e0 = (i % (E / 8)) * 8;
       const int src = min(r0 + row, M - 1);
       bf16x8 vv = *reinterpret_cast<const bf16x8*>(hidden + (size_t)src * E + e0);
       if (r0 + row >= M) vv = bf16x8{0};
+      *reinterpret_cast<bf16x8*>(hn + (size_t)row * HN_B + e0 * 2) = vv;
 #pragma unroll
       for (int j = 0; j < 8; ++j) ht[ht_off(e0 + j, row)] = ((const __hip_bfloat16*)&vv)[j];
     }
@@ -584,59 +592,63 @@ __global__ __launch_bounds__(256, 4) void ce_linear_wgrad_kernel(
     }
     __syncthreads();
 
-    // ---- S^T = W . hidden^T (B fragments straight from global hidden) ----
-    f32x4 sacc[4];
 #pragma unroll
-    for (int f = 0; f < 4; ++f) sacc[f] = f32x4{0.f, 0.f, 0.f, 0.f};
+    for (int mi = 0; mi < MI; ++mi) {
+      // ---- S^T = W . hidden^T: B fragments from the row-major LDS stage ----
+      f32x4 sacc[4];
 #pragma unroll
-    for (int ks = 0; ks < KSTEPS; ++ks) {
+      for (int f = 0; f < 4; ++f) sacc[f] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks) {
+#pragma unroll
+        for (int f = 0; f < 4; ++f) {
+          bf16x8 b_h = *reinterpret_cast<const bf16x8*>(
+              hn + (size_t)(f * 16 + (lane & 15)) * HN_B + (ks * 32 + (lane >> 4) * 8) * 2);
+          sacc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_w[mi][ks], b_h, sacc[f], 0, 0, 0);
+        }
+      }
+      // ---- dS^T -> wave-private bounce tile ----
 #pragma unroll
       for (int f = 0; f < 4; ++f) {
-        const int row = min(r0 + f * 16 + (lane & 15), M - 1);
-        bf16x8 b_h = *reinterpret_cast<const bf16x8*>(
-            hidden + (size_t)row * E + ks * 32 + (lane >> 4) * 8);
-        sacc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_w[ks], b_h, sacc[f], 0, 0, 0);
-      }
-    }
-    // ---- dS^T on the accumulators -> wave-private bounce tile ----
 #pragma unroll
-    for (int f = 0; f < 4; ++f) {
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int lrow = f * 16 + (lane & 15);
-        const int item = item0 + wave * 16 + (lane >> 4) * 4 + r;
-        float dl = 0.f;
-        if (item < V) {
-          dl = __expf(sacc[f][r] - adj_l[lrow]);
-          if (item == lab_l[lrow]) dl -= g_l[lrow];
+        for (int r = 0; r < 4; ++r) {
+          const int lrow = f * 16 + (lane & 15);
+          const int item = item0 + mi * 16 + (lane >> 4) * 4 + r;
+          float dl = 0.f;
+          if (item < V) {
+            dl = __expf(sacc[f][r] - adj_l[lrow]);
+            if (item == lab_l[lrow]) dl -= g_l[lrow];
+          }
+          my_ds[ds_off((lane >> 4) * 4 + r, lrow)] = __float2bfloat16(dl * gsign);
         }
-        my_ds[ds_off((lane >> 4) * 4 + r, lrow)] = __float2bfloat16(dl * gsign);
+      }
+      // ---- dW[mi] += dS^T . hidden (A from bounce, B from ht) ----
+#pragma unroll
+      for (int ks2 = 0; ks2 < 2; ++ks2) {
+        bf16x8 a_ds = *reinterpret_cast<const bf16x8*>(
+            my_ds + ds_off(lane & 15, ks2 * 32 + (lane >> 4) * 8));
+#pragma unroll
+        for (int f = 0; f < OF; ++f) {
+          bf16x8 b_ht = *reinterpret_cast<const bf16x8*>(
+              ht + ht_off(f * 16 + (lane & 15), ks2 * 32 + (lane >> 4) * 8));
+          dwacc[mi][f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_ds, b_ht, dwacc[mi][f], 0, 0, 0);
+        }
       }
     }
-    // ---- dW += dS^T . hidden (A from bounce, B from the hidden^T stage) ----
-#pragma unroll
-    for (int ks2 = 0; ks2 < 2; ++ks2) {
-      bf16x8 a_ds = *reinterpret_cast<const bf16x8*>(
-          my_ds + ds_off(lane & 15, ks2 * 32 + (lane >> 4) * 8));
-#pragma unroll
-      for (int f = 0; f < OF; ++f) {
-        bf16x8 b_ht = *reinterpret_cast<const bf16x8*>(
-            ht + ht_off(f * 16 + (lane & 15), ks2 * 32 + (lane >> 4) * 8));
-        dwacc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_ds, b_ht, dwacc[f], 0, 0, 0);
-      }
-    }
-    __syncthreads();  // everyone done with ht before the next stage
+    __syncthreads();  // everyone done with the stages before the next tile
   }
   // ---- store this (item-tile, stripe)'s dW partial ----
   float* slab = dw_slabs + (size_t)blockIdx.y * V * E;
 #pragma unroll
-  for (int f = 0; f < OF; ++f) {
+  for (int mi = 0; mi < MI; ++mi)
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int item = item0 + wave * 16 + (lane >> 4) * 4 + r;
-      if (item < V) slab[(size_t)item * E + f * 16 + (lane & 15)] = dwacc[f][r];
+    for (int f = 0; f < OF; ++f) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int item = item0 + mi * 16 + (lane >> 4) * 4 + r;
+        if (item < V) slab[(size_t)item * E + f * 16 + (lane & 15)] = dwacc[mi][f][r];
+      }
     }
-  }
 }
 
 }  // namespace
@@ -648,17 +660,18 @@ torch::Tensor ce_linear_wgrad(torch::Tensor hidden, torch::Tensor w, torch::Tens
   const int64_t V = w.size(0);
   TORCH_CHECK(V < (int64_t)INT32_MAX - 64, "catalog must fit int32");
   TORCH_CHECK(E == 64 || E == 128, "ce_linear_wgrad supports E in {64, 128}");
-  const int item_tiles = (int)((V + 63) / 64);
+  const int items_per_wg = (E == 64) ? 256 : 128;
+  const int item_tiles = (int)((V + items_per_wg - 1) / items_per_wg);
   const int stripes = std::max(1, std::min(16, 4096 / std::max(item_tiles, 1)));
   auto slabs = torch::empty({stripes, V, (int64_t)E}, hidden.options().dtype(torch::kFloat32));
   auto labels_c = labels.contiguous();
   auto stream = at::cuda::getCurrentHIPStream();
-  const size_t lds = (size_t)E * 64 * 2 + 64 * 12 + 4 * 16 * 64 * 2 + 64;
-#define LAUNCH_WG(EE)                                                                        hipLaunchKernelGGL((ce_linear_wgrad_kernel<EE>), dim3(item_tiles, stripes), dim3(256),                        lds, stream,                                                                               reinterpret_cast<const __hip_bfloat16*>(hidden.data_ptr()),                                reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),                                     labels_c.data_ptr<int64_t>(), lse.data_ptr<float>(),                                       gscale.data_ptr<float>(), (float)gsign, slabs.data_ptr<float>(),                           M, (int)V)
+  const size_t lds = (size_t)64 * (E * 2 + 16) + (size_t)E * 64 * 2 + 64 * 12 + 4 * 16 * 64 * 2 + 64;
+#define LAUNCH_WG(EE, MII) hipLaunchKernelGGL((ce_linear_wgrad_kernel<EE, MII>), dim3(item_tiles, stripes), dim3(256), lds, stream, reinterpret_cast<const __hip_bfloat16*>(hidden.data_ptr()), reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()), labels_c.data_ptr<int64_t>(), lse.data_ptr<float>(), gscale.data_ptr<float>(), (float)gsign, slabs.data_ptr<float>(), M, (int)V)
   if (E == 64) {
-    LAUNCH_WG(64);
+    LAUNCH_WG(64, 4);
   } else {
-    LAUNCH_WG(128);
+    LAUNCH_WG(128, 2);
   }
 #undef LAUNCH_WG
   return slabs.sum(0);
