@@ -104,11 +104,16 @@ class FusedLinearCrossEntropyFunction(torch.autograd.Function):
         if (
             hidden2d.shape[1] <= 128
             and hasattr(ext, "ce_linear_wgrad")
-            and os.environ.get("REPLAY_AMD_CE_WGRAD") != "0"
+            and os.environ.get("REPLAY_AMD_CE_WGRAD") == "1"
         ):
-            # phase-split backward: dW from the item-owner wgrad kernel and
-            # dhidden from the store-free fused-dh pass — the [M, Vp] bf16
-            # dlogits tensor (22 GB at the flagship shape) never exists
+            # OPT-IN phase-split backward (REPLAY_AMD_CE_WGRAD=1): dW from the
+            # item-owner wgrad kernel and dhidden from the store-free fused-dh
+            # pass — the [M, Vp] bf16 dlogits tensor never exists, but both
+            # kernels are LDS-transpose-stage bound and MEASURED SLOWER than
+            # the shipped dlogits+hipBLASLt path (11.6 + 11.5 ms vs ~12 ms
+            # total at the flagship shape, i.e. the memory saving costs ~7 ms
+            # of step time).  Kept for the T10 (ds_read_b64_tr_b16 hardware
+            # transpose) rework; parity-tested either way.
             dweight = ext.ce_linear_wgrad(hidden2d, weight, labels, lse, gscale, gsign)
             dhidden = ext.ce_linear_bwd_fused_dh(hidden2d, weight, labels, lse, gscale, gsign)
             return dhidden, dweight.to(weight.dtype), None, None
