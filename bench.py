@@ -510,12 +510,12 @@ def main() -> None:
         return
     if args.mode == "serve":
         serve_bench(args, device, rank, world)
-        if world > 1:
+        if torch.distributed.is_initialized():
             torch.distributed.destroy_process_group()
         return
     if args.mode == "twotower":
         twotower_bench(args, device, rank, world)
-        if world > 1:
+        if torch.distributed.is_initialized():
             torch.distributed.destroy_process_group()
         return
 
