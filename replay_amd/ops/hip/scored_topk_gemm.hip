@@ -603,7 +603,7 @@ __device__ __forceinline__ void waitcnt_vm() {
 
 // ABLATE: 0 full; 1 no-stage; 2 no-mfma; 3 no-epilogue; 4 no-epilogue +
 // no-barrier/vmcnt (pure ds_read+MFMA loop); 5 no-epilogue + no-stage
-template <int E, int MF, int NBUF = 3, int ABLATE = 0, int LOADERS = 0, int PAIR = 0>
+template <int E, int MF, int NBUF = 3, int ABLATE = 0, int LOADERS = 0>
 __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
     const __hip_bfloat16* __restrict__ q,  // [M, E]
     const __hip_bfloat16* __restrict__ w,  // [V, E]
@@ -683,8 +683,7 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
   }
 
   int cur = 0;
-  const int loop_step = (PAIR ? 2 : 1) * tile_stride;
-  for (int tile = tile0; tile < n_tiles; tile += loop_step) {
+  for (int tile = tile0; tile < n_tiles; tile += tile_stride) {
     // own glds for buf[cur] complete; allow NBUF-2 newer tiles in flight
     // (when they were actually staged — at the walk's tail, full drain)
     if constexpr (ABLATE != 4) {
@@ -706,11 +705,8 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
       __builtin_amdgcn_s_barrier();
     }
 
-    for (int sub = 0; sub <= (PAIR ? 1 : 0); ++sub) {
-    const int tile_s = PAIR ? (tile + sub * tile_stride) : tile;
-    if (PAIR && tile_s >= n_tiles) break;
-    const char* bbuf = smem + (size_t)((cur + sub) % NBUF) * TILE_B;
-    const int n0 = tile_s << 6;
+    const char* bbuf = smem + (size_t)cur * TILE_B;
+    const int n0 = tile << 6;
     // one 16-item column fragment at a time: acc live set = MF quads (not
     // MF x 4), which is what lets MF = 4 (512-row M-tile) fit in 256 VGPRs.
     // B fragments are prefetched a full column ahead (ping-pong bfr[2][8])
@@ -781,20 +777,11 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
     // stage tile+2 LAST so the next iteration's counted vmcnt drains this
     // iteration's (older) epilogue stores together with tile+1's glds while
     // leaving only the newest stage in flight
-    }  // sub (pair)
     if constexpr (ABLATE != 1 && ABLATE != 5) {
-      if constexpr (PAIR != 0) {
-#pragma unroll
-        for (int sub2 = 0; sub2 < 2; ++sub2) {
-          const int t2 = tile + (2 + sub2) * tile_stride;
-          if (t2 < n_tiles) stage_tile((cur + 2 + sub2) % NBUF, t2);
-        }
-      } else {
-        const int t2 = tile + (NBUF - 1) * tile_stride;
-        if (t2 < n_tiles) stage_tile((cur + NBUF - 1) % NBUF, t2);
-      }
+      const int t2 = tile + (NBUF - 1) * tile_stride;
+      if (t2 < n_tiles) stage_tile((cur + NBUF - 1) % NBUF, t2);
     }
-    cur = (cur + (PAIR ? 2 : 1)) % NBUF;
+    cur = (cur + 1) % NBUF;
   }
 }
 
@@ -1263,16 +1250,6 @@ std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
       const size_t lds_p = 3 * 64 * (256 * 2 + 32);
       hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 3, 0, 2>), grid, dim3(512), lds_p,
                          stream, reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
-                         reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
-                         thr.data_ptr<float>(), out_vals.data_ptr<float>(),
-                         out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
-                         (int)capacity);
-    } else if (variant != nullptr && variant[0] == 'q') {
-      // tile-PAIR processing: one barrier per two tiles (NBUF=4)
-      const size_t lds_q = 4 * 64 * (256 * 2 + 32);
-      hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 4, 0, 0, 1>), grid, dim3(512),
-                         lds_q, stream,
-                         reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
                          reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
                          thr.data_ptr<float>(), out_vals.data_ptr<float>(),
                          out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
