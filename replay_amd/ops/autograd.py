@@ -82,7 +82,14 @@ class FusedLinearCrossEntropyFunction(torch.autograd.Function):
         ext = hip_ext()
         hidden2d = hidden2d.contiguous()
         weight = weight.contiguous()
-        lse, lab_logit = ext.ce_linear_fwd(hidden2d, weight, labels)
+        if hasattr(ext, "ce_linear_lse"):
+            # LSE-only kernel (the in-kernel per-logit label compare was ~2%
+            # VALU issue); the label logit is a gather + row dot here
+            lse = ext.ce_linear_lse(hidden2d, weight)
+            lab_w = weight.index_select(0, labels.clamp(min=0))
+            lab_logit = (hidden2d.float() * lab_w.float()).sum(-1)
+        else:
+            lse, lab_logit = ext.ce_linear_fwd(hidden2d, weight, labels)
         valid = labels != ignore_index
         count = valid.sum()
         loss = torch.where(valid, lse - lab_logit, torch.zeros_like(lse)).sum()

@@ -59,7 +59,7 @@ __device__ __forceinline__ void lse_combine16(float& m, float& s) {
 // ---------------------------------------------------------------------------
 // forward: per-row logsumexp + label logit, no logits materialization
 // ---------------------------------------------------------------------------
-template <int E, bool RESIDENT>
+template <int E, bool RESIDENT, bool CAPTURE_LAB = true>
 __global__ __launch_bounds__(256, 2) void ce_linear_fwd_kernel(
     const __hip_bfloat16* __restrict__ hidden,  // [M, E]
     const __hip_bfloat16* __restrict__ w,       // [V, E]
@@ -200,19 +200,34 @@ __global__ __launch_bounds__(256, 2) void ce_linear_fwd_kernel(
     // epilogue: online LSE, branchless, one rescale per 4 values.  The VALU
     // exp throughput (1/4 rate) is this kernel's floor — 5 exps per 4 logits
     // beats the per-value online update (2 exps + divergent rescale each).
-    // OOB tail columns become -inf and exp to exactly 0.
+    // The OOB select and the label compare run only on the (single) tail
+    // tile / in CAPTURE_LAB builds: the fwd is VALU-ISSUE bound (PMC: 48%
+    // of wave cycles are VALU issue), so every epilogue op is ~2% of the
+    // kernel.
+    const bool tail_tile = (n0 + 64 > Vi);
 #pragma unroll
     for (int mf = 0; mf < MF; ++mf) {
       const int lrow0 = wave * 64 + mf * 16 + (lane >> 4) * 4;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int lab_r = lab_lds[lrow0 + r];
         float v4[4];
+        if (tail_tile) {
 #pragma unroll
-        for (int f = 0; f < 4; ++f) {
-          const int item = n0 + f * 16 + (lane & 15);
-          v4[f] = (item < Vi) ? acc[mf][f][r] : -INFINITY;
-          if (item == lab_r && item < Vi) rlab_lds[lrow0 + r] = v4[f];
+          for (int f = 0; f < 4; ++f) {
+            const int item = n0 + f * 16 + (lane & 15);
+            v4[f] = (item < Vi) ? acc[mf][f][r] : -INFINITY;
+          }
+        } else {
+#pragma unroll
+          for (int f = 0; f < 4; ++f) v4[f] = acc[mf][f][r];
+        }
+        if constexpr (CAPTURE_LAB) {
+          const int lab_r = lab_lds[lrow0 + r];
+#pragma unroll
+          for (int f = 0; f < 4; ++f) {
+            const int item = n0 + f * 16 + (lane & 15);
+            if (item == lab_r && item < Vi) rlab_lds[lrow0 + r] = v4[f];
+          }
         }
         const float m4 = fmaxf(fmaxf(v4[0], v4[1]), fmaxf(v4[2], v4[3]));
         const float nm = fmaxf(r_max[mf][r], m4);
@@ -699,6 +714,36 @@ torch::Tensor ce_linear_bwd_fused_dh(torch::Tensor hidden, torch::Tensor w,
   }
 #undef LAUNCH_BF
   return dhidden;
+}
+
+torch::Tensor ce_linear_lse(torch::Tensor hidden, torch::Tensor w) {
+  // LSE-only forward (CAPTURE_LAB compiled out): the label logit is two
+  // cheap eager ops on the host, and the in-kernel compare/select/LDS
+  // capture per logit goes away
+  TORCH_CHECK(hidden.is_cuda() && hidden.dim() == 2 && hidden.is_contiguous());
+  TORCH_CHECK(w.is_cuda() && w.dim() == 2 && w.is_contiguous());
+  const int M = (int)hidden.size(0);
+  const int E = (int)hidden.size(1);
+  const int64_t V = w.size(0);
+  TORCH_CHECK(V < (int64_t)INT32_MAX - 64, "catalog must fit int32");
+  auto opts_f = hidden.options().dtype(torch::kFloat32);
+  auto lse = torch::empty({M}, opts_f);
+  auto lab_logit = torch::empty({0}, opts_f);
+  auto labels = torch::zeros({M}, hidden.options().dtype(torch::kInt64));
+  const int m_tiles = (M + 255) / 256;
+  auto stream = at::cuda::getCurrentHIPStream();
+#define LAUNCH_CLL(EE) hipLaunchKernelGGL((ce_linear_fwd_kernel<EE, (EE <= 128), false>), dim3(m_tiles), dim3(256), 2048 + ((EE <= 128) ? 0 : (size_t)256 * EE * 2), stream, reinterpret_cast<const __hip_bfloat16*>(hidden.data_ptr()), reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()), labels.data_ptr<int64_t>(), lse.data_ptr<float>(), lse.data_ptr<float>(), M, V)
+  if (E == 64) {
+    LAUNCH_CLL(64);
+  } else if (E == 128) {
+    LAUNCH_CLL(128);
+  } else if (E == 256) {
+    LAUNCH_CLL(256);
+  } else {
+    TORCH_CHECK(false, "ce_linear supports E in {64, 128, 256}");
+  }
+#undef LAUNCH_CLL
+  return lse;
 }
 
 std::vector<torch::Tensor> ce_linear_fwd(torch::Tensor hidden, torch::Tensor w,

@@ -35,6 +35,7 @@ std::vector<torch::Tensor> attention_bwd_mfma(torch::Tensor q, torch::Tensor k, 
                                               bool causal);
 std::vector<torch::Tensor> ce_linear_fwd(torch::Tensor hidden, torch::Tensor w,
                                          torch::Tensor labels);
+torch::Tensor ce_linear_lse(torch::Tensor hidden, torch::Tensor w);
 std::vector<torch::Tensor> ce_linear_bwd(torch::Tensor hidden, torch::Tensor w,
                                          torch::Tensor labels, torch::Tensor lse,
                                          torch::Tensor gscale, double gsign);
@@ -60,6 +61,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused MFMA score-GEMM + top-k candidate selection (gfx950)");
   m.def("attention_fwd_mfma", &attention_fwd_mfma, "MFMA attention forward (gfx950)");
   m.def("attention_bwd_mfma", &attention_bwd_mfma, "MFMA attention backward (gfx950)");
+  m.def("ce_linear_lse", &ce_linear_lse,
+        "LSE-only fused linear forward (no label capture) (gfx950)");
   m.def("ce_linear_fwd", &ce_linear_fwd,
         "fused linear+CE forward: online LSE in the GEMM epilogue (gfx950)");
   m.def("ce_linear_bwd", &ce_linear_bwd,

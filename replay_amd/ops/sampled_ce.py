@@ -38,8 +38,11 @@ class NegPoolLSE(torch.autograd.Function):
         ext = hip_ext()
         h_b = h2d.to(torch.bfloat16).contiguous()
         w_b = wneg.to(torch.bfloat16).contiguous()
-        zeros = torch.zeros(h_b.shape[0], dtype=torch.long, device=h_b.device)
-        lse, _ = ext.ce_linear_fwd(h_b, w_b, zeros)
+        if hasattr(ext, "ce_linear_lse"):
+            lse = ext.ce_linear_lse(h_b, w_b)
+        else:
+            zeros = torch.zeros(h_b.shape[0], dtype=torch.long, device=h_b.device)
+            lse, _ = ext.ce_linear_fwd(h_b, w_b, zeros)
         ctx.save_for_backward(h_b, w_b, lse)
         ctx.h_dtype = h2d.dtype
         ctx.w_dtype = wneg.dtype
@@ -76,7 +79,7 @@ def can_fuse_sampled_ce(embeddings: torch.Tensor, negative_labels: torch.Tensor,
     return (
         embeddings.is_cuda
         and ext is not None
-        and hasattr(ext, "ce_linear_fwd")
+        and (hasattr(ext, "ce_linear_lse") or hasattr(ext, "ce_linear_fwd"))
         and negative_labels.dim() == 1
         and negative_labels.shape[0] >= 512
         and embeddings.shape[-1] in (64, 128, 256)
