@@ -246,9 +246,11 @@ __global__ __launch_bounds__(256, 2) void ce_linear_fwd_kernel(
       lse_combine16(r_max[mf][r], r_sum[mf][r]);
       const int row = m0 + mf * 16 + (lane >> 4) * 4 + r;
       if ((lane & 15) == 0 && row < M) {
-        const float rl = rlab_lds[wave * 64 + mf * 16 + (lane >> 4) * 4 + r];
         lse_out[row] = r_max[mf][r] + __logf(r_sum[mf][r]);
-        lab_out[row] = (rl == -INFINITY) ? 0.f : rl;
+        if constexpr (CAPTURE_LAB) {
+          const float rl = rlab_lds[wave * 64 + mf * 16 + (lane >> 4) * 4 + r];
+          lab_out[row] = (rl == -INFINITY) ? 0.f : rl;
+        }
       }
     }
   }
