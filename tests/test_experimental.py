@@ -247,3 +247,39 @@ def test_two_stages_deepened(ds):
         assert recs.groupby("query_id").size().max() <= 3
     # embedding features reached the ranker
     assert any(c.startswith("m0_fm") for c in sc._feature_cols)
+
+
+def test_rl_mdp_builder_and_save_load(ds, tmp_path):
+    """Episode construction (top-K reward shaping, terminals) + policy
+    save/load for the offline-RL models."""
+    import numpy as np
+
+    from replay_amd.experimental.models import CQL, DDPG
+    from replay_amd.experimental.models.rl import MdpDatasetBuilder
+
+    inter = ds.interactions
+    mdp = MdpDatasetBuilder(top_k=2).build(inter, "query_id", "item_id", "rating", "timestamp", seed=0)
+    n_users = inter["query_id"].nunique()
+    assert mdp["terminals"].sum() == n_users  # one terminal per user episode
+    per_user_rewards = {}
+    for u, r in zip(mdp["users"], mdp["rewards"]):
+        per_user_rewards.setdefault(u, 0)
+        per_user_rewards[u] += r
+    assert all(v <= 2 for v in per_user_rewards.values())  # top-K shaping
+
+    cql = CQL(epochs=2, embedding_dim=8, hidden_dim=8, device="cpu", seed=0)
+    cql.fit(ds)
+    recs1 = cql.predict(ds, k=3)
+    cql._save_model(str(tmp_path / "cql.pt"))
+    cql2 = CQL(embedding_dim=8, hidden_dim=8, device="cpu")
+    for attr in ("fit_queries", "fit_items", "_query_dim_size", "_item_dim_size",
+                 "query_column", "item_column", "rating_column", "timestamp_column"):
+        setattr(cql2, attr, getattr(cql, attr))
+    cql2._load_model(str(tmp_path / "cql.pt"))
+    recs2 = cql2.predict(ds, k=3)
+    assert (recs1["item_id"].to_numpy() == recs2["item_id"].to_numpy()).all()
+
+    ddpg = DDPG(epochs=2, embedding_dim=8, hidden_dim=8, device="cpu", seed=0)
+    ddpg.fit(ds)
+    ddpg._save_model(str(tmp_path / "ddpg.pt"))
+    assert len(ddpg.predict(ds, k=3)) > 0
