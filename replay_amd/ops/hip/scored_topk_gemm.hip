@@ -603,7 +603,8 @@ __device__ __forceinline__ void waitcnt_vm() {
 
 // ABLATE: 0 full; 1 no-stage; 2 no-mfma; 3 no-epilogue; 4 no-epilogue +
 // no-barrier/vmcnt (pure ds_read+MFMA loop); 5 no-epilogue + no-stage
-template <int E, int MF, int NBUF = 3, int ABLATE = 0, int LOADERS = 0>
+template <int E, int MF, int NBUF = 3, int ABLATE = 0, int LOADERS = 0, int XCD_MAP = 1,
+          int SPREAD = 1, int LDSEPI = 1>
 __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
     const __hip_bfloat16* __restrict__ q,  // [M, E]
     const __hip_bfloat16* __restrict__ w,  // [V, E]
@@ -617,11 +618,44 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
   constexpr int CHUNKS_ROW = ROW_B / 16;        // 34
   constexpr int TILE_PIECES = 64 * CHUNKS_ROW / 64;  // 34 wave-pieces
   constexpr int TILE_B = 64 * ROW_B;
+  constexpr size_t TILE_W = (size_t)64 * E * 2;  // W bytes per tile (unpadded)
+  // LDS hit-append epilogue (LDSEPI): per-wave per-row candidate lists.
+  // Hits are appended with a ds_add_rtn + ds_write (~100 cyc, no vmem)
+  // instead of a global atomicAdd whose result feeds the store (~1 L2
+  // round trip serialized under the per-tile barrier — PMC showed 53% of
+  // wave cycles parked with the global path).  The lists flush to the
+  // global candidate buffer ONCE per block walk, one lane-parallel
+  // atomicAdd per row.  RCAP=8 covers the expected ~1 hit/row/walk at
+  // serving thresholds; overflow falls back to the direct global append
+  // (correct, just slower — only matters when few stripes make walks
+  // long, e.g. very large M).
+  constexpr int RCAP = 8;
+  constexpr int ROWS_W = 16 * MF;  // rows owned by one wave
+  constexpr size_t EPI_PER_WAVE = (size_t)ROWS_W * 4 + (size_t)ROWS_W * RCAP * 8;
   const int V = (int)V64;  // host asserts V < 2^31 (out_idx is int32 anyway);
                            // 32-bit index math saves ~30 VGPRs at MF=4
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
-  const int m0 = blockIdx.x * (128 * MF) + wave * (16 * MF);
+  // XCD-aware (m_tile, stripe) assignment.  Workgroups dispatch round-robin
+  // over the 8 XCDs, so the naive (x = m_tile, y = stripe) grid puts the
+  // m_tiles that share a B stripe on DIFFERENT XCDs: every XCD streams the
+  // whole item table through its own (private) L2 and the kernel is
+  // HBM-bound on the 8x re-read (measured 5.6 ms at B=1024/V=10M — the
+  // 40 GB the 8 XCDs pull at ~7 TB/s).  Remap (bijective, guide §"XCD
+  // swizzle") so each XCD owns a contiguous band of stripes together with
+  // ALL their m_tiles: each B tile is pulled from HBM by exactly one XCD
+  // and the co-resident m_tile blocks hit it in that XCD's L2.
+  int bx = blockIdx.x, by = blockIdx.y;
+  if constexpr (XCD_MAP) {
+    const int nwg = (int)(gridDim.x * gridDim.y);
+    const int orig = (int)(blockIdx.y * gridDim.x + blockIdx.x);
+    const int qq = nwg >> 3, rr = nwg & 7, xcd = orig & 7, slot = orig >> 3;
+    const int rid =
+        (xcd < rr ? xcd * (qq + 1) : rr * (qq + 1) + (xcd - rr) * qq) + slot;
+    bx = rid % (int)gridDim.x;  // m_tile: fastest within an XCD's band
+    by = rid / (int)gridDim.x;  // stripe
+  }
+  const int m0 = bx * (128 * MF) + wave * (16 * MF);
 
   extern __shared__ __attribute__((aligned(16))) char smem[];  // 3 x TILE_B
 
@@ -639,43 +673,98 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
       for (int ks = 0; ks < KSTEPS; ++ks) a_frag[mf][ks] = bf16x8{0};
     }
   }
-  float t_reg[MF][4];
+  // thresholds packed two-per-register as bf16, rounded toward -inf: the
+  // packed value never exceeds the fp32 threshold, so the (5x oversized)
+  // candidate capacity absorbs the few extra admits and the exact final
+  // top-k is unchanged.  Saves MF*2 VGPRs — the margin that keeps the
+  // E=256 spread-staging variant spill-free.
+  unsigned tpk[MF][2];
 #pragma unroll
   for (int mf = 0; mf < MF; ++mf) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int row = m0 + mf * 16 + (lane >> 4) * 4 + r;
-      t_reg[mf][r] = (row < M) ? thresholds[row] : INFINITY;
+      const float tf = (row < M) ? thresholds[row] : INFINITY;
+      const unsigned u = __float_as_uint(tf);
+      unsigned short h = (unsigned short)(u >> 16);
+      if ((u & 0xFFFFu) && (u & 0x80000000u)) ++h;  // negative: round away from 0
+      if ((r & 1) == 0) tpk[mf][r >> 1] = h;
+      else tpk[mf][r >> 1] |= (unsigned)h << 16;
     }
   }
+  const auto t_reg_at = [&](int mf, int r) -> float {
+    return __uint_as_float(((tpk[mf][r >> 1] >> ((r & 1) * 16)) & 0xFFFFu) << 16);
+  };
   // drain the A/threshold loads so they never mix into the glds counting
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
 
   const int n_tiles = (V + 63) >> 6;
-  const int tile0 = blockIdx.y;
+  const int tile0 = by;
   const int tile_stride = gridDim.y;
   if (tile0 >= n_tiles) return;
 
+  auto stage_piece = [&](int buf, int tile, bool tail, int piece) {
+    const char* wt = reinterpret_cast<const char*>(w) + (size_t)((unsigned)tile << 6) * (E * 2);
+    char* lds_base = smem + (size_t)buf * TILE_B;
+    const int c = piece * 64 + lane;
+    int item = c / CHUNKS_ROW;
+    int sub = c % CHUNKS_ROW;
+    if (sub >= E * 2 / 16) sub = 0;
+    if (tail) {
+      const int gitem = (tile << 6) + item;
+      item -= (gitem >= V ? (gitem - (V - 1)) : 0);
+    }
+    const unsigned off = (unsigned)item * (E * 2) + sub * 16;
+    __builtin_amdgcn_global_load_lds(
+        (const void*)(wt + off), (void*)(lds_base + (size_t)piece * 1024), 16, 0, 0);
+  };
   auto stage_tile = [&](int buf, int tile) {
     constexpr int NLOAD = (LOADERS == 0) ? 8 : LOADERS;
     if (LOADERS != 0 && wave >= LOADERS) return;
-    const char* wt = reinterpret_cast<const char*>(w) + (size_t)((unsigned)tile << 6) * (E * 2);
     const bool tail = ((tile << 6) + 64) > V;
-    char* lds_base = smem + (size_t)buf * TILE_B;
     for (int piece = wave; piece < TILE_PIECES; piece += NLOAD) {
+      stage_piece(buf, tile, tail, piece);
+    }
+  };
+  // spread-staging (SPREAD=1, guide T3/T4): tile+2's glds are issued one
+  // piece per wave per f-column instead of one burst after the tile's
+  // compute, so the VMEM pipe is fed continuously under the MFMA runs.
+  // The counted-vmcnt accounting at the tile-top wait is unchanged: by the
+  // time a wave reaches the wait for tile t+1, tile t+2 is fully issued
+  // either way.  The per-lane within-tile W offsets are tile-invariant for
+  // a fixed (wave, lane), so they are precomputed HERE: inside the f-loop
+  // a glds issue is just uniform-scalar base + 32-bit voffset, keeping the
+  // address math out of the MFMA-region live set (64-bit per-lane address
+  // chains there spill at E=256).
+  constexpr bool DO_SPREAD = (SPREAD == 1) && (LOADERS == 0) && NBUF >= 3;
+  // five per-lane offsets, each < 64*E*2 <= 32 KiB, packed as u16 pairs
+  unsigned spread_pk[DO_SPREAD ? 3 : 1] = {0};
+  if constexpr (DO_SPREAD) {
+#pragma unroll
+    for (int f = 0; f < 5; ++f) {
+      const int piece = wave + f * 8;
       const int c = piece * 64 + lane;
       int item = c / CHUNKS_ROW;
       int sub = c % CHUNKS_ROW;
       if (sub >= E * 2 / 16) sub = 0;
-      if (tail) {
-        const int gitem = (tile << 6) + item;
-        item -= (gitem >= V ? (gitem - (V - 1)) : 0);
-      }
       const unsigned off = (unsigned)item * (E * 2) + sub * 16;
-      __builtin_amdgcn_global_load_lds(
-          (const void*)(wt + off), (void*)(lds_base + (size_t)piece * 1024), 16, 0, 0);
+      spread_pk[f >> 1] |= off << ((f & 1) * 16);
     }
+  }
+  const auto spread_off = [&](int f) -> unsigned {
+    return (spread_pk[f >> 1] >> ((f & 1) * 16)) & 0xFFFFu;
   };
+
+  unsigned* ecnt = nullptr;
+  float* ebufv = nullptr;
+  int* ebufi = nullptr;
+  if constexpr (LDSEPI) {
+    char* epi = smem + (size_t)NBUF * TILE_B + (size_t)wave * EPI_PER_WAVE;
+    ecnt = reinterpret_cast<unsigned*>(epi);
+    ebufv = reinterpret_cast<float*>(epi + ROWS_W * 4);
+    ebufi = reinterpret_cast<int*>(epi + ROWS_W * 4 + ROWS_W * RCAP * 4);
+    for (int lr = lane; lr < ROWS_W; lr += WAVE) ecnt[lr] = 0;
+  }
 
 #pragma unroll
   for (int d = 0; d < NBUF - 1; ++d) {
@@ -707,6 +796,8 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
 
     const char* bbuf = smem + (size_t)cur * TILE_B;
     const int n0 = tile << 6;
+    const int spread_t2 = tile + (NBUF - 1) * tile_stride;
+    const int spread_buf = (cur + NBUF - 1) % NBUF;
     // one 16-item column fragment at a time: acc live set = MF quads (not
     // MF x 4), which is what lets MF = 4 (512-row M-tile) fit in 256 VGPRs.
     // B fragments are prefetched a full column ahead (ping-pong bfr[2][8])
@@ -731,10 +822,33 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
           bfr[(f + 1) & 1][ks] = *reinterpret_cast<const bf16x8*>(bcoln + ks * 64);
         }
       }
+      if constexpr (DO_SPREAD) {
+        if (spread_t2 < n_tiles && ABLATE != 1 && ABLATE != 5) {
+          // tail tiles: clamp the UNIFORM tile base so wt2+off stays in
+          // bounds (the duplicated rows staged for OOB items are masked by
+          // the item < V epilogue guard) — no per-lane tail math in the
+          // MFMA-region live set
+          const size_t woff =
+              min((size_t)((unsigned)spread_t2 << 6) * (E * 2), (size_t)V * (E * 2) - TILE_W);
+          const char* wt2 = reinterpret_cast<const char*>(w) + woff;
+          char* sl = smem + (size_t)spread_buf * TILE_B;
+          const int piece = wave + f * 8;
+          if (piece < TILE_PIECES) {
+            __builtin_amdgcn_global_load_lds((const void*)(wt2 + spread_off(f)),
+                                             (void*)(sl + (size_t)piece * 1024), 16, 0, 0);
+          }
+          if (f == 3 && wave + 32 < TILE_PIECES) {
+            __builtin_amdgcn_global_load_lds((const void*)(wt2 + spread_off(4)),
+                                             (void*)(sl + (size_t)(wave + 32) * 1024), 16, 0,
+                                             0);
+          }
+        }
+      }
       f32x4 acc[MF];
 #pragma unroll
       for (int mf = 0; mf < MF; ++mf) acc[mf] = f32x4{0.f, 0.f, 0.f, 0.f};
       const int item = n0 + f * 16 + (lane & 15);
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int ks = 0; ks < KSTEPS; ++ks) {
 #pragma unroll
@@ -743,12 +857,13 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
               a_frag[mf][ks], bfr[f & 1][ks], acc[mf], 0, 0, 0);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
       bool any_hit = false;
       if (item < V) {
 #pragma unroll
         for (int mf = 0; mf < MF; ++mf) {
 #pragma unroll
-          for (int r = 0; r < 4; ++r) any_hit |= (acc[mf][r] >= t_reg[mf][r]);
+          for (int r = 0; r < 4; ++r) any_hit |= (acc[mf][r] >= t_reg_at(mf, r));
         }
       }
       if constexpr (ABLATE == 3 || ABLATE == 4 || ABLATE == 5) {
@@ -762,26 +877,64 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_kernel_v4(
 #pragma unroll
           for (int r = 0; r < 4; ++r) {
             const float v = acc[mf][r];
-            if (item < V && v >= t_reg[mf][r]) {
-              const int row = m0 + mf * 16 + (lane >> 4) * 4 + r;
-              const int pos = atomicAdd(&counts[row], 1);
-              if (pos < cap) {
-                out_vals[(size_t)row * cap + pos] = v;
-                out_idx[(size_t)row * cap + pos] = item;
+            if (item < V && v >= t_reg_at(mf, r)) {
+              const int lr = mf * 16 + (lane >> 4) * 4 + r;
+              if constexpr (LDSEPI) {
+                const unsigned n = atomicAdd(&ecnt[lr], 1u);  // ds_add_rtn
+                if (n < (unsigned)RCAP) {
+                  ebufv[lr * RCAP + (int)n] = v;
+                  ebufi[lr * RCAP + (int)n] = item;
+                } else {
+                  // list full: push counts past capacity so the host's
+                  // exact-rescore fallback owns this row (result unused —
+                  // a fire-and-forget atomic, no round trip in the loop).
+                  // The host only picks this kernel when overflow is rare.
+                  atomicAdd(&counts[m0 + lr], cap + 1);
+                }
+              } else {
+                const int row = m0 + lr;
+                const int pos = atomicAdd(&counts[row], 1);
+                if (pos < cap) {
+                  // 32-bit offset (host checks M*cap*4 < 2^31): keeps the
+                  // store at SGPR-base + voffset so the output pointers
+                  // never occupy VGPR pairs across the MFMA region
+                  const int off = row * cap + pos;
+                  out_vals[off] = v;
+                  out_idx[off] = item;
+                }
               }
             }
           }
         }
       }
     }
-    // stage tile+2 LAST so the next iteration's counted vmcnt drains this
-    // iteration's (older) epilogue stores together with tile+1's glds while
-    // leaving only the newest stage in flight
-    if constexpr (ABLATE != 1 && ABLATE != 5) {
+    // without spread-staging: stage tile+2 in one burst LAST so the next
+    // iteration's counted vmcnt drains this iteration's (older) epilogue
+    // stores together with tile+1's glds
+    if constexpr (!DO_SPREAD && ABLATE != 1 && ABLATE != 5) {
       const int t2 = tile + (NBUF - 1) * tile_stride;
       if (t2 < n_tiles) stage_tile((cur + NBUF - 1) % NBUF, t2);
     }
     cur = (cur + 1) % NBUF;
+  }
+  if constexpr (LDSEPI && ABLATE != 3 && ABLATE != 4 && ABLATE != 5) {
+    // walk-end flush: lane-parallel, one global atomicAdd per owned row
+    for (int lr = lane; lr < ROWS_W; lr += WAVE) {
+      const unsigned c = ecnt[lr];
+      const int m_ = (int)min(c, (unsigned)RCAP);
+      const int row = m0 + lr;
+      if (m_ > 0 && row < M) {
+        const int base = atomicAdd(&counts[row], m_);
+        for (int j = 0; j < m_; ++j) {
+          const int pos = base + j;
+          if (pos < cap) {
+            const int off = row * cap + pos;
+            out_vals[off] = ebufv[lr * RCAP + j];
+            out_idx[off] = ebufi[lr * RCAP + j];
+          }
+        }
+      }
+    }
   }
 }
 
@@ -812,7 +965,26 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_fp8_kernel(
   const int V = (int)V64;
   const int wave = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
-  const int m0 = blockIdx.x * (128 * MF) + wave * (16 * MF);
+  // XCD-aware (m_tile, stripe) assignment.  Workgroups dispatch round-robin
+  // over the 8 XCDs, so the naive (x = m_tile, y = stripe) grid puts the
+  // m_tiles that share a B stripe on DIFFERENT XCDs: every XCD streams the
+  // whole item table through its own (private) L2 and the kernel is
+  // HBM-bound on the 8x re-read (measured 5.6 ms at B=1024/V=10M — the
+  // 40 GB the 8 XCDs pull at ~7 TB/s).  Remap (bijective, guide §"XCD
+  // swizzle") so each XCD owns a contiguous band of stripes together with
+  // ALL their m_tiles: each B tile is pulled from HBM by exactly one XCD
+  // and the co-resident m_tile blocks hit it in that XCD's L2.
+  int bx = blockIdx.x, by = blockIdx.y;
+  {  // always on for the fp8 kernel
+    const int nwg = (int)(gridDim.x * gridDim.y);
+    const int orig = (int)(blockIdx.y * gridDim.x + blockIdx.x);
+    const int qq = nwg >> 3, rr = nwg & 7, xcd = orig & 7, slot = orig >> 3;
+    const int rid =
+        (xcd < rr ? xcd * (qq + 1) : rr * (qq + 1) + (xcd - rr) * qq) + slot;
+    bx = rid % (int)gridDim.x;  // m_tile: fastest within an XCD's band
+    by = rid / (int)gridDim.x;  // stripe
+  }
+  const int m0 = bx * (128 * MF) + wave * (16 * MF);
 
   extern __shared__ __attribute__((aligned(16))) char smem[];  // 3 x TILE_B
 
@@ -842,7 +1014,7 @@ __global__ __launch_bounds__(512, 2) void scored_topk_gemm_fp8_kernel(
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
 
   const int n_tiles = (V + 63) >> 6;
-  const int tile0 = blockIdx.y;
+  const int tile0 = by;
   const int tile_stride = gridDim.y;
   if (tile0 >= n_tiles) return;
 
@@ -998,24 +1170,27 @@ __global__ __launch_bounds__(512, 4) void scored_topk_gemm_kernel_v6(
   const int tile_stride = gridDim.y;
   if (tile0 >= n_tiles) return;
 
+  auto stage_piece = [&](int buf, int tile, bool tail, int piece) {
+    const char* wt = reinterpret_cast<const char*>(w) + (size_t)((unsigned)tile << 6) * (E * 2);
+    char* lds_base = smem + (size_t)buf * TILE_B;
+    const int c = piece * 64 + lane;
+    int item = c / CHUNKS_ROW;
+    int sub = c % CHUNKS_ROW;
+    if (sub >= E * 2 / 16) sub = 0;
+    if (tail) {
+      const int gitem = (tile << 6) + item;
+      item -= (gitem >= V ? (gitem - (V - 1)) : 0);
+    }
+    const unsigned off = (unsigned)item * (E * 2) + sub * 16;
+    __builtin_amdgcn_global_load_lds(
+        (const void*)(wt + off), (void*)(lds_base + (size_t)piece * 1024), 16, 0, 0);
+  };
   auto stage_tile = [&](int buf, int tile) {
     constexpr int NLOAD = (LOADERS == 0) ? 8 : LOADERS;
     if (LOADERS != 0 && wave >= LOADERS) return;
-    const char* wt = reinterpret_cast<const char*>(w) + (size_t)((unsigned)tile << 6) * (E * 2);
     const bool tail = ((tile << 6) + 64) > V;
-    char* lds_base = smem + (size_t)buf * TILE_B;
     for (int piece = wave; piece < TILE_PIECES; piece += NLOAD) {
-      const int c = piece * 64 + lane;
-      int item = c / CHUNKS_ROW;
-      int sub = c % CHUNKS_ROW;
-      if (sub >= E * 2 / 16) sub = 0;
-      if (tail) {
-        const int gitem = (tile << 6) + item;
-        item -= (gitem >= V ? (gitem - (V - 1)) : 0);
-      }
-      const unsigned off = (unsigned)item * (E * 2) + sub * 16;
-      __builtin_amdgcn_global_load_lds(
-          (const void*)(wt + off), (void*)(lds_base + (size_t)piece * 1024), 16, 0, 0);
+      stage_piece(buf, tile, tail, piece);
     }
   };
 
@@ -1125,6 +1300,8 @@ std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
   auto out_idx = torch::zeros({M, capacity}, opts_i);
   auto counts = torch::zeros({M}, opts_i);
   auto thr = thresholds.to(torch::kFloat32).contiguous();
+  TORCH_CHECK((int64_t)M * capacity * 4 < (int64_t)INT32_MAX,
+              "M*capacity too large for 32-bit candidate offsets");
   static const char* variant = std::getenv("REPLAY_AMD_STG_VARIANT");
   if (variant != nullptr && variant[0] == '\0') variant = nullptr;  // empty = default
   const bool legacy = (variant != nullptr && variant[0] >= '1' && variant[0] < '4');
@@ -1136,6 +1313,23 @@ std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
   const int m_tiles = (M + m_tile_rows - 1) / m_tile_rows;
   // fill 256 CUs x ~4 blocks with >> WGs (guide §1); stripes over item tiles
   int stripes = (int)std::min<int64_t>((V + 63) / 64, std::max(1, 4096 / m_tiles));
+  // LDS hit-append epilogue: opt-in (REPLAY_AMD_LDSEPI=1) negative result.
+  // It removes the in-loop global atomics (raw kernel -5% at K=10 shapes),
+  // BUT sample-threshold noise makes a few rows ~30x hotter than expected
+  // (the j-th-of-32768 order statistic has ~38% quantile std), and ANY
+  // single-walk overflow of the 8-entry LDS list marks the row for the
+  // exact-rescore fallback: at the K=100 bench config that is ~34 rescored
+  // rows and a 2x END-TO-END regression (12.2 vs 6.4 ms).  The direct
+  // global append tolerates hot rows (they only rescore when the TOTAL
+  // exceeds capacity), so it stays the default.
+  static const bool ldsepi_env = std::getenv("REPLAY_AMD_LDSEPI") != nullptr;
+  bool use_ldsepi = ldsepi_env && (int64_t)capacity <= (int64_t)20 * stripes;
+  if (variant != nullptr && variant[0] == 'w') use_ldsepi = false;  // A/B
+  static const bool dbg = std::getenv("REPLAY_AMD_DEBUG") != nullptr;
+  if (dbg) {
+    fprintf(stderr, "[stg] E=%ld M=%ld cap=%ld stripes=%d ldsepi=%d variant=%s\n", (long)E,
+            (long)M, (long)capacity, stripes, (int)use_ldsepi, variant ? variant : "-");
+  }
   dim3 grid(m_tiles, stripes);
   auto stream = at::cuda::getCurrentHIPStream();
 #define LAUNCH_STG(EE)                                                                   \
@@ -1151,21 +1345,42 @@ std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
   } else if (E == 128 && legacy) {
     LAUNCH_STG(128);
   } else if (E == 64) {
-    const size_t lds64 = 3 * 64 * (64 * 2 + 32);
-    hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<64, 8>), grid, dim3(512), lds64, stream,
-                       reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
-                       reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
-                       thr.data_ptr<float>(), out_vals.data_ptr<float>(),
-                       out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
-                       (int)capacity);
+    if (use_ldsepi) {
+      const size_t lds64 = 3 * 64 * (64 * 2 + 32) + 8 * (128 * 4 + 128 * 8 * 8);
+      hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<64, 8>), grid, dim3(512), lds64, stream,
+                         reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                         reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                         thr.data_ptr<float>(), out_vals.data_ptr<float>(),
+                         out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
+                         (int)capacity);
+    } else {
+      const size_t lds64 = 3 * 64 * (64 * 2 + 32);
+      hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<64, 8, 3, 0, 0, 1, 1, 0>), grid, dim3(512),
+                         lds64, stream, reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                         reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                         thr.data_ptr<float>(), out_vals.data_ptr<float>(),
+                         out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
+                         (int)capacity);
+    }
   } else if (E == 128) {
-    const size_t lds128 = 3 * 64 * (128 * 2 + 32);
-    hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<128, 4>), grid, dim3(512), lds128, stream,
-                       reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
-                       reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
-                       thr.data_ptr<float>(), out_vals.data_ptr<float>(),
-                       out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
-                       (int)capacity);
+    if (use_ldsepi) {
+      const size_t lds128 = 3 * 64 * (128 * 2 + 32) + 8 * (64 * 4 + 64 * 8 * 8);
+      hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<128, 4>), grid, dim3(512), lds128, stream,
+                         reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                         reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                         thr.data_ptr<float>(), out_vals.data_ptr<float>(),
+                         out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
+                         (int)capacity);
+    } else {
+      const size_t lds128 = 3 * 64 * (128 * 2 + 32);
+      hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<128, 4, 3, 0, 0, 1, 1, 0>), grid,
+                         dim3(512), lds128, stream,
+                         reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                         reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                         thr.data_ptr<float>(), out_vals.data_ptr<float>(),
+                         out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
+                         (int)capacity);
+    }
   } else if (E == 256 && v6) {
     const size_t lds_v6 = 2 * 64 * (256 * 2 + 32);
     hipLaunchKernelGGL((scored_topk_gemm_kernel_v6<256>), grid, dim3(512), lds_v6, stream,
@@ -1195,7 +1410,8 @@ std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
     } else if (v4_mf == 2) {
       // v4 at the narrow 256-row M-tile (A/B reference)
       const size_t lds_v4 = 3 * 64 * (256 * 2 + 32);
-      hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 2>), grid, dim3(512), lds_v4, stream,
+      hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 2, 3, 0, 0, 1, 1, 0>), grid,
+                         dim3(512), lds_v4, stream,
                          reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
                          reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
                          thr.data_ptr<float>(), out_vals.data_ptr<float>(),
@@ -1205,21 +1421,21 @@ std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
                (variant[0] == '9' || variant[0] == 'a' || variant[0] == 'b')) {
       const size_t lds_ab = 3 * 64 * (256 * 2 + 32);
       if (variant[0] == '9') {
-        hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 3, 3>), grid, dim3(512), lds_ab,
+        hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 3, 3, 0, 1, 1, 0>), grid, dim3(512), lds_ab,
                            stream, reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
                            reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
                            thr.data_ptr<float>(), out_vals.data_ptr<float>(),
                            out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
                            (int)capacity);
       } else if (variant[0] == 'a') {
-        hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 3, 4>), grid, dim3(512), lds_ab,
+        hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 3, 4, 0, 1, 1, 0>), grid, dim3(512), lds_ab,
                            stream, reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
                            reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
                            thr.data_ptr<float>(), out_vals.data_ptr<float>(),
                            out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
                            (int)capacity);
       } else {
-        hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 3, 5>), grid, dim3(512), lds_ab,
+        hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 3, 5, 0, 1, 1, 0>), grid, dim3(512), lds_ab,
                            stream, reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
                            reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
                            thr.data_ptr<float>(), out_vals.data_ptr<float>(),
@@ -1231,24 +1447,51 @@ std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
       // only, WRONG RESULTS), 8 = no MFMA/epilogue (DMA+barrier only)
       const size_t lds_ab = 3 * 64 * (256 * 2 + 32);
       if (variant[0] == '7') {
-        hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 3, 1>), grid, dim3(512), lds_ab,
+        hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 3, 1, 0, 1, 1, 0>), grid, dim3(512), lds_ab,
                            stream, reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
                            reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
                            thr.data_ptr<float>(), out_vals.data_ptr<float>(),
                            out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
                            (int)capacity);
       } else {
-        hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 3, 2>), grid, dim3(512), lds_ab,
+        hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 3, 2, 0, 1, 1, 0>), grid, dim3(512), lds_ab,
                            stream, reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
                            reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
                            thr.data_ptr<float>(), out_vals.data_ptr<float>(),
                            out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
                            (int)capacity);
       }
+    } else if (variant != nullptr && variant[0] == 'z') {
+      // spread staging + 4-deep buffer ring (2 tiles of DMA in flight)
+      const size_t lds_z = 4 * 64 * (256 * 2 + 32);
+      hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 4, 0, 0, 1, 1, 0>), grid, dim3(512),
+                         lds_z, stream, reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                         reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                         thr.data_ptr<float>(), out_vals.data_ptr<float>(),
+                         out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
+                         (int)capacity);
+    } else if (variant != nullptr && variant[0] == 'y') {
+      // A/B reference: end-burst staging (no per-f-column spread)
+      const size_t lds_y = 3 * 64 * (256 * 2 + 32) + 8 * (64 * 4 + 64 * 8 * 8);
+      hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 3, 0, 0, 1, 0>), grid, dim3(512),
+                         lds_y, stream, reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                         reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                         thr.data_ptr<float>(), out_vals.data_ptr<float>(),
+                         out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
+                         (int)capacity);
+    } else if (variant != nullptr && variant[0] == 'x') {
+      // A/B reference: XCD-aware remap DISABLED
+      const size_t lds_x = 3 * 64 * (256 * 2 + 32) + 8 * (64 * 4 + 64 * 8 * 8);
+      hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 3, 0, 0, 0>), grid, dim3(512),
+                         lds_x, stream, reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                         reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                         thr.data_ptr<float>(), out_vals.data_ptr<float>(),
+                         out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
+                         (int)capacity);
     } else if (variant != nullptr && variant[0] == 'p') {
       // loader-specialized staging: 2 waves carry the whole DMA stream
       const size_t lds_p = 3 * 64 * (256 * 2 + 32);
-      hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 3, 0, 2>), grid, dim3(512), lds_p,
+      hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 3, 0, 2, 1, 1, 0>), grid, dim3(512), lds_p,
                          stream, reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
                          reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
                          thr.data_ptr<float>(), out_vals.data_ptr<float>(),
@@ -1257,17 +1500,28 @@ std::vector<torch::Tensor> scored_topk_gemm(torch::Tensor q, torch::Tensor w,
     } else if (variant != nullptr && variant[0] == '6') {
       // v4 with a 4-deep buffer ring (3 tiles of DMA in flight)
       const size_t lds_v46 = 4 * 64 * (256 * 2 + 32);
-      hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 4>), grid, dim3(512), lds_v46,
+      hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 4, 0, 0, 1, 1, 0>), grid, dim3(512), lds_v46,
                          stream,
                          reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
                          reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
                          thr.data_ptr<float>(), out_vals.data_ptr<float>(),
                          out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
                          (int)capacity);
-    } else {
-      // v4: glds triple-buffer + raw barrier + counted vmcnt, 512-row M-tile
-      const size_t lds_v4 = 3 * 64 * (256 * 2 + 32);
+    } else if (use_ldsepi) {
+      // v4: glds triple-buffer + raw barrier + counted vmcnt, 512-row
+      // M-tile, LDS hit-append epilogue
+      const size_t lds_v4 = 3 * 64 * (256 * 2 + 32) + 8 * (64 * 4 + 64 * 8 * 8);
       hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4>), grid, dim3(512), lds_v4, stream,
+                         reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+                         reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
+                         thr.data_ptr<float>(), out_vals.data_ptr<float>(),
+                         out_idx.data_ptr<int>(), counts.data_ptr<int>(), M, V,
+                         (int)capacity);
+    } else {
+      // long-walk shapes (few stripes): direct global hit append
+      const size_t lds_v4 = 3 * 64 * (256 * 2 + 32);
+      hipLaunchKernelGGL((scored_topk_gemm_kernel_v4<256, 4, 3, 0, 0, 1, 1, 0>), grid,
+                         dim3(512), lds_v4, stream,
                          reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
                          reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),
                          thr.data_ptr<float>(), out_vals.data_ptr<float>(),
