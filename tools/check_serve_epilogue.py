@@ -14,7 +14,7 @@ def parity(M, V, E, k=10, seed=0):
     for r in range(M):
         if not torch.equal(torch.sort(i[r])[0], torch.sort(ri[r])[0]):
             # allow tie-at-boundary rows
-            vs, rs = torch.sort(v[r])[0], torch.sort(rv[r])[0]
+            vs, rs = torch.sort(v[r].float())[0], torch.sort(rv[r].float())[0]
             if not torch.allclose(vs, rs, rtol=2e-2, atol=2e-2):
                 bad += 1
     print(f"parity M={M} V={V} E={E}: mismatched rows = {bad}")

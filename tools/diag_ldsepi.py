@@ -18,7 +18,8 @@ qr = sample.shape[1] / V
 j = max(1, math.ceil(k * qr + 3.0 * math.sqrt(max(k * qr, 1e-9)) + 2))
 j += math.ceil(seen.shape[1] * qr) + 1
 j = min(j, sample.shape[1])
-thr = sample.topk(j, dim=1).values[:, -1]
+from replay_amd.ops.topk import _tail_threshold
+thr = _tail_threshold(sample, j)
 capacity = max(4 * k, int(5.0 * j / qr))
 print(f"j={j} capacity={capacity} stripes~{4096//2}")
 
