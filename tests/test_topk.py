@@ -99,3 +99,37 @@ def test_fused_fp8_matches_chunked():
             len(set(i_f[r].tolist()) & set(i_c[r].tolist())) for r in range(M)
         )
         assert inter / (M * 100) > 0.97
+
+
+def test_tail_threshold_statistics():
+    """_tail_threshold targets ~j/qr admits with FAR lower spread than the
+    raw j-th order statistic (the hot-row pathology behind the rescore
+    fallback).  Pure-CPU statistical check on Gaussian scores."""
+    import torch
+
+    from replay_amd.ops.topk import _tail_threshold
+
+    torch.manual_seed(0)
+    B, n, V = 256, 32768, 10_000_000
+    qr = n / V
+    sample = torch.randn(B, n)
+    j = 7
+    thr = _tail_threshold(sample, j)
+    assert thr.shape == (B,)
+    # expected admits per row over the FULL population ~ N(0,1) tail mass
+    from scipy.stats import norm
+
+    admits = torch.tensor([(1 - norm.cdf(t)) * V for t in thr.tolist()])
+    target = j / qr
+    # conservative curvature: mean admits in [0.3, 1.5] x target, and no
+    # row more than ~4x the mean (the raw order stat shows 30x outliers)
+    assert 0.3 * target < admits.mean() < 1.5 * target
+    assert admits.max() < 4.0 * admits.mean()
+    # guard branches: large j and tiny samples fall back to the order stat
+    small = torch.randn(4, 100)
+    t_small = _tail_threshold(small, 3)
+    ref = small.topk(3, dim=1).values[:, -1]
+    torch.testing.assert_close(t_small, ref)
+    t_large = _tail_threshold(sample[:4], 100)
+    ref_large = sample[:4].topk(100, dim=1).values[:, -1]
+    torch.testing.assert_close(t_large, ref_large)
