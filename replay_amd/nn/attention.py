@@ -46,8 +46,10 @@ class MultiheadAttention(torch.nn.Module):
         self.num_heads = num_heads
         self.head_dim = embed_dim // num_heads
         self.dropout = dropout
-        self.in_proj = torch.nn.Linear(embed_dim, 3 * embed_dim, bias=bias)
-        self.out_proj = torch.nn.Linear(embed_dim, embed_dim, bias=bias)
+        from replay_amd.ops.fast_linear import ChunkedWgradLinear
+
+        self.in_proj = ChunkedWgradLinear(embed_dim, 3 * embed_dim, bias=bias)
+        self.out_proj = ChunkedWgradLinear(embed_dim, embed_dim, bias=bias)
 
     def _can_use_flash(self, x: torch.Tensor, attn_mask) -> bool:
         from replay_amd.nn.mask import MaskSpec

@@ -26,8 +26,10 @@ class PointWiseFeedForward(torch.nn.Module):
         hidden = hidden_dim or embedding_dim
         if activation not in ("relu", "gelu"):
             raise ValueError("activation must be relu or gelu")
-        self.w1 = torch.nn.Linear(embedding_dim, hidden)
-        self.w2 = torch.nn.Linear(hidden, embedding_dim)
+        from replay_amd.ops.fast_linear import ChunkedWgradLinear
+
+        self.w1 = ChunkedWgradLinear(embedding_dim, hidden)
+        self.w2 = ChunkedWgradLinear(hidden, embedding_dim)
         self.dropout1 = torch.nn.Dropout(dropout)
         self.dropout2 = torch.nn.Dropout(dropout)
         self.activation = torch.nn.ReLU() if activation == "relu" else torch.nn.GELU()
@@ -44,9 +46,11 @@ class SwiGLU(torch.nn.Module):
     def __init__(self, embedding_dim: int, hidden_dim: int = None, dropout: float = 0.0) -> None:
         super().__init__()
         hidden = hidden_dim or embedding_dim * 2
-        self.WG = torch.nn.Linear(embedding_dim, hidden)
-        self.W1 = torch.nn.Linear(embedding_dim, hidden)
-        self.W2 = torch.nn.Linear(hidden, embedding_dim)
+        from replay_amd.ops.fast_linear import ChunkedWgradLinear
+
+        self.WG = ChunkedWgradLinear(embedding_dim, hidden)
+        self.W1 = ChunkedWgradLinear(embedding_dim, hidden)
+        self.W2 = ChunkedWgradLinear(hidden, embedding_dim)
         self.dropout = torch.nn.Dropout(dropout)
 
     def reset_parameters(self) -> None:

@@ -133,3 +133,34 @@ def test_tail_threshold_statistics():
     t_large = _tail_threshold(sample[:4], 100)
     ref_large = sample[:4].topk(100, dim=1).values[:, -1]
     torch.testing.assert_close(t_large, ref_large)
+
+
+def test_chunked_wgrad_linear_matches_plain():
+    """ChunkedWgradLinear must produce identical forward and matching
+    gradients vs torch.nn.Linear (the chunked dW path forced by a small
+    _MIN_ROWS)."""
+    import torch
+
+    import replay_amd.ops.fast_linear as fl
+
+    torch.manual_seed(0)
+    old = fl._MIN_ROWS
+    fl._MIN_ROWS = 128
+    try:
+        for n, fi, fo in ((256, 16, 48), (320, 24, 8)):
+            ref = torch.nn.Linear(fi, fo)
+            fast = fl.ChunkedWgradLinear(fi, fo)
+            fast.load_state_dict(ref.state_dict())
+            x1 = torch.randn(n, fi, requires_grad=True)
+            x2 = x1.detach().clone().requires_grad_(True)
+            y1 = ref(x1)
+            y2 = fast(x2)
+            torch.testing.assert_close(y1, y2)
+            g = torch.randn_like(y1)
+            y1.backward(g)
+            y2.backward(g)
+            torch.testing.assert_close(x1.grad, x2.grad, rtol=1e-5, atol=1e-6)
+            torch.testing.assert_close(ref.weight.grad, fast.weight.grad, rtol=1e-4, atol=1e-5)
+            torch.testing.assert_close(ref.bias.grad, fast.bias.grad, rtol=1e-5, atol=1e-6)
+    finally:
+        fl._MIN_ROWS = old
