@@ -164,3 +164,30 @@ def test_chunked_wgrad_linear_matches_plain():
             torch.testing.assert_close(ref.bias.grad, fast.bias.grad, rtol=1e-5, atol=1e-6)
     finally:
         fl._MIN_ROWS = old
+
+
+def test_chunked_wgrad_linear_3d_and_module_swap():
+    """3-D activations reshape correctly, and the swapped-in projections in
+    MultiheadAttention / PointWiseFeedForward / SwiGLU keep nn.Linear
+    state-dict layout."""
+    import torch
+
+    from replay_amd.nn.attention import MultiheadAttention
+    from replay_amd.nn.ffn import PointWiseFeedForward, SwiGLU
+    from replay_amd.ops.fast_linear import ChunkedWgradLinear
+
+    lin = ChunkedWgradLinear(8, 12)
+    x = torch.randn(4, 5, 8, requires_grad=True)
+    y = lin(x)
+    assert y.shape == (4, 5, 12)
+    y.sum().backward()
+    assert lin.weight.grad.shape == (12, 8)
+    assert x.grad.shape == x.shape
+
+    mha = MultiheadAttention(16, 2)
+    sd = mha.state_dict()
+    assert "in_proj.weight" in sd and sd["in_proj.weight"].shape == (48, 16)
+    ffn = PointWiseFeedForward(16)
+    assert ffn.state_dict()["w1.weight"].shape == (16, 16)
+    sw = SwiGLU(16)
+    assert sw.state_dict()["WG.weight"].shape == (32, 16)
