@@ -291,14 +291,20 @@ __global__ __launch_bounds__(256, 2) void ce_linear_bwd_kernel(
   // free: a WG-shared stage needed 2 __syncthreads per tile, which
   // serialized the workgroup)] [4 waves x dl bounce 16x64 bf16]
   // [per-row state: 3 x 256 words] [!RESIDENT: 256 x E bf16 A tile]
+  // dl bounce rows per wave: E <= 128 holds ALL MF row-fragments (64 rows)
+  // so the tile needs ONE lgkm drain between the epilogue writes and the
+  // coalesced store sweep — the 16-row per-mf bounce serialized ~70
+  // full lgkmcnt(0) round trips per tile (measured, .s).  E = 256 keeps
+  // the 16-row tile (the !RESIDENT q stage already uses 128 KB of LDS).
+  constexpr int DLROWS = (E >= 256) ? 16 : 64;
   __hip_bfloat16* wt_lds = reinterpret_cast<__hip_bfloat16*>(smem);
   __hip_bfloat16* my_wt = wt_lds + (FUSE_DH ? (size_t)wave * E * 64 : 0);
   __hip_bfloat16* dl_lds = wt_lds + (FUSE_DH ? 4 * E * 64 : 0);
-  float* g_lds = reinterpret_cast<float*>(dl_lds + 4 * 16 * 64);
+  float* g_lds = reinterpret_cast<float*>(dl_lds + 4 * DLROWS * 64);
   float* adj_lds = g_lds + 256;
   int* lab_lds = reinterpret_cast<int*>(adj_lds + 256);
   __hip_bfloat16* q_lds = reinterpret_cast<__hip_bfloat16*>(lab_lds + 256);
-  __hip_bfloat16* my_dl = dl_lds + wave * 16 * 64;
+  __hip_bfloat16* my_dl = dl_lds + wave * DLROWS * 64;
   // 16-B-granular XOR swizzles (8 bf16 granules stay contiguous)
   auto dl_off = [&](int row, int col) {  // [16][64] bounce tile, elements
     return row * 64 + (col ^ ((row & 7) << 3));
@@ -458,10 +464,15 @@ __global__ __launch_bounds__(256, 2) void ce_linear_bwd_kernel(
       const int lrow0 = wave * 64 + mf * 16 + (lane >> 4) * 4;
       float adj4[4];
       int lab4[4];
+      float g4[4];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         adj4[r] = adj_lds[lrow0 + r];
         lab4[r] = lab_lds[lrow0 + r];
+        // hoisted: an in-branch g_lds read compiles to a ds_read +
+        // lgkmcnt(0) at EVERY (f, r) use — 64 serialized LDS round trips
+        // per tile (measured dominant stall in the .s)
+        g4[r] = g_lds[lrow0 + r];
       }
 #pragma unroll
       for (int f = 0; f < 4; ++f) {
@@ -471,9 +482,10 @@ __global__ __launch_bounds__(256, 2) void ce_linear_bwd_kernel(
           float dl = 0.f;
           if (item < Vi) {
             dl = __expf(acc[m2][f][r] - adj4[r]);
-            if (item == lab4[r]) dl -= g_lds[lrow0 + r];
+            if (item == lab4[r]) dl -= g4[r];
           }
-          my_dl[dl_off((lane >> 4) * 4 + r, f * 16 + (lane & 15))] =
+          const int dlr0 = (DLROWS == 64) ? mf * 16 : 0;
+          my_dl[dl_off(dlr0 + (lane >> 4) * 4 + r, f * 16 + (lane & 15))] =
               __float2bfloat16(dl * gsign);
         }
       }
@@ -482,7 +494,8 @@ __global__ __launch_bounds__(256, 2) void ce_linear_bwd_kernel(
 #pragma unroll
         for (int ks2 = 0; ks2 < 2; ++ks2) {
           bf16x8 a_dl = *reinterpret_cast<const bf16x8*>(
-              my_dl + dl_off(lane & 15, ks2 * 32 + (lane >> 4) * 8));
+              my_dl + dl_off(((DLROWS == 64) ? mf * 16 : 0) + (lane & 15),
+                             ks2 * 32 + (lane >> 4) * 8));
 #pragma unroll
           for (int f = 0; f < OF; ++f) {
             bf16x8 b_w = *reinterpret_cast<const bf16x8*>(
@@ -492,7 +505,7 @@ __global__ __launch_bounds__(256, 2) void ce_linear_bwd_kernel(
           }
         }
       }
-      if constexpr (WRITE_DL) {
+      if constexpr (WRITE_DL && DLROWS == 16) {
         // coalesced store: each lane writes 16-B chunks of the bounce tile
         for (int i = lane; i < 16 * 8; i += WAVE) {
           const int row = row0 + i / 8;
@@ -504,6 +517,19 @@ __global__ __launch_bounds__(256, 2) void ce_linear_bwd_kernel(
         }
       }
     }
+    }
+    if constexpr (WRITE_DL && DLROWS == 64) {
+      // single store sweep over ALL 64 bounce rows: one lgkm drain covers
+      // every epilogue ds_write of this tile
+#pragma unroll
+      for (int i = lane; i < 64 * 8; i += WAVE) {
+        const int row = m0 + i / 8;
+        const int c0 = (i % 8) * 8;
+        if (row < M) {
+          *reinterpret_cast<bf16x8*>(dlogits + (size_t)row * ldd + n0 + c0) =
+              *reinterpret_cast<const bf16x8*>(my_dl + dl_off(i / 8, c0));
+        }
+      }
     }
   }
   if constexpr (FUSE_DH) {
@@ -708,7 +734,7 @@ torch::Tensor ce_linear_bwd_fused_dh(torch::Tensor hidden, torch::Tensor w,
   auto labels_c = labels.contiguous();
   const int m_tiles = (M + 255) / 256;
   auto stream = at::cuda::getCurrentHIPStream();
-#define LAUNCH_BF(EE)                                                                         do {                                                                                          const size_t lds = (size_t)4 * EE * 64 * 2 + 4 * 16 * 64 * 2 + 3 * 256 * 4 + 64;            hipLaunchKernelGGL((ce_linear_bwd_kernel<EE, true, true, false>), dim3(m_tiles),                               dim3(256), lds, stream,                                                                     reinterpret_cast<const __hip_bfloat16*>(hidden.data_ptr()),                                 reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),                                      labels_c.data_ptr<int64_t>(), lse.data_ptr<float>(),                                        gscale.data_ptr<float>(), (float)gsign, nullptr,                                            reinterpret_cast<__hip_bfloat16*>(dhidden.data_ptr()), M, V, 0);       } while (0)
+#define LAUNCH_BF(EE)                                                                         do {                                                                                          const size_t lds = (size_t)4 * EE * 64 * 2 + 4 * 64 * 64 * 2 + 3 * 256 * 4 + 64;            hipLaunchKernelGGL((ce_linear_bwd_kernel<EE, true, true, false>), dim3(m_tiles),                               dim3(256), lds, stream,                                                                     reinterpret_cast<const __hip_bfloat16*>(hidden.data_ptr()),                                 reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),                                      labels_c.data_ptr<int64_t>(), lse.data_ptr<float>(),                                        gscale.data_ptr<float>(), (float)gsign, nullptr,                                            reinterpret_cast<__hip_bfloat16*>(dhidden.data_ptr()), M, V, 0);       } while (0)
   if (E == 64) {
     LAUNCH_BF(64);
   } else {
@@ -813,8 +839,9 @@ std::vector<torch::Tensor> ce_linear_bwd(torch::Tensor hidden, torch::Tensor w,
 #define LAUNCH_CLB(EE, FDH)                                                                  \
   do {                                                                                       \
     constexpr bool RES = (EE <= 128);                                                        \
-    size_t lds = (FDH ? (size_t)4 * EE * 64 * 2 : 0) + 4 * 16 * 64 * 2 + 3 * 256 * 4 +       \
-                 (RES ? 0 : (size_t)256 * EE * 2);                                           \
+    constexpr int DLR = (EE >= 256) ? 16 : 64;                                               \
+    size_t lds = (FDH ? (size_t)4 * EE * 64 * 2 : 0) + (size_t)4 * DLR * 64 * 2 +            \
+                 3 * 256 * 4 + (RES ? 0 : (size_t)256 * EE * 2);                             \
     hipLaunchKernelGGL((ce_linear_bwd_kernel<EE, RES, FDH>), dim3(m_tiles), dim3(256), lds,  \
                        stream, reinterpret_cast<const __hip_bfloat16*>(hidden.data_ptr()),   \
                        reinterpret_cast<const __hip_bfloat16*>(w.data_ptr()),                \
