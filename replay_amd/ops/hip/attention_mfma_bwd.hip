@@ -130,22 +130,38 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_mfma_kernel(
     for (int f = 0; f < OF; ++f) dq_acc[f] = f32x4{0.f, 0.f, 0.f, 0.f};
 
     const int kt_end = causal ? ((q0 + 15) >> 5) + 1 : n_t32;
+    // K/V prefetched one key-tile ahead in registers: the direct load->MFMA
+    // form compiled to a full vmcnt(0) drain per use (~10 per tile,
+    // measured 80% wave parking); with the ring the waits are counted and
+    // overlap the previous tile's epilogue.
+    auto load_kv = [&](bf16x8 (&bk)[2][KS], bf16x8 (&bv)[2][KS], int kt32) {
+      const int kk0 = (lane >> 4) * 8;
+#pragma unroll
+      for (int half = 0; half < 2; ++half) {
+        const int key = (kt32 << 5) + half * 16 + (lane & 15);
+        const size_t kbase = base + (size_t)min(key, L - 1) * DH;
+#pragma unroll
+        for (int s = 0; s < KS; ++s) {
+          bk[half][s] = *reinterpret_cast<const bf16x8*>(k + kbase + s * 32 + kk0);
+          bv[half][s] = *reinterpret_cast<const bf16x8*>(v + kbase + s * 32 + kk0);
+        }
+      }
+    };
+    bf16x8 bk_cur[2][KS], bv_cur[2][KS], bk_nxt[2][KS], bv_nxt[2][KS];
+    load_kv(bk_cur, bv_cur, 0);
     for (int ktile = 0; ktile < kt_end; ++ktile) {  // NOT "kt": shadows the LDS stage ptr
       const int k0 = ktile << 5;
+      if (ktile + 1 < kt_end) load_kv(bk_nxt, bv_nxt, ktile + 1);
       f32x4 s_acc[2] = {f32x4{0.f, 0.f, 0.f, 0.f}, f32x4{0.f, 0.f, 0.f, 0.f}};
       f32x4 dp_acc[2] = {f32x4{0.f, 0.f, 0.f, 0.f}, f32x4{0.f, 0.f, 0.f, 0.f}};
 #pragma unroll
       for (int half = 0; half < 2; ++half) {
-        const int key = k0 + half * 16 + (lane & 15);
-        const int kk0 = (lane >> 4) * 8;
-        const size_t kbase = base + (size_t)min(key, L - 1) * DH;
 #pragma unroll
         for (int s = 0; s < KS; ++s) {
-          bf16x8 b_k = *reinterpret_cast<const bf16x8*>(k + kbase + s * 32 + kk0);
-          bf16x8 b_v = *reinterpret_cast<const bf16x8*>(v + kbase + s * 32 + kk0);
-          s_acc[half] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_q[s], b_k, s_acc[half], 0, 0, 0);
-          dp_acc[half] =
-              __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_do[s], b_v, dp_acc[half], 0, 0, 0);
+          s_acc[half] =
+              __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_q[s], bk_cur[half][s], s_acc[half], 0, 0, 0);
+          dp_acc[half] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_do[s], bv_cur[half][s],
+                                                                 dp_acc[half], 0, 0, 0);
         }
       }
       // dS = P * (dP - delta) * scale, written to my_ds [16][32]
@@ -171,6 +187,14 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_mfma_kernel(
         const size_t byte = tr_off(dh, k0 + (lane >> 4) * 8);
         bf16x8 kb = *reinterpret_cast<const bf16x8*>(reinterpret_cast<const char*>(kt) + byte);
         dq_acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsa, kb, dq_acc[f], 0, 0, 0);
+      }
+#pragma unroll
+      for (int half = 0; half < 2; ++half) {
+#pragma unroll
+        for (int s = 0; s < KS; ++s) {
+          bk_cur[half][s] = bk_nxt[half][s];
+          bv_cur[half][s] = bv_nxt[half][s];
+        }
       }
     }
 #pragma unroll
@@ -207,29 +231,45 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_mfma_kernel(
     }
     // causal: queries >= k0 contribute; start at the 32-tile containing k0
     const int qt_start = causal ? (k0 >> 5) : 0;
+    auto load_qdo = [&](bf16x8 (&bq)[2][KS], bf16x8 (&bdo)[2][KS], int qt32) {
+      const int kk0 = (lane >> 4) * 8;
+#pragma unroll
+      for (int half = 0; half < 2; ++half) {
+        const int qrow = (qt32 << 5) + half * 16 + (lane & 15);
+        const size_t qbase = base + (size_t)min(qrow, L - 1) * DH;
+#pragma unroll
+        for (int s = 0; s < KS; ++s) {
+          bq[half][s] = *reinterpret_cast<const bf16x8*>(q + qbase + s * 32 + kk0);
+          bdo[half][s] = *reinterpret_cast<const bf16x8*>(dout + qbase + s * 32 + kk0);
+        }
+      }
+    };
+    bf16x8 bq_cur[2][KS], bdo_cur[2][KS], bq_nxt[2][KS], bdo_nxt[2][KS];
+    if (qt_start < n_t32) load_qdo(bq_cur, bdo_cur, qt_start);
     for (int qtile = qt_start; qtile < n_t32; ++qtile) {
       const int qq0 = qtile << 5;
+      if (qtile + 1 < n_t32) load_qdo(bq_nxt, bdo_nxt, qtile + 1);
       f32x4 st_acc[2] = {f32x4{0.f, 0.f, 0.f, 0.f}, f32x4{0.f, 0.f, 0.f, 0.f}};
       f32x4 dpt_acc[2] = {f32x4{0.f, 0.f, 0.f, 0.f}, f32x4{0.f, 0.f, 0.f, 0.f}};
 #pragma unroll
       for (int half = 0; half < 2; ++half) {
-        const int qrow = qq0 + half * 16 + (lane & 15);
-        const int kk0 = (lane >> 4) * 8;
-        const size_t qbase = base + (size_t)min(qrow, L - 1) * DH;
 #pragma unroll
         for (int s = 0; s < KS; ++s) {
-          bf16x8 b_q = *reinterpret_cast<const bf16x8*>(q + qbase + s * 32 + kk0);
-          bf16x8 b_do = *reinterpret_cast<const bf16x8*>(dout + qbase + s * 32 + kk0);
           // S^T[k, q] = K . Q^T ; dP^T[k, q] = V . dO^T
-          st_acc[half] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_k[s], b_q, st_acc[half], 0, 0, 0);
-          dpt_acc[half] =
-              __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_v[s], b_do, dpt_acc[half], 0, 0, 0);
+          st_acc[half] =
+              __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_k[s], bq_cur[half][s], st_acc[half], 0, 0, 0);
+          dpt_acc[half] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_v[s], bdo_cur[half][s],
+                                                                  dpt_acc[half], 0, 0, 0);
         }
       }
-      // P^T and dS^T elementwise (rows = keys, cols = queries)
+      // P^T and dS^T elementwise (rows = keys, cols = queries); the per-query
+      // lse/delta are r-invariant — hoisted out of the r loop (a per-(r)
+      // global lse gather serialized 4 redundant loads per half)
 #pragma unroll
       for (int half = 0; half < 2; ++half) {
         const int qcol = qq0 + half * 16 + (lane & 15);
+        const float l_q = (qcol < L) ? lse[(size_t)bh * L + qcol] : 0.f;
+        const float d_q = (qcol < L) ? delta_s[qcol] : 0.f;
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int krow = k0 + (lane >> 4) * 4 + r;
@@ -237,8 +277,6 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_mfma_kernel(
               (qcol == krow) ||
               ((!causal || krow <= qcol) && krow < L && val_s[min(krow, L - 1)]);
           if (qcol >= L) allowed = false;
-          const float l_q = (qcol < L) ? lse[(size_t)bh * L + qcol] : 0.f;
-          const float d_q = (qcol < L) ? delta_s[qcol] : 0.f;
           float p = allowed ? __expf(st_acc[half][r] * scale - l_q) : 0.f;
           float ds_v = p * (dpt_acc[half][r] - d_q) * scale;
           const int idx = ((lane >> 4) * 4 + r) * 32 + half * 16 + (lane & 15);
@@ -258,6 +296,14 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_mfma_kernel(
             *reinterpret_cast<const bf16x8*>(reinterpret_cast<const char*>(dot_s) + byte);
         dk_acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsa, qb, dk_acc[f], 0, 0, 0);
         dv_acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, dob, dv_acc[f], 0, 0, 0);
+      }
+#pragma unroll
+      for (int half = 0; half < 2; ++half) {
+#pragma unroll
+        for (int s = 0; s < KS; ++s) {
+          bq_cur[half][s] = bq_nxt[half][s];
+          bdo_cur[half][s] = bdo_nxt[half][s];
+        }
       }
     }
 #pragma unroll
