@@ -27,20 +27,26 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 // C-layout row groups: row = (l>>4)*4 + r, col = l&15 -> the 16 lanes of a
-// row are those sharing l>>4, i.e. xor offsets 1,2,4,8 reduce over columns.
+// row are consecutive, i.e. one DPP "row".  The reduction runs on DPP
+// row_ror permutes (pure VALU, 0x120|N ctrl): the __shfl_xor form
+// compiled to ds_bpermute + a full lgkmcnt(0) drain per step — ~32
+// serialized LDS round trips per key tile in the online softmax
+// (measured dominant stall in the .s).  ror by 8/4/2/1 leaves every
+// lane holding the full 16-lane reduction, same as the xor ladder.
+#define ROW_ROR_DPP(x, N)   __builtin_amdgcn_update_dpp(0, (x), 0x120 | (N), 0xF, 0xF, true)
 __device__ __forceinline__ float group16_max(float v) {
-#pragma unroll
-  for (int off = 1; off < 16; off <<= 1) {
-    v = fmaxf(v, __shfl_xor(v, off, WAVE));
-  }
+  v = fmaxf(v, __int_as_float(ROW_ROR_DPP(__float_as_int(v), 8)));
+  v = fmaxf(v, __int_as_float(ROW_ROR_DPP(__float_as_int(v), 4)));
+  v = fmaxf(v, __int_as_float(ROW_ROR_DPP(__float_as_int(v), 2)));
+  v = fmaxf(v, __int_as_float(ROW_ROR_DPP(__float_as_int(v), 1)));
   return v;
 }
 
 __device__ __forceinline__ float group16_sum(float v) {
-#pragma unroll
-  for (int off = 1; off < 16; off <<= 1) {
-    v += __shfl_xor(v, off, WAVE);
-  }
+  v += __int_as_float(ROW_ROR_DPP(__float_as_int(v), 8));
+  v += __int_as_float(ROW_ROR_DPP(__float_as_int(v), 4));
+  v += __int_as_float(ROW_ROR_DPP(__float_as_int(v), 2));
+  v += __int_as_float(ROW_ROR_DPP(__float_as_int(v), 1));
   return v;
 }
 
