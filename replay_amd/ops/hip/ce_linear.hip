@@ -317,6 +317,12 @@ __global__ __launch_bounds__(256, 2) void ce_linear_bwd_kernel(
   };
 
   constexpr int A_KS = RESIDENT ? KSTEPS : 1;
+  // A fragments are TILE-INVARIANT; at E <= 128 they fit in registers for
+  // the whole walk.  The per-tile reload was not a latency problem (L1
+  // hits) but a STORE SERIALIZER: vmcnt retires in issue order, so a wait
+  // on an A load issued after the tile's dlogits stores must drain those
+  // stores too — one full store round trip per tile on the critical path.
+  constexpr bool A_HOIST = RESIDENT && (E <= 128);
   if constexpr (!RESIDENT) {
     const int row_q0 = blockIdx.x * 256;
     for (int i = threadIdx.x; i < 256 * (E * 2 / 16); i += blockDim.x) {
@@ -355,6 +361,22 @@ __global__ __launch_bounds__(256, 2) void ce_linear_bwd_kernel(
   const int Vi = (int)V;  // checked < 2^31 host-side; int index math
   const int n_tiles = (Vi + 63) >> 6;
   const int bk0 = (lane >> 4) * 8;
+  bf16x8 a_res[A_HOIST ? 4 : 1][A_HOIST ? KSTEPS : 1];
+  if constexpr (A_HOIST) {
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf) {
+      const int row = m0 + mf * 16 + (lane & 15);
+      const __hip_bfloat16* qr = hidden + (size_t)min(row, M - 1) * E + (lane >> 4) * 8;
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks) {
+        a_res[mf][ks] = *reinterpret_cast<const bf16x8*>(qr + ks * 32);
+      }
+      if (row >= M) {
+#pragma unroll
+        for (int ks = 0; ks < KSTEPS; ++ks) a_res[mf][ks] = bf16x8{0};
+      }
+    }
+  }
   // pipelined B stream: groups are (tile, half, ks); the address depends on
   // (tile, ks) only, so a prefetch landing on the other half of the same
   // tile is an L1 hit.  wait/busy was 31x with direct load-use.
@@ -403,8 +425,8 @@ __global__ __launch_bounds__(256, 2) void ce_linear_bwd_kernel(
     // B fragments are re-read per half from L1.
 #pragma unroll
     for (int half = 0; half < 2; ++half) {
-      bf16x8 a2[2][A_KS];
-      if constexpr (RESIDENT) {
+      bf16x8 a2[A_HOIST ? 1 : 2][A_KS];
+      if constexpr (RESIDENT && !A_HOIST) {
 #pragma unroll
         for (int m2 = 0; m2 < 2; ++m2) {
           const int row = m0 + (half * 2 + m2) * 16 + (lane & 15);
@@ -452,8 +474,13 @@ __global__ __launch_bounds__(256, 2) void ce_linear_bwd_kernel(
         for (int f = 0; f < 4; ++f) {
 #pragma unroll
           for (int m2 = 0; m2 < 2; ++m2) {
-            acc[m2][f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                a2[m2][RESIDENT ? ks : 0], b_frag[f], acc[m2][f], 0, 0, 0);
+            if constexpr (A_HOIST) {
+              acc[m2][f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  a_res[half * 2 + m2][ks], b_frag[f], acc[m2][f], 0, 0, 0);
+            } else {
+              acc[m2][f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  a2[m2][RESIDENT ? ks : 0], b_frag[f], acc[m2][f], 0, 0, 0);
+            }
           }
         }
       }
@@ -519,15 +546,21 @@ __global__ __launch_bounds__(256, 2) void ce_linear_bwd_kernel(
     }
     }
     if constexpr (WRITE_DL && DLROWS == 64) {
-      // single store sweep over ALL 64 bounce rows: one lgkm drain covers
-      // every epilogue ds_write of this tile
+      // single store sweep over ALL 64 bounce rows: all 8 ds_reads are
+      // issued into registers first (counted waits), then the 8 stores —
+      // the read->wait->store 1:1 form serialized 8 full lgkm drains
+      bf16x8 sw[8];
 #pragma unroll
-      for (int i = lane; i < 64 * 8; i += WAVE) {
+      for (int it = 0; it < 8; ++it) {
+        const int i = lane + it * WAVE;
+        sw[it] = *reinterpret_cast<const bf16x8*>(my_dl + dl_off(i / 8, (i % 8) * 8));
+      }
+#pragma unroll
+      for (int it = 0; it < 8; ++it) {
+        const int i = lane + it * WAVE;
         const int row = m0 + i / 8;
-        const int c0 = (i % 8) * 8;
         if (row < M) {
-          *reinterpret_cast<bf16x8*>(dlogits + (size_t)row * ldd + n0 + c0) =
-              *reinterpret_cast<const bf16x8*>(my_dl + dl_off(i / 8, c0));
+          *reinterpret_cast<bf16x8*>(dlogits + (size_t)row * ldd + n0 + (i % 8) * 8) = sw[it];
         }
       }
     }
