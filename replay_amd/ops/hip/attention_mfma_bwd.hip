@@ -168,11 +168,13 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_mfma_kernel(
 #pragma unroll
       for (int half = 0; half < 2; ++half) {
         const int col = k0 + half * 16 + (lane & 15);
+        // r-invariant hoists: the in-loop val_s byte read compiled to a
+        // ds_read_u8 + lgkmcnt(0) per (half, r)
+        const bool vcol = (col < L) && val_s[min(col, L - 1)];
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int row = q0 + (lane >> 4) * 4 + r;
-          bool allowed = (col == row) ||
-                         ((!causal || col <= row) && col < L && val_s[min(col, L - 1)]);
+          bool allowed = (col == row) || ((!causal || col <= row) && vcol);
           if (row >= L) allowed = false;
           float p = allowed ? __expf(s_acc[half][r] * scale - my_lse[r]) : 0.f;
           float ds_v = p * (dp_acc[half][r] - my_delta[r]) * scale;
@@ -229,6 +231,14 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_mfma_kernel(
       dk_acc[f] = f32x4{0.f, 0.f, 0.f, 0.f};
       dv_acc[f] = f32x4{0.f, 0.f, 0.f, 0.f};
     }
+    // qtile-INVARIANT per-key validity, hoisted: the in-loop byte read
+    // compiled to ds_read_u8 + lgkmcnt(0) at every (half, r)
+    bool vrow[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int krow = k0 + (lane >> 4) * 4 + r;
+      vrow[r] = (krow < L) && val_s[min(krow, L - 1)];
+    }
     // causal: queries >= k0 contribute; start at the 32-tile containing k0
     const int qt_start = causal ? (k0 >> 5) : 0;
     auto load_qdo = [&](bf16x8 (&bq)[2][KS], bf16x8 (&bdo)[2][KS], int qt32) {
@@ -249,6 +259,16 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_mfma_kernel(
     for (int qtile = qt_start; qtile < n_t32; ++qtile) {
       const int qq0 = qtile << 5;
       if (qtile + 1 < n_t32) load_qdo(bq_nxt, bdo_nxt, qtile + 1);
+      // clamped unconditional lse/delta loads issued BEFORE the MFMA block
+      // so their waits hide under it (the exec-masked conditional form
+      // waited vmcnt(0) at first use)
+      float lq2[2], dq2[2];
+#pragma unroll
+      for (int half = 0; half < 2; ++half) {
+        const int qcol = qq0 + half * 16 + (lane & 15);
+        lq2[half] = lse[(size_t)bh * L + min(qcol, L - 1)];
+        dq2[half] = delta_s[min(qcol, L - 1)];
+      }
       f32x4 st_acc[2] = {f32x4{0.f, 0.f, 0.f, 0.f}, f32x4{0.f, 0.f, 0.f, 0.f}};
       f32x4 dpt_acc[2] = {f32x4{0.f, 0.f, 0.f, 0.f}, f32x4{0.f, 0.f, 0.f, 0.f}};
 #pragma unroll
@@ -268,14 +288,12 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_mfma_kernel(
 #pragma unroll
       for (int half = 0; half < 2; ++half) {
         const int qcol = qq0 + half * 16 + (lane & 15);
-        const float l_q = (qcol < L) ? lse[(size_t)bh * L + qcol] : 0.f;
-        const float d_q = (qcol < L) ? delta_s[qcol] : 0.f;
+        const float l_q = (qcol < L) ? lq2[half] : 0.f;
+        const float d_q = (qcol < L) ? dq2[half] : 0.f;
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int krow = k0 + (lane >> 4) * 4 + r;
-          bool allowed =
-              (qcol == krow) ||
-              ((!causal || krow <= qcol) && krow < L && val_s[min(krow, L - 1)]);
+          bool allowed = (qcol == krow) || ((!causal || krow <= qcol) && vrow[r]);
           if (qcol >= L) allowed = false;
           float p = allowed ? __expf(st_acc[half][r] * scale - l_q) : 0.f;
           float ds_v = p * (dpt_acc[half][r] - d_q) * scale;
