@@ -191,3 +191,25 @@ def test_chunked_wgrad_linear_3d_and_module_swap():
     assert ffn.state_dict()["w1.weight"].shape == (16, 16)
     sw = SwiGLU(16)
     assert sw.state_dict()["WG.weight"].shape == (32, 16)
+
+
+def test_catalog_topk_cpu_chunked_with_seen_and_offset():
+    """The CPU/chunked path (no HIP ext): exact top-k with filter_seen and a
+    global item offset, vs a brute-force reference."""
+    import torch
+
+    from replay_amd.ops.topk import catalog_topk
+
+    torch.manual_seed(3)
+    B, V, E, k, off = 17, 300, 8, 5, 1000
+    q = torch.randn(B, E)
+    w = torch.randn(V, E)
+    seen = torch.randint(off, off + V, (B, 6))
+    s, i = catalog_topk(q, w, k, seen=seen, chunk_items=64, item_offset=off)
+    ref = q @ w.T
+    for b in range(B):
+        for sid in seen[b]:
+            ref[b, sid - off] = float("-inf")
+    rs, ri = torch.topk(ref, k, dim=1)
+    torch.testing.assert_close(s.float(), rs, rtol=1e-4, atol=1e-5)
+    assert torch.equal(i, ri + off)
